@@ -1,0 +1,43 @@
+"""AutoEnsembleEstimator: learn to ensemble a pool of models.
+
+Reference: adanet/autoensemble/estimator.py:28-220. An adanet Estimator
+whose Generator is synthesized from a candidate_pool of estimator-like
+objects (dict name->candidate, list, or callable(config[, iteration])).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from adanet_amd.autoensemble.common import _GeneratorFromCandidatePool
+from adanet_amd.core.estimator import Estimator
+
+
+class AutoEnsembleEstimator(Estimator):
+
+    def __init__(self, head, candidate_pool, max_iteration_steps,
+                 logits_fn=None, last_layer_fn=None, ensemblers=None,
+                 ensemble_strategies=None, evaluator=None,
+                 metric_fn=None, force_grow=False,
+                 adanet_loss_decay: float = 0.9, model_dir=None,
+                 config=None, max_iterations: Optional[int] = None,
+                 replay_config=None, **kwargs):
+        if not candidate_pool:
+            raise ValueError("candidate_pool can't be empty.")
+        generator = _GeneratorFromCandidatePool(candidate_pool,
+                                                logits_fn=logits_fn,
+                                                last_layer_fn=last_layer_fn)
+        super().__init__(head=head,
+                         subnetwork_generator=generator,
+                         max_iteration_steps=max_iteration_steps,
+                         ensemblers=ensemblers,
+                         ensemble_strategies=ensemble_strategies,
+                         evaluator=evaluator,
+                         metric_fn=metric_fn,
+                         force_grow=force_grow,
+                         adanet_loss_decay=adanet_loss_decay,
+                         model_dir=model_dir,
+                         config=config,
+                         max_iterations=max_iterations,
+                         replay_config=replay_config,
+                         **kwargs)

@@ -1,0 +1,511 @@
+"""_Iteration: one AdaNet round — all candidate subnetworks + ensembles
+training simultaneously.
+
+Reference: adanet/core/iteration.py (the _IterationBuilder/_Iteration/
+_TrainManager machinery, :40-1230) and adanet/core/ensemble_builder.py
+(_SubnetworkManager/_EnsembleBuilder, :258-805). The reference builds ONE
+TF graph per iteration and gates each candidate's train op with hooks; the
+MI355X-native engine is define-by-run:
+
+  * every training step runs each ACTIVE subnetwork's fwd+bwd+fused-optimizer
+    on its own HIP stream (overlapping candidates on the 256-CU chip),
+  * frozen previous-iteration subnetworks run ONCE per batch under no_grad
+    in eval mode (reference "dropout off for frozen members",
+    iteration.py:567-579) and their logits are cached in HBM for every
+    candidate ensemble's fused mixer call,
+  * candidate ensembles train only their mixture weights on the AdaNet
+    objective (reference weighted.py:606-617 var_list=mixture weights),
+  * per-candidate AdaNet losses accumulate in a device ring buffer and are
+    flushed to host EMAs in chunks, so the steady-state step issues no
+    host<->device synchronization.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import json
+import logging
+import math
+import os
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
+
+import torch
+from torch import nn
+
+from adanet_amd.core.architecture import _Architecture
+from adanet_amd.core.candidate import _Candidate
+from adanet_amd.core.summary import _ScopedSummary
+from adanet_amd.distributed import comm
+from adanet_amd.ensemble.strategy import Candidate as EnsembleCandidate
+from adanet_amd.subnetwork.generator import Builder, Subnetwork
+
+log = logging.getLogger("adanet_amd")
+
+_LOSS_FLUSH_STEPS = 16
+
+
+class _TrainManager(object):
+    """Persists per-spec "done training" state as JSON files.
+
+    Reference: adanet/core/iteration.py:40-118 — same on-disk layout
+    (<model_dir>/train_manager/t<t>/<spec_name>.json) so restarts and
+    preemptions resume correctly; only the chief writes.
+    """
+
+    def __init__(self, model_dir: Optional[str], iteration_number: int,
+                 is_chief: bool = True):
+        self._is_chief = is_chief
+        self._dir = None
+        self._stopped: Dict[str, str] = {}
+        if model_dir:
+            self._dir = os.path.join(model_dir, "train_manager",
+                                     "t{}".format(iteration_number))
+            if is_chief:
+                os.makedirs(self._dir, exist_ok=True)
+            if os.path.isdir(self._dir):
+                for fname in os.listdir(self._dir):
+                    if fname.endswith(".json"):
+                        name = fname[:-len(".json")]
+                        try:
+                            with open(os.path.join(self._dir, fname)) as f:
+                                payload = json.load(f)
+                        except (json.JSONDecodeError, OSError):
+                            payload = {}
+                        self._stopped[name] = payload.get("reason", "")
+
+    def should_train(self, spec_name: str) -> bool:
+        return spec_name not in self._stopped
+
+    def request_stop(self, spec_name: str, reason: str = ""):
+        if spec_name in self._stopped:
+            return
+        self._stopped[spec_name] = reason
+        if self._dir and self._is_chief:
+            path = os.path.join(self._dir, spec_name + ".json")
+            with open(path, "w") as f:
+                json.dump({"reason": reason}, f)
+
+    def is_over(self, spec_names: Sequence[str]) -> bool:
+        return all(n in self._stopped for n in spec_names)
+
+    @property
+    def stopped(self) -> Dict[str, str]:
+        return dict(self._stopped)
+
+
+@dataclasses.dataclass
+class _SubnetworkSpec:
+    """One candidate subnetwork under training (reference
+    adanet/core/ensemble_builder.py:586-641 _SubnetworkSpec)."""
+
+    name: str
+    builder: Builder
+    subnetwork: Optional[Subnetwork]  # None on non-owner ranks (round-robin)
+    optimizer: Optional[object]
+    owner_rank: int = 0
+    step: int = 0
+    last_loss: float = float("nan")
+    stream: Optional[object] = None
+    summary: Optional[_ScopedSummary] = None
+    # Private input pipeline for bagging (reference autoensemble/common.py:
+    # 43-56 _SecondaryTrainOpRunnerHook: the candidate trains on its own
+    # batches; ensembles see its outputs on the shared batch).
+    train_input_fn: Optional[Callable] = None
+    train_iter: Optional[object] = None
+    # per-batch cached outputs (detached)
+    out_logits: Optional[torch.Tensor] = None
+    out_last_layer: Optional[torch.Tensor] = None
+
+
+@dataclasses.dataclass
+class _EnsembleSpec:
+    """One candidate ensemble (reference ensemble_builder.py:43-255)."""
+
+    name: str
+    candidate: Optional[EnsembleCandidate]
+    ensemble: Optional[nn.Module]
+    ensembler_name: str
+    architecture: _Architecture
+    optimizer: Optional[object] = None
+    owner_rank: int = 0
+    step: int = 0
+    is_previous_best: bool = False
+    # member resolution: list of ("frozen", name) | ("new", subnetwork name)
+    members: Tuple = ()
+    summary: Optional[_ScopedSummary] = None
+    eval_loss: Optional[float] = None
+
+    @property
+    def new_subnetwork_names(self):
+        return [m[1] for m in self.members if m[0] == "new"]
+
+
+class _Iteration(object):
+    """All candidates of one AdaNet round, training in lockstep steps."""
+
+    def __init__(self, number: int, head, subnetwork_specs, ensemble_specs,
+                 frozen_subnetworks: Dict[str, Subnetwork], train_manager,
+                 max_iteration_steps: Optional[int], adanet_loss_decay: float,
+                 device: torch.device, placement, use_streams: bool = True,
+                 replicate_ensemble_in_training: bool = False,
+                 to_device: Optional[Callable] = None):
+        self.number = number
+        self.head = head
+        self.subnetwork_specs: List[_SubnetworkSpec] = list(subnetwork_specs)
+        self.ensemble_specs: List[_EnsembleSpec] = list(ensemble_specs)
+        self.candidates = [
+            _Candidate(spec, adanet_loss_decay) for spec in self.ensemble_specs
+        ]
+        self.frozen_subnetworks = dict(frozen_subnetworks)
+        self.train_manager = train_manager
+        self.max_iteration_steps = max_iteration_steps
+        self.device = device
+        self.placement = placement
+        self.replicate_ensemble_in_training = replicate_ensemble_in_training
+        self.to_device = to_device or (lambda f, l: (f, l))
+        self.step = 0
+        self._use_streams = use_streams and device.type == "cuda"
+        if self._use_streams:
+            for spec in self.subnetwork_specs:
+                if spec.subnetwork is not None:
+                    spec.stream = torch.cuda.Stream(device=device)
+        # Device ring buffer of per-candidate adanet losses + per-subnetwork
+        # losses: flushed to host EMAs every _LOSS_FLUSH_STEPS.
+        n = len(self.ensemble_specs) + len(self.subnetwork_specs)
+        self._loss_buf = torch.full((_LOSS_FLUSH_STEPS, max(n, 1)),
+                                    float("nan"), device=device,
+                                    dtype=torch.float32)
+        self._loss_buf_rows = 0
+        self._row_recorded: List[set] = []  # names actually written per row
+        self._frozen_event = None
+
+    # ------------------------------------------------------------------
+    # training
+    # ------------------------------------------------------------------
+
+    @property
+    def spec_names(self) -> List[str]:
+        return ([s.name for s in self.subnetwork_specs] +
+                [s.name for s in self.ensemble_specs])
+
+    def _owned(self, spec) -> bool:
+        return spec.owner_rank == comm.rank() or (
+            self.placement is not None and self.placement.data_parallel)
+
+    def compute_frozen_outputs(self, features, training: bool = False):
+        """Runs every frozen previous-iteration subnetwork once per batch
+        under no_grad; the resulting HBM-resident logits feed every
+        candidate's fused mixer call (north star: frozen logits cached in
+        288 GB HBM, never recomputed per candidate)."""
+        outputs = {}
+        with torch.no_grad():
+            for name, sub in self.frozen_subnetworks.items():
+                train_mode = training and self.replicate_ensemble_in_training
+                sub.module.train(train_mode)
+                last, logits = sub(features)
+                outputs[name] = (last, logits)
+        return outputs
+
+    def train_step(self, features, labels) -> None:
+        """One lockstep training step for every still-active spec
+        (the reference's single session.run over all candidate train ops,
+        iteration.py:779-804 + hooks)."""
+        tm = self.train_manager
+        frozen_out = self.compute_frozen_outputs(features, training=True)
+        if self._use_streams:
+            self._frozen_event = torch.cuda.Event()
+            self._frozen_event.record()
+
+        losses_row = {}
+        # --- subnetworks: fwd + own head loss + fused optimizer ---
+        for i, spec in enumerate(self.subnetwork_specs):
+            if spec.subnetwork is None:
+                continue  # not owned (round-robin)
+            active = tm.should_train(spec.name)
+            if self._use_streams:
+                ctx = torch.cuda.stream(spec.stream)
+                self._frozen_event.wait(spec.stream)
+            else:
+                ctx = _nullcontext()
+            with ctx:
+                spec.subnetwork.module.train(active)
+                private_batch = None
+                if active and spec.train_input_fn is not None:
+                    if spec.train_iter is None:
+                        spec.train_iter = iter(spec.train_input_fn())
+                    try:
+                        private_batch = next(spec.train_iter)
+                    except StopIteration:
+                        tm.request_stop(spec.name, "OutOfRange")
+                        active = False
+                if active:
+                    if private_batch is not None:
+                        pf, pl = private_batch
+                        pf, pl = self.to_device(pf, pl)
+                        last, logits = spec.subnetwork(pf)
+                        loss = self.head.loss(logits, pl)
+                    else:
+                        last, logits = spec.subnetwork(features)
+                        loss = self.head.loss(logits, labels)
+                    if spec.optimizer is not None:
+                        spec.optimizer.zero_grad(set_to_none=True)
+                        loss.backward()
+                        if (self.placement is not None
+                                and self.placement.data_parallel
+                                and comm.is_initialized()):
+                            comm.allreduce_gradients(
+                                list(spec.subnetwork.module.parameters()))
+                        spec.optimizer.step()
+                        sched = getattr(spec.optimizer, "_adanet_lr_sched",
+                                        None)
+                        if sched is not None:
+                            sched.step()
+                    spec.step += 1
+                    losses_row[spec.name] = loss.detach()
+                    if (self.max_iteration_steps is not None
+                            and spec.step >= self.max_iteration_steps):
+                        tm.request_stop(spec.name, "Training is over.")
+                else:
+                    with torch.no_grad():
+                        spec.subnetwork.module.eval()
+                        last, logits = spec.subnetwork(features)
+                if private_batch is not None:
+                    # Ensembles consume outputs on the SHARED batch
+                    # (reference common.py:146-180: model_fn invoked twice
+                    # when bagging).
+                    with torch.no_grad():
+                        was_training = spec.subnetwork.module.training
+                        spec.subnetwork.module.eval()
+                        last, logits = spec.subnetwork(features)
+                        spec.subnetwork.module.train(was_training)
+                spec.out_logits = logits.detach()
+                spec.out_last_layer = last.detach()
+
+        # --- candidate ensembles: mixture weights on the AdaNet objective ---
+        for spec in self.ensemble_specs:
+            if spec.ensemble is None:
+                continue  # not owned
+            if self._use_streams:
+                stream = self._member_stream(spec)
+                ctx = torch.cuda.stream(stream) if stream else _nullcontext()
+            else:
+                ctx = _nullcontext()
+            with ctx:
+                loss_t = self._ensemble_adanet_loss(spec, frozen_out, labels,
+                                                    train=True)
+                if loss_t is not None:
+                    losses_row[spec.name] = loss_t.detach()
+            spec.step += 1
+            if (self.max_iteration_steps is not None
+                    and spec.step >= self.max_iteration_steps):
+                tm.request_stop(spec.name, "Training is over.")
+
+        if self._use_streams:
+            # Device-side join: the default stream waits on every candidate
+            # stream (no host synchronization in the steady-state step).
+            cur = torch.cuda.current_stream(self.device)
+            for spec in self.subnetwork_specs:
+                if spec.stream is not None:
+                    cur.wait_stream(spec.stream)
+
+        self._record_losses(losses_row)
+        self.step += 1
+        if (self.max_iteration_steps is not None
+                and self.step >= self.max_iteration_steps):
+            for name in self.spec_names:
+                tm.request_stop(name, "Training is over.")
+
+    def _member_stream(self, spec: _EnsembleSpec):
+        for kind, name in spec.members:
+            if kind == "new":
+                for s in self.subnetwork_specs:
+                    if s.name == name and s.stream is not None:
+                        return s.stream
+        return None
+
+    def _gather_member_outputs(self, spec: _EnsembleSpec, frozen_out):
+        sub_logits, sub_last = [], []
+        for kind, name in spec.members:
+            if kind == "frozen":
+                last, logits = frozen_out[name]
+            else:
+                s = self._subnetwork_spec(name)
+                logits, last = s.out_logits, s.out_last_layer
+                if logits is None:
+                    return None, None
+            sub_logits.append(logits)
+            sub_last.append(last)
+        return sub_logits, sub_last
+
+    def _subnetwork_spec(self, name: str) -> _SubnetworkSpec:
+        """Lookup by spec name or by the builder's (candidate member) name."""
+        for s in self.subnetwork_specs:
+            if s.name == name or s.builder.name == name:
+                return s
+        raise KeyError(name)
+
+    def _ensemble_adanet_loss(self, spec: _EnsembleSpec, frozen_out, labels,
+                              train: bool):
+        """adanet_loss = head loss + complexity regularization
+        (reference ensemble_builder.py:420-426)."""
+        sub_logits, sub_last = self._gather_member_outputs(spec, frozen_out)
+        if sub_logits is None:
+            return None
+        train_mixture = (train and spec.optimizer is not None
+                         and not spec.is_previous_best
+                         and self.train_manager.should_train(spec.name))
+        if train_mixture:
+            logits = spec.ensemble.logits_from(sub_logits, sub_last)
+            loss = self.head.loss(logits, labels)
+            creg = spec.ensemble.complexity_regularization()
+            adanet_loss = loss + creg
+            spec.optimizer.zero_grad(set_to_none=True)
+            adanet_loss.backward()
+            spec.optimizer.step()
+            return adanet_loss
+        with torch.no_grad():
+            logits = spec.ensemble.logits_from(sub_logits, sub_last)
+            loss = self.head.loss(logits, labels)
+            creg = spec.ensemble.complexity_regularization()
+            return loss + creg
+
+    # ------------------------------------------------------------------
+    # loss ring buffer -> host EMAs
+    # ------------------------------------------------------------------
+
+    def _record_losses(self, losses_row: Dict[str, torch.Tensor]):
+        row = self._loss_buf_rows
+        names = self.spec_names
+        for i, name in enumerate(names):
+            t = losses_row.get(name)
+            if t is not None:
+                self._loss_buf[row, i] = t
+        self._row_recorded.append(set(losses_row.keys()))
+        self._loss_buf_rows += 1
+        if self._loss_buf_rows >= _LOSS_FLUSH_STEPS:
+            self.flush_losses()
+
+    def flush_losses(self):
+        """Reads the device loss buffer once and updates host-side EMAs /
+        NaN bookkeeping (reference _NanLossHook warns on NaN,
+        iteration.py:121-147; EMA in candidate.py:117-129). NaN for a
+        RECORDED spec means genuine divergence (the buffer's un-recorded
+        slots are also NaN but tracked separately)."""
+        if self._loss_buf_rows == 0:
+            return
+        rows = self._loss_buf[:self._loss_buf_rows].cpu().numpy()
+        n_sub = len(self.subnetwork_specs)
+        for r in range(rows.shape[0]):
+            recorded = self._row_recorded[r]
+            for i, spec in enumerate(self.subnetwork_specs):
+                if spec.name not in recorded:
+                    continue
+                v = float(rows[r, i])
+                spec.last_loss = v
+                if math.isnan(v):
+                    log.warning("'%s' diverged with loss = NaN.", spec.name)
+            for j, cand in enumerate(self.candidates):
+                if cand.ensemble_spec.name not in recorded:
+                    continue
+                # NaN while training poisons the EMA so selection surfaces
+                # divergence (reference iteration.py:1040-1046).
+                cand.update(float(rows[r, n_sub + j]))
+        self._loss_buf.fill_(float("nan"))
+        self._loss_buf_rows = 0
+        self._row_recorded = []
+
+    def is_over(self) -> bool:
+        return self.train_manager.is_over(self.spec_names)
+
+    # ------------------------------------------------------------------
+    # selection
+    # ------------------------------------------------------------------
+
+    def adanet_losses(self) -> List[float]:
+        """Per-candidate EMA adanet losses, all-gathered across owners in
+        round-robin placement."""
+        self.flush_losses()
+        local = {}
+        for i, cand in enumerate(self.candidates):
+            if cand.ensemble_spec.ensemble is not None:
+                local[i] = cand.adanet_loss
+        gathered = comm.all_gather_objects(local)
+        merged = {}
+        for d in gathered:
+            merged.update(d)
+        return [
+            merged.get(i, float("inf")) for i in range(len(self.candidates))
+        ]
+
+    def best_candidate_index(self,
+                             override: Optional[int] = None,
+                             losses: Optional[Sequence[float]] = None) -> int:
+        """argmin over adanet losses with NaN -> -inf (NaN candidates WIN so
+        divergence surfaces — reference iteration.py:1011-1047)."""
+        if override is not None:
+            return int(override)
+        if len(self.candidates) == 1:
+            return 0
+        vals = list(losses) if losses is not None else self.adanet_losses()
+        vals = [-float("inf") if math.isnan(v) else v for v in vals]
+        best, best_i = None, 0
+        for i, v in enumerate(vals):
+            if best is None or v < best:
+                best, best_i = v, i
+        return best_i
+
+    # ------------------------------------------------------------------
+    # evaluation
+    # ------------------------------------------------------------------
+
+    def evaluate_candidates(self, input_iter, steps: Optional[int],
+                            to_device: Callable) -> List[float]:
+        """Mean adanet_loss per candidate over shared eval batches
+        (reference evaluator.py:97-140: same batches for all candidates).
+        Round-robin: each rank evaluates the candidates it owns; results
+        are merged by adanet_losses()-style all-gather in the caller."""
+        sums = [0.0] * len(self.ensemble_specs)
+        count = 0
+        step = 0
+        while steps is None or step < steps:
+            try:
+                features, labels = next(input_iter)
+            except StopIteration:
+                break
+            features, labels = to_device(features, labels)
+            frozen_out = self.compute_frozen_outputs(features)
+            with torch.no_grad():
+                for spec in self.subnetwork_specs:
+                    if spec.subnetwork is None:
+                        continue
+                    spec.subnetwork.module.eval()
+                    last, logits = spec.subnetwork(features)
+                    spec.out_logits, spec.out_last_layer = logits, last
+                for i, spec in enumerate(self.ensemble_specs):
+                    if spec.ensemble is None:
+                        continue
+                    loss = self._ensemble_adanet_loss(spec, frozen_out,
+                                                      labels, train=False)
+                    if loss is not None:
+                        sums[i] += float(loss)
+            count += 1
+            step += 1
+        if count == 0:
+            return [float("nan")] * len(self.ensemble_specs)
+        out = []
+        for i, spec in enumerate(self.ensemble_specs):
+            if spec.ensemble is None:
+                out.append(float("nan"))
+            else:
+                spec.eval_loss = sums[i] / count
+                out.append(spec.eval_loss)
+        return out
+
+
+class _nullcontext(object):
+
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *a):
+        return False

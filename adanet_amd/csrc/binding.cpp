@@ -1,0 +1,63 @@
+// pybind11 bindings for the adanet_amd CDNA4 kernel library (_adanet_hip).
+// Built in-tree with torch.utils.cpp_extension under PYTORCH_ROCM_ARCH=gfx950
+// (see setup.py / __graft_entry__.build()).
+
+#include <torch/extension.h>
+
+void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+                  const c10::optional<at::Tensor>& bias, int64_t act);
+void transpose_bf16(const at::Tensor& in, at::Tensor& out);
+void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
+                      at::Tensor& loss, at::Tensor& probs, double eps);
+void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
+                      const at::Tensor& grad_rows, at::Tensor& dlogits,
+                      double eps);
+void mixer_fwd(const at::Tensor& ptrs, const at::Tensor& weights,
+               const c10::optional<at::Tensor>& bias, at::Tensor& out,
+               int64_t B, int64_t C, int64_t ldl, int64_t vector_mode);
+void mixer_bwd_dw(const at::Tensor& ptrs, const at::Tensor& dY, at::Tensor& dw,
+                  int64_t B, int64_t C, int64_t ldl, int64_t vector_mode);
+void mixer_bwd_dlogits(const at::Tensor& dY, const at::Tensor& weights,
+                       at::Tensor& dL, int64_t j, int64_t vector_mode);
+void fused_sgd(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,
+               const c10::optional<at::Tensor>& momentum_buf, double lr,
+               double momentum, double dampening, double weight_decay,
+               bool nesterov, double grad_scale);
+void fused_adam(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,
+                at::Tensor& m_buf, at::Tensor& v_buf, double lr, double beta1,
+                double beta2, double eps, double weight_decay, int64_t step,
+                double grad_scale);
+void layernorm_fwd(const at::Tensor& x, const c10::optional<at::Tensor>& gamma,
+                   const c10::optional<at::Tensor>& beta, at::Tensor& y,
+                   at::Tensor& mean, at::Tensor& rstd, double eps);
+void layernorm_bwd(const at::Tensor& x, const at::Tensor& dy,
+                   const c10::optional<at::Tensor>& gamma,
+                   const at::Tensor& mean, const at::Tensor& rstd,
+                   at::Tensor& dx, const c10::optional<at::Tensor>& dgamma,
+                   const c10::optional<at::Tensor>& dbeta);
+void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed);
+void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed);
+void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
+void colsum_bf16(const at::Tensor& x, at::Tensor& out);
+void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
+                    const c10::optional<at::Tensor>& pred, at::Tensor& correct);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "adanet_amd gfx950 (MI355X/CDNA4) kernels";
+  m.def("gemm_nt_bf16", &gemm_nt_bf16, "C[M,N] = A[M,K] @ B[N,K]^T, fused bias/relu");
+  m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_bwd", &softmax_xent_bwd);
+  m.def("mixer_fwd", &mixer_fwd);
+  m.def("mixer_bwd_dw", &mixer_bwd_dw);
+  m.def("mixer_bwd_dlogits", &mixer_bwd_dlogits);
+  m.def("fused_sgd", &fused_sgd);
+  m.def("fused_adam", &fused_adam);
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
+  m.def("relu_bwd", &relu_bwd);
+  m.def("colsum_bf16", &colsum_bf16);
+  m.def("argmax_correct", &argmax_correct);
+}

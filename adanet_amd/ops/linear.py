@@ -1,0 +1,157 @@
+"""HipLinear: bf16 dense layer on the fused MFMA GEMM (K1).
+
+Forward is one gemm_nt_bf16 call with the bias+ReLU epilogue fused into the
+GEMM (adanet_amd/csrc/gemm.hip). Backward runs two more MFMA GEMMs with
+K-minor operands materialized by the LDS-tiled transpose kernel, plus a
+column-sum for the bias gradient:
+
+    dX = dReLU(dY) @ W          (gemm_nt with W^T)
+    dW = dReLU(dY)^T @ X        (gemm_nt with dY^T, X^T)
+    db = colsum(dReLU(dY))
+
+The ReLU mask is recovered from the saved forward OUTPUT (y > 0), so the
+fused epilogue never materializes a mask tensor.
+
+Out-features are padded up to a multiple of 8 internally so every GEMM and
+logit buffer keeps an 8-element-aligned row stride (the gemm/global_load_lds
+fast path needs 16 B alignment); callers see the narrow [B, N] view.
+
+Reference dependency being replaced: tf.layers.dense in subnetworks
+(adanet/examples/simple_dnn.py:74-86).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+from torch import nn
+
+from adanet_amd.ops import _extension
+
+
+def _pad8(n: int) -> int:
+    return (n + 7) // 8 * 8
+
+
+def gemm_nt(a: torch.Tensor, b: torch.Tensor,
+            bias: Optional[torch.Tensor] = None,
+            activation: Optional[str] = None) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ B[N,K]^T (+bias, +relu). bf16 on GPU, fp32 on CPU."""
+    if a.is_cuda:
+        ext = _extension.require()
+        out = torch.empty((a.shape[0], b.shape[0]), device=a.device,
+                          dtype=torch.bfloat16)
+        ext.gemm_nt_bf16(a, b, out, bias, 1 if activation == "relu" else 0)
+        return out
+    out = a.float() @ b.float().t()
+    if bias is not None:
+        out = out + bias.float()
+    if activation == "relu":
+        out = torch.relu(out)
+    return out.to(a.dtype)
+
+
+def transpose2d(x: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        ext = _extension.require()
+        out = torch.empty((x.shape[1], x.shape[0]), device=x.device,
+                          dtype=x.dtype)
+        ext.transpose_bf16(x, out)
+        return out
+    return x.t().contiguous()
+
+
+class _LinearFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, weight, bias, activation):
+        x = x.contiguous()
+        y = gemm_nt(x, weight, bias, activation)
+        ctx.activation = activation
+        ctx.has_bias = bias is not None
+        ctx.save_for_backward(x, weight, y if activation == "relu" else None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if dy.is_cuda:
+            ext = _extension.require()
+            if ctx.activation == "relu":
+                dz = torch.empty_like(dy)
+                ext.relu_bwd(dy, y, dz)
+            else:
+                dz = dy
+            wt = transpose2d(weight)      # [K, N]
+            dzt = transpose2d(dz)         # [N, B]
+            xt = transpose2d(x)           # [K, B]
+            dx = gemm_nt(dz, wt)          # [B, K]
+            dw = gemm_nt(dzt, xt)         # [N, K]
+            db = None
+            if ctx.has_bias:
+                db = torch.empty((dz.shape[1],), device=dz.device,
+                                 dtype=torch.float32)
+                ext.colsum_bf16(dz, db)
+        else:
+            dzf = dy.float()
+            if ctx.activation == "relu":
+                dzf = dzf * (y > 0).float() if y is not None else dzf
+            dx = (dzf @ weight.float()).to(x.dtype)
+            dw = (dzf.t() @ x.float()).to(weight.dtype)
+            db = dzf.sum(dim=0) if ctx.has_bias else None
+        return dx, dw, db, None
+
+
+class HipLinear(nn.Module):
+    """bf16 Linear with fused bias+ReLU epilogue and padded out-features."""
+
+    def __init__(self, in_features: int, out_features: int, bias: bool = True,
+                 activation: Optional[str] = None, device=None,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        if activation not in (None, "relu"):
+            raise ValueError("activation must be None or 'relu'")
+        self.in_features = in_features
+        self.out_features = out_features
+        self.padded_out = _pad8(out_features)
+        self.activation = activation
+        self.weight = nn.Parameter(
+            torch.empty((self.padded_out, in_features), device=device,
+                        dtype=dtype))
+        if bias:
+            self.bias = nn.Parameter(
+                torch.zeros((self.padded_out,), device=device,
+                            dtype=torch.float32))
+        else:
+            self.register_parameter("bias", None)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        with torch.no_grad():
+            w = torch.empty((self.padded_out, self.in_features),
+                            dtype=torch.float32)
+            # Kaiming-uniform as torch.nn.Linear default.
+            bound = 1.0 / math.sqrt(self.in_features)
+            w.uniform_(-bound, bound)
+            if self.padded_out != self.out_features:
+                w[self.out_features:].zero_()
+            self.weight.copy_(w.to(self.weight.dtype))
+            if self.bias is not None:
+                self.bias.zero_()
+
+    def forward(self, x):
+        if x.dim() > 2:
+            x = x.reshape(x.shape[0], -1)
+        y = _LinearFn.apply(x.to(self.weight.dtype), self.weight, self.bias,
+                            self.activation)
+        if self.padded_out != self.out_features:
+            y = y[:, :self.out_features]
+        return y
+
+    def extra_repr(self):
+        return "in=%d, out=%d (padded %d), act=%s" % (
+            self.in_features, self.out_features, self.padded_out,
+            self.activation)

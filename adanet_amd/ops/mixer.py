@@ -1,0 +1,96 @@
+"""Fused weighted-sum-of-logits ensemble mixer (K5).
+
+out[b,c] = bias[c] + sum_j w_j (*) L_j[b,c] over the J member logit buffers
+in ONE kernel (csrc/mixer.hip); the frozen members' logits come from the
+iteration's HBM cache (no recompute). Autograd flows to the weights, the
+bias, and to any member logits that require grad (the newly-training
+subnetwork); frozen members get no gradient buffers at all.
+
+Reference behavior: adanet/ensemble/weighted.py:427-454,545-561.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+
+from adanet_amd.ops import _extension
+
+
+class _MixerFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, stacked_w, bias, vector_mode, *logits):
+        ext = _extension.require()
+        J = len(logits)
+        B, C = logits[0].shape
+        ldl = logits[0].stride(0)
+        ptrs = torch.tensor([t.data_ptr() for t in logits],
+                            dtype=torch.int64).to(logits[0].device,
+                                                  non_blocking=True)
+        out = torch.empty((B, C), device=logits[0].device,
+                          dtype=torch.bfloat16)
+        ext.mixer_fwd(ptrs, stacked_w, bias, out, B, C, ldl,
+                      1 if vector_mode else 0)
+        ctx.vector_mode = vector_mode
+        ctx.has_bias = bias is not None
+        ctx.logit_requires = [t.requires_grad for t in logits]
+        # Save the member tensors too: ptrs holds raw device addresses, so
+        # the buffers must stay alive until backward runs.
+        ctx.save_for_backward(stacked_w, ptrs, *logits)
+        ctx.shape = (B, C, ldl)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _extension.require()
+        stacked_w, ptrs = ctx.saved_tensors[:2]
+        B, C, ldl = ctx.shape
+        dy = dy.contiguous()
+        J = ptrs.numel()
+        dw = torch.zeros_like(stacked_w)
+        ext.mixer_bwd_dw(ptrs, dy, dw, B, C, ldl,
+                         1 if ctx.vector_mode else 0)
+        dbias = None
+        if ctx.has_bias:
+            dbias = torch.empty((C,), device=dy.device, dtype=torch.float32)
+            ext.colsum_bf16(dy, dbias)
+        dlogits = []
+        for j, req in enumerate(ctx.logit_requires):
+            if req:
+                dl = torch.empty((B, C), device=dy.device,
+                                 dtype=torch.bfloat16)
+                ext.mixer_bwd_dlogits(dy, stacked_w, dl, j,
+                                      1 if ctx.vector_mode else 0)
+                dlogits.append(dl)
+            else:
+                dlogits.append(None)
+        return (dw, dbias, None) + tuple(dlogits)
+
+
+def weighted_sum_logits(logits: Sequence[torch.Tensor],
+                        weights: Sequence[torch.Tensor],
+                        bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """bias + sum_j w_j * logits_j with scalar or per-class vector weights."""
+    assert len(logits) == len(weights) and len(logits) > 0
+    vector_mode = weights[0].dim() >= 1 and weights[0].numel() > 1
+    if logits[0].is_cuda:
+        # Stack weights (differentiable) into the flat fp32 buffer the
+        # kernel reads; autograd un-stacks dw back to each parameter.
+        stacked = torch.stack([w.reshape(-1) for w in weights]).contiguous()
+        logits = [l if (l.stride(-1) == 1 and l.dtype == torch.bfloat16
+                        and l.stride(0) == logits[0].stride(0))
+                  else l.to(torch.bfloat16).contiguous() for l in logits]
+        # All members must share a row stride for the pointer-table kernel.
+        if len({l.stride(0) for l in logits}) != 1:
+            logits = [l.contiguous() for l in logits]
+        return _MixerFn.apply(stacked, bias, vector_mode, *logits)
+    # CPU reference path (fp32).
+    total = None
+    for w, l in zip(weights, logits):
+        term = l.float() * w.float()
+        total = term if total is None else total + term
+    if bias is not None:
+        total = total + bias.float()
+    return total.to(logits[0].dtype)

@@ -1,0 +1,61 @@
+"""Fused softmax cross-entropy (K2) with label smoothing.
+
+GPU: one-pass fused kernel producing per-row fp32 loss + bf16 softmax probs
+(csrc/softmax_xent.hip); backward is closed-form (p - target) * grad.
+CPU: fp32 torch reference (the numerics oracle used by tests/).
+
+Replaces the reference head loss (adanet/core/ensemble_builder.py:571-583,
+tf.losses.softmax_cross_entropy with label_smoothing in
+research/improve_nas/trainer/improve_nas.py:160-181).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from adanet_amd.ops import _extension
+
+
+class _SoftmaxXentFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, logits, labels, label_smoothing):
+        ext = _extension.require()
+        B, C = logits.shape
+        loss = torch.empty((B,), device=logits.device, dtype=torch.float32)
+        probs = torch.empty((B, C), device=logits.device,
+                            dtype=torch.bfloat16)
+        ext.softmax_xent_fwd(logits, labels, loss, probs,
+                             float(label_smoothing))
+        ctx.save_for_backward(probs, labels)
+        ctx.label_smoothing = float(label_smoothing)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_rows):
+        probs, labels = ctx.saved_tensors
+        ext = _extension.require()
+        dlogits = torch.empty_like(probs)
+        ext.softmax_xent_bwd(probs, labels, grad_rows.contiguous().float(),
+                             dlogits, ctx.label_smoothing)
+        return dlogits, None, None
+
+
+def softmax_xent(logits: torch.Tensor, labels: torch.Tensor,
+                 label_smoothing: float = 0.0,
+                 reduction: str = "mean") -> torch.Tensor:
+    """Per-row (or reduced) softmax cross-entropy loss, fp32."""
+    if logits.is_cuda:
+        if logits.stride(1) != 1:
+            logits = logits.contiguous()
+        per_row = _SoftmaxXentFn.apply(logits, labels, label_smoothing)
+    else:
+        per_row = F.cross_entropy(logits.float(), labels,
+                                  label_smoothing=label_smoothing,
+                                  reduction="none")
+    if reduction == "mean":
+        return per_row.mean()
+    if reduction == "sum":
+        return per_row.sum()
+    return per_row
