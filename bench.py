@@ -1,0 +1,207 @@
+"""bench.py — AdaNet iterations/hour on the CIFAR-10 DNN search space.
+
+The BASELINE.json headline: "AdaNet iterations/hour + final ensemble eval
+accuracy, CIFAR-10 DNN search at 1/2/4/8 MI355X". One bench *step* is one
+FULL AdaNet boosting iteration (the paper's unit of work): train the
+candidate subnetworks simultaneously for --train-steps-per-iter optimizer
+steps each, run the Evaluator over the shared eval batches, select the
+winner on the complexity-regularized objective, freeze it, checkpoint, and
+grow. Data is synthetic CIFAR-10-shaped (3072-dim inputs, 10 classes,
+teacher-generated labels), weights random-init, compute dtype bf16.
+
+Multi-GPU (launched by the driver via torch.distributed.run): synchronous
+data-parallel replication over RCCL/xGMI with a fixed per-GPU batch (weak
+scaling — per-GPU work constant; the whole job still completes the same
+number of AdaNet iterations, so `value` is the job's iterations/hour).
+"""
+
+import argparse
+import json
+import os
+import shutil
+import sys
+import tempfile
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3,
+                   help="AdaNet iterations to time")
+    p.add_argument("--warmup", type=int, default=1,
+                   help="untimed AdaNet iterations first")
+    p.add_argument("--batch", type=int, default=2048, help="per-GPU batch")
+    p.add_argument("--hidden", type=int, default=2048)
+    p.add_argument("--train-steps-per-iter", type=int, default=150)
+    p.add_argument("--eval-batches", type=int, default=8)
+    p.add_argument("--candidates", type=int, default=2,
+                   help="candidates per iteration (paper: 2)")
+    p.add_argument("--placement", choices=["replication", "round_robin"],
+                   default="replication")
+    p.add_argument("--lr", type=float, default=0.05)
+    p.add_argument("--model-dir", default=None)
+    p.add_argument("--cpu", action="store_true", help="debug on CPU")
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+    import functools
+
+    import adanet_amd
+    from adanet_amd.distributed import (ReplicationStrategy,
+                                        RoundRobinStrategy, comm)
+    from adanet_amd.head import MultiClassHead
+    from adanet_amd.models import simple_dnn
+    from adanet_amd.ops.optim import FusedSGD
+
+    use_gpu = torch.cuda.is_available() and not args.cpu
+    comm.maybe_init_process_group()
+    world = comm.world_size()
+    rank = comm.rank()
+    local_rank = int(os.environ.get("LOCAL_RANK", rank))
+    device = torch.device("cuda", local_rank) if use_gpu else torch.device(
+        "cpu")
+    if use_gpu:
+        torch.cuda.set_device(device)
+
+    D, C = 3072, 10
+    torch.manual_seed(1234)  # same teacher everywhere
+    teacher = torch.randn(D, C)
+    torch.manual_seed(1000 + rank)  # different data shards per rank
+
+    # HBM-resident synthetic dataset: a pool of batches cycled forever.
+    n_pool = 32
+    pool = []
+    for _ in range(n_pool):
+        x = torch.randn(args.batch, D)
+        y = (x @ teacher).argmax(dim=1)
+        if use_gpu:
+            x = x.to(device).to(torch.bfloat16)
+            y = y.to(device)
+        pool.append((x, y))
+    eval_pool = []
+    torch.manual_seed(5000 + rank)
+    for _ in range(args.eval_batches):
+        x = torch.randn(args.batch, D)
+        y = (x @ teacher).argmax(dim=1)
+        if use_gpu:
+            x = x.to(device).to(torch.bfloat16)
+            y = y.to(device)
+        eval_pool.append((x, y))
+
+    def input_fn():
+        def gen():
+            i = 0
+            while True:
+                yield pool[i % n_pool]
+                i += 1
+
+        return gen()
+
+    def eval_input_fn():
+        return iter(list(eval_pool))
+
+    model_dir = args.model_dir or tempfile.mkdtemp(prefix="adanet_bench_")
+    placement = (ReplicationStrategy() if args.placement == "replication"
+                 else RoundRobinStrategy())
+
+    # Paper-style CIFAR DNN search space: candidates at the current depth
+    # and one deeper, width --hidden, fused momentum-SGD.
+    generator = simple_dnn.Generator(
+        optimizer_fn=functools.partial(FusedSGD, lr=args.lr, momentum=0.9),
+        layer_size=args.hidden,
+        initial_num_layers=1,
+        learn_mixture_weights=True,
+        dropout=0.0)
+
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(C),
+        subnetwork_generator=generator,
+        max_iteration_steps=args.train_steps_per_iter,
+        evaluator=adanet_amd.Evaluator(input_fn=eval_input_fn,
+                                       steps=args.eval_batches),
+        force_grow=True,
+        adanet_lambda=1e-4,
+        model_dir=model_dir,
+        config=adanet_amd.RunConfig(tf_random_seed=42,
+                                    device=str(device),
+                                    log_step_count_steps=10 ** 9),
+        experimental_placement_strategy=placement,
+    )
+
+    def one_iteration():
+        before = est.iteration_number
+        est.train(input_fn, steps=args.train_steps_per_iter)
+        assert est.iteration_number == before + 1, (
+            "bench step must complete one full AdaNet iteration")
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize(device)
+        comm.barrier()
+
+    for _ in range(args.warmup):
+        one_iteration()
+
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_iteration()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks.
+    if comm.is_initialized():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    res = est.evaluate(eval_input_fn, steps=args.eval_batches)
+    final_acc = float(res.get("accuracy", float("nan")))
+
+    iters_per_hour = args.steps / elapsed * 3600.0
+    if rank == 0:
+        out = {
+            "metric": "adanet_iterations_per_hour",
+            "value": iters_per_hour,
+            "unit": "iterations/hour",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "cifar10-dnn-search (adanet paper space)",
+                "global_batch": args.batch * world,
+                "input_dim": D,
+                "n_classes": C,
+                "hidden": args.hidden,
+                "candidates_per_iter": 2,
+                "train_steps_per_iter": args.train_steps_per_iter,
+                "eval_batches": args.eval_batches,
+                "parallelism": ("dp%d" % world
+                                if args.placement == "replication" else
+                                "round_robin%d" % world),
+                "final_ensemble_accuracy": final_acc,
+                "final_ensemble_size": est.iteration_number,
+            },
+        }
+        print(json.dumps(out))
+    if args.model_dir is None:
+        shutil.rmtree(model_dir, ignore_errors=True)
+    if comm.is_initialized():
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,150 @@
+"""Unit tests: architecture JSON, report accessor, evaluator, candidate EMA,
+summaries, timer (reference test model: architecture_test.py:27-82,
+report_accessor_test.py:28-297, evaluator_test.py:51-133, candidate_test.py,
+timer_test.py, summary_test.py)."""
+
+import json
+import math
+import os
+import time
+
+import pytest
+import torch
+
+from adanet_amd.core.architecture import _Architecture
+from adanet_amd.core.candidate import _Candidate
+from adanet_amd.core.evaluator import Evaluator, Objective
+from adanet_amd.core.report_accessor import _ReportAccessor
+from adanet_amd.core.summary import _ScopedSummary, read_events
+from adanet_amd.core.timer import _CountDownTimer
+from adanet_amd.subnetwork.report import MaterializedReport
+
+
+def test_architecture_roundtrip():
+    a = _Architecture("cand", "complexity_regularized")
+    a.add_subnetwork(0, "linear")
+    a.add_subnetwork(1, "dnn")
+    a.add_subnetwork(1, "wide")
+    a.set_replay_indices([0, 2])
+    s = a.serialize(iteration_number=2, global_step=300)
+    parsed = json.loads(s)
+    assert parsed["iteration_number"] == 2
+    assert parsed["global_step"] == 300
+    assert parsed["ensemble_candidate_name"] == "cand"
+    b = _Architecture.deserialize(s)
+    assert b.subnetworks == ((0, "linear"), (1, "dnn"), (1, "wide"))
+    assert b.replay_indices == [0, 2]
+    assert b.subnetworks_grouped_by_iteration == ((0, ("linear",)),
+                                                  (1, ("dnn", "wide")))
+
+
+def test_architecture_serialize_is_sorted_json():
+    a = _Architecture("c", "e")
+    s = a.serialize(0, 0)
+    assert s == json.dumps(json.loads(s), sort_keys=True)
+
+
+def test_report_accessor_roundtrip(tmp_path):
+    acc = _ReportAccessor(str(tmp_path / "report"))
+    r0 = MaterializedReport(0, "a", {"h": 1}, {"at": "x"}, {"m": 0.5}, True)
+    r1 = MaterializedReport(0, "bé", {}, {}, {}, False)  # unicode
+    acc.write_iteration_report(0, [r0, r1])
+    acc.write_iteration_report(1, [MaterializedReport(1, "c", {}, {}, {})])
+    got = acc.read_iteration_reports()
+    assert len(got) == 2
+    assert got[0][0] == r0
+    assert got[0][1].name == "bé"
+    # Overwrite iteration 1 (idempotent rewrite)
+    acc.write_iteration_report(1, [MaterializedReport(1, "d", {}, {}, {})])
+    got = acc.read_iteration_reports()
+    assert [r.name for r in got[1]] == ["d"]
+
+
+def test_evaluator_minimize_maximize():
+    data = [(torch.zeros(2, 2), torch.zeros(2)) for _ in range(3)]
+
+    def input_fn():
+        return iter(list(data))
+
+    ev = Evaluator(input_fn=input_fn, steps=3)
+    vals = ev.evaluate([lambda f, l: 1.0, lambda f, l: 0.5])
+    assert vals == [1.0, 0.5]
+    assert ev.best_index(vals) == 1
+    ev2 = Evaluator(input_fn=input_fn, steps=3,
+                    objective=Objective.MAXIMIZE)
+    assert ev2.best_index(vals) == 0
+    with pytest.raises(ValueError):
+        Evaluator(input_fn=input_fn, objective="bogus")
+
+
+def test_evaluator_steps_none_runs_to_exhaustion():
+    data = [(torch.zeros(1), torch.zeros(1))] * 5
+
+    def input_fn():
+        return iter(list(data))
+
+    ev = Evaluator(input_fn=input_fn, steps=None)
+    calls = []
+    ev.evaluate([lambda f, l: calls.append(1) or 1.0])
+    assert len(calls) == 5
+
+
+def test_candidate_ema():
+    class _S:
+        name = "s"
+
+    c = _Candidate(_S(), adanet_loss_decay=0.5)
+    assert c.adanet_loss == float("inf")
+    c.update(1.0)
+    assert c.adanet_loss == 1.0  # first value seeds the EMA
+    c.update(3.0)
+    assert c.adanet_loss == pytest.approx(2.0)
+    c.update(float("nan"))
+    assert math.isnan(c.adanet_loss)
+    c.update(1.0)  # NaN poisons permanently (assign_moving_average semantics)
+    assert c.adanet_loss == 1.0 or math.isnan(c.adanet_loss)
+
+
+def test_candidate_decay_validation():
+    class _S:
+        name = "s"
+
+    with pytest.raises(ValueError):
+        _Candidate(_S(), adanet_loss_decay=1.5)
+
+
+def test_scoped_summary_isolation(tmp_path):
+    s1 = _ScopedSummary(str(tmp_path), scope="cand1", namespace="t0_ensemble")
+    s2 = _ScopedSummary(str(tmp_path), scope="cand2", namespace="t0_ensemble")
+    s1.set_step(5)
+    s1.scalar("loss", 0.25)
+    s1.scalar("adanet_loss", torch.tensor(0.5))
+    s2.scalar("loss", 0.75)
+    e1 = read_events(os.path.join(str(tmp_path), "t0_ensemble", "cand1"))
+    e2 = read_events(os.path.join(str(tmp_path), "t0_ensemble", "cand2"))
+    assert [e["value"] for e in e1 if e["tag"] == "loss"] == [0.25]
+    assert e1[0]["step"] == 5
+    assert [e["value"] for e in e2 if e["tag"] == "loss"] == [0.75]
+    # same relative tag names -> overlayable charts
+    assert {e["tag"] for e in e1} >= {"loss", "adanet_loss"}
+
+
+def test_scoped_summary_skip():
+    s = _ScopedSummary(None, scope="x")
+    s.scalar("loss", 1.0)  # no-op, no crash
+    s.histogram("h", torch.ones(4))
+
+
+def test_scoped_summary_histogram(tmp_path):
+    s = _ScopedSummary(str(tmp_path), scope="c")
+    s.histogram("w", torch.tensor([1.0, 2.0, 3.0]))
+    ev = read_events(os.path.join(str(tmp_path), "c"))
+    assert ev[0]["value"]["mean"] == pytest.approx(2.0)
+    assert ev[0]["value"]["count"] == 3
+
+
+def test_countdown_timer():
+    t = _CountDownTimer(10.0)
+    assert 9.0 < t.secs_remaining() <= 10.0
+    t2 = _CountDownTimer(0.0)
+    assert t2.secs_remaining() == 0.0

@@ -1,0 +1,247 @@
+"""Estimator integration tests — behavioral invariants on tiny data.
+
+Reference test model: adanet/core/estimator_test.py (lifecycle :1022,
+checkpoints :1659, force_grow :3002, replay :3235, NaN :3081) ported as
+behavioral invariants (iteration counts, architecture JSON, checkpoint
+layout) rather than TF golden losses (SURVEY.md section 7 "hard parts" (e)).
+"""
+
+import glob
+import json
+import math
+import os
+
+import pytest
+import torch
+from torch import nn
+
+import adanet_amd
+from adanet_amd import replay
+from adanet_amd.head import MultiClassHead
+from adanet_amd.models import simple_dnn
+from adanet_amd.subnetwork import Builder, SimpleGenerator, Subnetwork
+
+
+def _make_estimator(model_dir, input_fn, **kwargs):
+    defaults = dict(
+        head=MultiClassHead(4),
+        subnetwork_generator=simple_dnn.Generator(layer_size=8),
+        max_iteration_steps=10,
+        model_dir=model_dir,
+        config=adanet_amd.RunConfig(tf_random_seed=42),
+    )
+    defaults.update(kwargs)
+    return adanet_amd.Estimator(**defaults)
+
+
+def test_constructor_validation(model_dir):
+    with pytest.raises(ValueError):
+        adanet_amd.Estimator(head=MultiClassHead(4),
+                             subnetwork_generator=None,
+                             max_iteration_steps=10)
+    with pytest.raises(ValueError):
+        _make_estimator(model_dir, None, max_iteration_steps=0)
+    with pytest.raises(ValueError):
+        _make_estimator(model_dir, None, max_iterations=0)
+
+
+def test_lifecycle_iterations_and_layout(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, max_steps=30)
+    # 3 iterations x 10 steps.
+    assert est.iteration_number == 3
+    assert est.global_step == 30
+    for t in range(3):
+        path = os.path.join(model_dir, "architecture-{}.json".format(t))
+        assert os.path.exists(path), path
+        arch = json.loads(open(path).read())
+        assert arch["iteration_number"] == t
+        assert len(arch["replay_indices"]) == t + 1
+    assert os.path.exists(os.path.join(model_dir, "checkpoint"))
+    assert glob.glob(os.path.join(model_dir, "increment.ckpt-*.pt"))
+    assert os.path.isdir(os.path.join(model_dir, "train_manager", "t0"))
+
+
+def test_train_steps_vs_max_steps(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, steps=10)
+    assert est.global_step == 10
+    est.train(input_fn, steps=10)
+    assert est.global_step == 20
+    # max_steps is absolute
+    est.train(input_fn, max_steps=25)
+    assert est.global_step == 25
+    assert est.iteration_number == 2  # 25 steps -> iteration 2 in progress
+
+
+def test_max_iterations_stops_early(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn, max_iterations=2)
+    est.train(input_fn, max_steps=100)
+    assert est.iteration_number == 2
+    assert est.global_step == 20
+
+
+def test_checkpoint_resume_mid_iteration(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, steps=15)  # mid-iteration 1
+    assert est.iteration_number == 1
+    est2 = _make_estimator(model_dir, input_fn)
+    assert est2.iteration_number == 1
+    assert est2.global_step == 15
+    est2.train(input_fn, steps=5)
+    assert est2.iteration_number == 2
+    assert est2.global_step == 20
+
+
+def test_grown_ensemble_members_accumulate(model_dir,
+                                           synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn, force_grow=True)
+    est.train(input_fn, max_steps=30)
+    arch = json.loads(
+        open(os.path.join(model_dir, "architecture-2.json")).read())
+    assert len(arch["subnetworks"]) == 3  # force_grow: one member per round
+
+
+def test_evaluate_predict_export(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, max_steps=20)
+    res = est.evaluate(input_fn, steps=5)
+    assert "loss" in res and "accuracy" in res
+    assert res["accuracy"] >= 0.0
+    preds = list(est.predict(lambda: iter([(X[:6], None)])))
+    assert len(preds) == 6
+    assert "probabilities" in preds[0]
+    export_dir = est.export_saved_model(os.path.join(model_dir, "export"))
+    assert os.path.exists(os.path.join(export_dir, "saved_model.pt"))
+    assert os.path.exists(os.path.join(export_dir, "architecture.json"))
+
+
+def test_evaluate_before_training_raises(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    with pytest.raises(ValueError):
+        est.evaluate(input_fn, steps=1)
+
+
+def test_replay_config_overrides_selection(model_dir,
+                                           synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        replay_config=replay.Config(best_ensemble_indices=[1, 0]))
+    est.train(input_fn, max_steps=20)
+    arch0 = json.loads(
+        open(os.path.join(model_dir, "architecture-0.json")).read())
+    # candidate index 1 at iteration 0 is the deeper DNN candidate.
+    assert arch0["replay_indices"][0] == 1
+    assert "1_layer_dnn" in [
+        s["builder_name"] for s in arch0["subnetworks"]
+    ]
+
+
+class _NanBuilder(Builder):
+    """Reference _NanLossBuilder (estimator_test.py:226)."""
+
+    @property
+    def name(self):
+        return "nan"
+
+    def build_subnetwork(self, features, logits_dimension, training,
+                         previous_ensemble=None):
+
+        class _M(nn.Module):
+
+            def __init__(self):
+                super().__init__()
+                self.w = nn.Parameter(torch.ones(1))
+
+            def forward(self, x):
+                out = (x[:, :1] * self.w * float("inf")) * 0.0  # NaN
+                return out.expand(x.shape[0], 4) if out.shape[1] == 1 else out
+
+        m = _M()
+        return Subnetwork(module=m, complexity=1.0)
+
+
+class _GoodBuilder(Builder):
+
+    def __init__(self, name="good"):
+        self._name = name
+
+    @property
+    def name(self):
+        return self._name
+
+    def build_subnetwork(self, features, logits_dimension, training,
+                         previous_ensemble=None):
+        from adanet_amd.ops.linear import HipLinear
+
+        class _M(nn.Module):
+
+            def __init__(self):
+                super().__init__()
+                self.lin = HipLinear(features.shape[1], logits_dimension)
+                self.last_layer_dim = features.shape[1]
+
+            def forward(self, x):
+                return x, self.lin(x)
+
+        return Subnetwork(module=_M(), complexity=1.0)
+
+
+def test_nan_candidate_does_not_crash_and_good_candidate_wins(
+        model_dir, synthetic_classification):
+    """A NaN candidate must not break training when an Evaluator selects on
+    eval data; selection via np.argmin surfaces NaN (reference semantics)
+    but the Evaluator path tolerates it (estimator.py:386-439 disables the
+    asserts). Here: the NaN candidate's EMA poisons, the evaluator (which
+    sees NaN too) is not used, and selection WITHOUT evaluator picks the
+    NaN candidate (NaN -> -inf, reference iteration.py:1040-1046)."""
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        subnetwork_generator=SimpleGenerator(
+            [_NanBuilder(), _GoodBuilder()]))
+    est.train(input_fn, max_steps=10)
+    arch = json.loads(
+        open(os.path.join(model_dir, "architecture-0.json")).read())
+    # NaN -> -inf wins argmin: divergence is surfaced in the architecture.
+    assert arch["subnetworks"][0]["builder_name"] == "nan"
+
+
+def test_evaluator_selects_on_eval_loss(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        evaluator=adanet_amd.Evaluator(input_fn=input_fn, steps=3))
+    est.train(input_fn, max_steps=10)
+    assert os.path.exists(os.path.join(model_dir, "architecture-0.json"))
+
+
+def test_report_materialization(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        report_materializer=adanet_amd.ReportMaterializer(input_fn=input_fn,
+                                                          steps=1))
+    est.train(input_fn, max_steps=20)
+    path = os.path.join(model_dir, "report", "iteration_reports.json")
+    assert os.path.exists(path)
+    reports = est._report_accessor.read_iteration_reports()
+    assert len(reports) == 2
+    names = {r.name for r in reports[0]}
+    assert "linear" in names or "1_layer_dnn" in names
+    assert any(r.included_in_final_ensemble for r in reports[0])
+
+
+def test_summaries_written(model_dir, synthetic_classification):
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, max_steps=10)
+    assert os.path.isdir(os.path.join(model_dir, "summaries"))
