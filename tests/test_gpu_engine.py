@@ -29,20 +29,26 @@ def test_gpu_lifecycle(tmp_path):
 
         return gen()
 
+    import functools
+
+    from adanet_amd.ops.optim import FusedSGD
+
     md = str(tmp_path / "model")
     est = adanet_amd.Estimator(
         head=MultiClassHead(C),
-        subnetwork_generator=simple_dnn.Generator(layer_size=64,
-                                                  learn_mixture_weights=True),
-        max_iteration_steps=20,
+        subnetwork_generator=simple_dnn.Generator(
+            optimizer_fn=functools.partial(FusedSGD, lr=0.2, momentum=0.9),
+            layer_size=256, learn_mixture_weights=True),
+        max_iteration_steps=40,
         evaluator=adanet_amd.Evaluator(input_fn=input_fn, steps=4),
         model_dir=md,
         config=adanet_amd.RunConfig(tf_random_seed=42),
     )
-    est.train(input_fn, max_steps=60)
+    est.train(input_fn, max_steps=120)
     assert est.iteration_number == 3
     res = est.evaluate(input_fn, steps=8)
-    assert res["accuracy"] > 0.3, res  # learnable synthetic task
+    # CPU fp32 reference of the identical config reaches ~0.98.
+    assert res["accuracy"] > 0.8, res
     arch = json.loads(res["architecture/adanet/ensembles"])
     assert len(arch["subnetworks"]) >= 1
     preds = list(est.predict(lambda: iter([(X[:4], None)])))
