@@ -508,18 +508,8 @@ class Estimator(object):
         return prev_ensemble, frozen
 
     def _restore_fp32_params(self, module):
-        """After casting a module to bf16 keep fp32 leaves fp32 (biases and
-        norm affines are fp32 by design in adanet_amd ops)."""
-        # HipLinear/HipLayerNorm declare fp32 params explicitly; a blanket
-        # .to(bf16) would downcast them, so rebuild their dtype here.
-        from adanet_amd.ops.layernorm import HipLayerNorm
-        from adanet_amd.ops.linear import HipLinear
-        for m in module.modules():
-            if isinstance(m, HipLinear) and m.bias is not None:
-                m.bias.data = m.bias.data.float()
-            if isinstance(m, HipLayerNorm) and m.weight is not None:
-                m.weight.data = m.weight.data.float()
-                m.bias.data = m.bias.data.float()
+        from adanet_amd.ops.linear import restore_fp32_params
+        restore_fp32_params(module)
 
     def _ensembler_by_name(self, name):
         for e in self._ensemblers:

@@ -395,6 +395,7 @@ class _Iteration(object):
             return
         rows = self._loss_buf[:self._loss_buf_rows].cpu().numpy()
         n_sub = len(self.subnetwork_specs)
+        base_step = self.step - self._loss_buf_rows
         for r in range(rows.shape[0]):
             recorded = self._row_recorded[r]
             for i, spec in enumerate(self.subnetwork_specs):
@@ -410,6 +411,19 @@ class _Iteration(object):
                 # NaN while training poisons the EMA so selection surfaces
                 # divergence (reference iteration.py:1040-1046).
                 cand.update(float(rows[r, n_sub + j]))
+        # TensorBoard-style per-candidate charts (the reference logs `loss`
+        # and `adanet_loss` under each candidate's scope,
+        # summary.py:262-296): one event per flush to keep IO off the step.
+        step = self.step
+        for i, spec in enumerate(self.subnetwork_specs):
+            if spec.summary is not None and not math.isnan(spec.last_loss):
+                spec.summary.set_step(step)
+                spec.summary.scalar("loss", spec.last_loss)
+        for spec, cand in zip(self.ensemble_specs, self.candidates):
+            if spec.summary is not None and cand.adanet_loss not in (
+                    float("inf"),):
+                spec.summary.set_step(step)
+                spec.summary.scalar("adanet_loss", cand.adanet_loss)
         self._loss_buf.fill_(float("nan"))
         self._loss_buf_rows = 0
         self._row_recorded = []

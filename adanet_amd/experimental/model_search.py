@@ -1,0 +1,25 @@
+"""ModelSearch: the ModelFlow entry point.
+
+Reference: adanet/experimental/keras/model_search.py:29-51.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+from adanet_amd.experimental.controllers import Controller
+from adanet_amd.experimental.schedulers import InProcessScheduler, Scheduler
+
+
+class ModelSearch(object):
+
+    def __init__(self, controller: Controller,
+                 scheduler: Optional[Scheduler] = None):
+        self._controller = controller
+        self._scheduler = scheduler or InProcessScheduler()
+
+    def run(self):
+        self._scheduler.schedule(self._controller.work_units())
+
+    def get_best_models(self, num_models: int = 1) -> List:
+        return self._controller.get_best_models(num_models)

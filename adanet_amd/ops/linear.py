@@ -35,6 +35,22 @@ def _pad8(n: int) -> int:
     return (n + 7) // 8 * 8
 
 
+def restore_fp32_params(module) -> None:
+    """Re-float the deliberately-fp32 leaves after a blanket .to(bf16).
+
+    HipLinear biases and HipLayerNorm affines are fp32 by design (they feed
+    fp32 epilogues/statistics); `module.to(torch.bfloat16)` downcasts them,
+    so call this afterwards.
+    """
+    from adanet_amd.ops.layernorm import HipLayerNorm
+    for m in module.modules():
+        if isinstance(m, HipLinear) and m.bias is not None:
+            m.bias.data = m.bias.data.float()
+        if isinstance(m, HipLayerNorm) and m.weight is not None:
+            m.weight.data = m.weight.data.float()
+            m.bias.data = m.bias.data.float()
+
+
 def gemm_nt(a: torch.Tensor, b: torch.Tensor,
             bias: Optional[torch.Tensor] = None,
             activation: Optional[str] = None) -> torch.Tensor:
