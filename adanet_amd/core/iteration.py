@@ -243,10 +243,12 @@ class _Iteration(object):
                         pf, pl = private_batch
                         pf, pl = self.to_device(pf, pl)
                         last, logits = spec.subnetwork(pf)
-                        loss = self.head.loss(logits, pl)
+                        loss = self._subnetwork_loss(spec, logits, pl, pf,
+                                                     frozen_out)
                     else:
                         last, logits = spec.subnetwork(features)
-                        loss = self.head.loss(logits, labels)
+                        loss = self._subnetwork_loss(spec, logits, labels,
+                                                     features, frozen_out)
                     if spec.optimizer is not None:
                         spec.optimizer.zero_grad(set_to_none=True)
                         loss.backward()
@@ -314,6 +316,46 @@ class _Iteration(object):
                 and self.step >= self.max_iteration_steps):
             for name in self.spec_names:
                 tm.request_stop(name, "Training is over.")
+
+    def _subnetwork_loss(self, spec, logits, labels, features, frozen_out):
+        """Head loss, or the builder's custom loss hook when provided
+        (enables knowledge distillation: the improve_nas search space adds
+        a distillation term against the previous ensemble's logits —
+        reference research/improve_nas/trainer/improve_nas.py:41-59,
+        166-181)."""
+        hook = getattr(spec.builder, "build_subnetwork_loss", None)
+        if hook is None:
+            return self.head.loss(logits, labels)
+
+        def frozen_outputs():
+            if spec.train_input_fn is not None:
+                # Private (bagging) batch: the shared-batch frozen cache
+                # doesn't apply — run the frozen members on this batch.
+                return self.compute_frozen_outputs(features)
+            return frozen_out
+
+        def prev_ensemble_logits():
+            prev = self.previous_best_spec
+            if prev is None or prev.ensemble is None:
+                return None
+            sub_logits, sub_last = self._gather_member_outputs(
+                prev, frozen_outputs())
+            if sub_logits is None:
+                return None
+            with torch.no_grad():
+                return prev.ensemble.logits_from(sub_logits, sub_last)
+
+        return hook(head=self.head, logits=logits, labels=labels,
+                    features=features,
+                    previous_ensemble_logits_fn=prev_ensemble_logits,
+                    frozen_outputs_fn=frozen_outputs)
+
+    @property
+    def previous_best_spec(self) -> Optional[_EnsembleSpec]:
+        for spec in self.ensemble_specs:
+            if spec.is_previous_best:
+                return spec
+        return None
 
     def _member_stream(self, spec: _EnsembleSpec):
         for kind, name in spec.members:
