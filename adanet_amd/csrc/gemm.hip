@@ -5,14 +5,16 @@
 // tf.layers.dense / tf.matmul in adanet/examples/simple_dnn.py:74-86 and
 // the MATRIX mixture weights in adanet/ensemble/weighted.py:449) —
 // re-designed CDNA4-native rather than ported:
-//   * 128x128 output tile, BK=32, 256 threads = 4 waves (2x2), each wave
-//     computing a 64x64 sub-tile as 4x4 MFMA f32_16x16x32_bf16 fragments.
-//   * double-buffered LDS staged with global_load_lds (16 B per lane,
-//     wave-uniform LDS base: the direct HBM->LDS path, no VGPR round-trip).
+//   * templated tile config: 128x128 (4 waves, 64x64/wave, 4x4 MFMA
+//     fragments) for large problems; 64x64 (4 waves, 32x32/wave) when the
+//     128-tile grid would underfill the 256-CU chip — the launch needs
+//     well over 256 workgroups before the big tile pays.
+//   * BK=32 K-steps staged double-buffered with global_load_lds (16 B per
+//     lane, wave-uniform LDS base: the direct HBM->LDS path).
 //   * one s_barrier per K-tile; the compiler's vmcnt drain at the barrier
-//     makes the staged tile visible (m97 structure from the CDNA4 guide).
+//     publishes the staged tile (the m97 structure from the CDNA4 guide).
 //   * XCD-aware bijective blockIdx swizzle so neighboring output tiles
-//     share a chiplet-local L2 (8 XCDs on MI355X).
+//     share a chiplet-local L2 (8 XCDs).
 // All A/B/C layouts are row-major with the reduction dim (K) minor; the
 // python wrappers materialize transposes for the backward GEMMs with the
 // LDS-tiled transpose kernel (transpose.hip).
@@ -21,21 +23,20 @@
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
-#define BM 128
-#define BN 128
 #define BK 32
 #define THREADS 256
 
 typedef s16x8 frag_ab;
 
+// Stage a [ROWS x BK] bf16 tile into LDS: each wave issues ROWS/16/(4 waves)
+// (or strided) 1 KiB global_load_lds ops (64 lanes x 16 B, LDS-linear).
+template <int ROWS>
 __device__ __forceinline__ void stage_tile_nt(
     const bf16_t* __restrict__ G, int ld, int tile_row0, int max_row, int k0,
-    bf16_t* __restrict__ lds /* [128*32] */, int wid, int lane) {
-  // Each wave stages two 16-row segments (16 rows x 32 cols bf16 = 1 KiB
-  // per global_load_lds: 64 lanes x 16 B, LDS-linear).
+    bf16_t* __restrict__ lds, int wid, int lane) {
+  constexpr int SEGMENTS = ROWS / 16;
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int seg = wid * 2 + i;
+  for (int seg = wid; seg < SEGMENTS; seg += 4) {
     const int row_in_tile = seg * 16 + (lane >> 2);
     int grow = tile_row0 + row_in_tile;
     grow = grow < max_row ? grow : max_row - 1;  // clamp; masked on C-store
@@ -47,14 +48,16 @@ __device__ __forceinline__ void stage_tile_nt(
   }
 }
 
-__global__ __launch_bounds__(THREADS, 2) void gemm_nt_bf16_kernel(
+// BM x BN tile, 2x2 wave grid, FM x FN 16x16 MFMA fragments per wave.
+template <int BM, int BN, int FM, int FN, int MINWAVES>
+__global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles) {
   __shared__ bf16_t As[2][BM * BK];
   __shared__ bf16_t Bs[2][BN * BK];
 
-  // Bijective XCD-aware swizzle (guide m204): contiguous tile chunks per XCD.
+  // Bijective XCD-aware swizzle (guide m204): contiguous tile chunks/XCD.
   const int nwg = mtiles * ntiles;
   const int orig = blockIdx.x;
   const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
@@ -63,65 +66,64 @@ __global__ __launch_bounds__(THREADS, 2) void gemm_nt_bf16_kernel(
   const int tile_m = wg / ntiles, tile_n = wg % ntiles;
 
   const int tid = threadIdx.x;
-  const int wid = tid >> 6;          // wave 0..3
+  const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int wm = wid >> 1, wn = wid & 1;  // 2x2 wave grid, 64x64 out each
+  const int wm = wid >> 1, wn = wid & 1;  // 2x2 wave grid
 
   const int row0 = tile_m * BM;
   const int col0 = tile_n * BN;
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < FM; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int ktiles = K / BK;
-  stage_tile_nt(A, lda, row0, M, 0, As[0], wid, lane);
-  stage_tile_nt(B, ldb, col0, N, 0, Bs[0], wid, lane);
+  stage_tile_nt<BM>(A, lda, row0, M, 0, As[0], wid, lane);
+  stage_tile_nt<BN>(B, ldb, col0, N, 0, Bs[0], wid, lane);
 
   int buf = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();  // staged tile `buf` visible; prior reads of buf^1 done
     if (kt + 1 < ktiles) {
       const int k0 = (kt + 1) * BK;
-      stage_tile_nt(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
-      stage_tile_nt(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
+      stage_tile_nt<BM>(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
+      stage_tile_nt<BN>(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
     }
-    // Fragment loads: lane reads 8 contiguous bf16 (16 B -> ds_read_b128).
-    // A-frag row = wm*64 + fm*16 + (lane&15); k = (lane>>4)*8.
-    frag_ab a[4], b[4];
+    // Fragment loads: 8 contiguous bf16 per lane -> ds_read_b128.
+    frag_ab a[FM], b[FN];
     const int kofs = (lane >> 4) * 8;
-    const int arow = wm * 64 + (lane & 15);
-    const int brow = wn * 64 + (lane & 15);
+    const int arow = wm * (FM * 16) + (lane & 15);
+    const int brow = wn * (FN * 16) + (lane & 15);
 #pragma unroll
-    for (int f = 0; f < 4; ++f) {
+    for (int f = 0; f < FM; ++f)
       a[f] = *(const frag_ab*)&As[buf][(arow + f * 16) * BK + kofs];
+#pragma unroll
+    for (int f = 0; f < FN; ++f)
       b[f] = *(const frag_ab*)&Bs[buf][(brow + f * 16) * BK + kofs];
-    }
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int i = 0; i < FM; ++i)
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
+      for (int j = 0; j < FN; ++j)
         acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
                                                             acc[i][j], 0, 0, 0);
     buf ^= 1;
   }
 
-  // Epilogue: C/D fragment layout (16x16x32): col = lane&15,
-  // row = (lane>>4)*4 + reg. Fused bias + ReLU, bf16 store with guards.
+  // Epilogue: C/D layout (16x16x32): col = lane&15, row = (lane>>4)*4 + reg.
   const int c_col_in_frag = lane & 15;
   const int c_row_base = (lane >> 4) * 4;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < FM; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int col = col0 + wn * 64 + j * 16 + c_col_in_frag;
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
       if (col >= N) continue;
       const float bv = bias ? bias[col] : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
-        const int row = row0 + wm * 64 + i * 16 + c_row_base + rr;
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
         if (row >= M) continue;
         float v = acc[i][j][rr] + bv;
         if (act == 1) v = v > 0.f ? v : 0.f;
@@ -178,10 +180,22 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
 
   const bool fast = (K % BK == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
   if (fast) {
-    const int mtiles = (M + BM - 1) / BM, ntiles = (N + BN - 1) / BN;
-    hipLaunchKernelGGL(gemm_nt_bf16_kernel, dim3(mtiles * ntiles),
-                       dim3(THREADS), 0, stream.stream(), a, b, c, bias_ptr, M,
-                       N, K, lda, ldb, ldc, (int)act, mtiles, ntiles);
+    const int mt128 = (M + 127) / 128, nt128 = (N + 127) / 128;
+    // 64-tile when the 128-tile grid underfills the chip (256 CUs want
+    // >=2 blocks each before the big tile's per-wave efficiency wins).
+    const bool small = (int64_t)mt128 * nt128 < 384;
+    if (small) {
+      const int mt = (M + 63) / 64, nt = (N + 63) / 64;
+      hipLaunchKernelGGL((gemm_nt_bf16_kernel<64, 64, 2, 2, 4>),
+                         dim3(mt * nt), dim3(THREADS), 0, stream.stream(), a,
+                         b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt,
+                         nt);
+    } else {
+      hipLaunchKernelGGL((gemm_nt_bf16_kernel<128, 128, 4, 4, 2>),
+                         dim3(mt128 * nt128), dim3(THREADS), 0,
+                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,
+                         ldb, ldc, (int)act, mt128, nt128);
+    }
   } else {
     const int64_t total = (int64_t)M * N;
     const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);

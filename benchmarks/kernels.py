@@ -32,9 +32,10 @@ def main():
     dev = "cuda:0"
     out = {}
 
-    shapes = [(4096, 4096, 4096), (2048, 2048, 3072), (8192, 8192, 8192)]
+    shapes = [(4096, 4096, 4096), (2048, 2048, 3072), (2048, 3072, 2048),
+              (4096, 2048, 3072), (8192, 8192, 8192)]
     if args.quick:
-        shapes = shapes[:2]
+        shapes = shapes[:4]
     for (M, N, K) in shapes:
         a = torch.randn(M, K, device=dev).to(torch.bfloat16)
         b = torch.randn(N, K, device=dev).to(torch.bfloat16)
