@@ -27,6 +27,14 @@ void fused_adam(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,
                 at::Tensor& m_buf, at::Tensor& v_buf, double lr, double beta1,
                 double beta2, double eps, double weight_decay, int64_t step,
                 double grad_scale);
+void fused_sgd_fp32(at::Tensor& param, const at::Tensor& grad,
+                    const c10::optional<at::Tensor>& momentum_buf, double lr,
+                    double momentum, double dampening, double weight_decay,
+                    bool nesterov, double grad_scale);
+void fused_adam_fp32(at::Tensor& param, const at::Tensor& grad,
+                     at::Tensor& m_buf, at::Tensor& v_buf, double lr,
+                     double beta1, double beta2, double eps,
+                     double weight_decay, int64_t step, double grad_scale);
 void layernorm_fwd(const at::Tensor& x, const c10::optional<at::Tensor>& gamma,
                    const c10::optional<at::Tensor>& beta, at::Tensor& y,
                    at::Tensor& mean, at::Tensor& rstd, double eps);
@@ -53,6 +61,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mixer_bwd_dlogits", &mixer_bwd_dlogits);
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_adam", &fused_adam);
+  m.def("fused_sgd_fp32", &fused_sgd_fp32);
+  m.def("fused_adam_fp32", &fused_adam_fp32);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("dropout_fwd", &dropout_fwd);

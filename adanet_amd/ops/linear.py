@@ -32,7 +32,11 @@ from adanet_amd.ops import _extension
 
 
 def _pad8(n: int) -> int:
-    return (n + 7) // 8 * 8
+    # Pad out-features to a multiple of 32: keeps the row stride 16B-aligned
+    # for global_load_lds AND keeps N usable as the reduction dim of the
+    # backward dX GEMM (the MFMA fast path needs K % 32 == 0 — a 16-wide
+    # padded head sent dX to the generic VALU kernel, 7% of step time).
+    return (n + 31) // 32 * 32
 
 
 def restore_fp32_params(module) -> None:
@@ -94,6 +98,8 @@ class _LinearFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, weight, y = ctx.saved_tensors
         dy = dy.contiguous()
+        need_dx = ctx.needs_input_grad[0]
+        need_dw = ctx.needs_input_grad[1]
         if dy.is_cuda:
             ext = _extension.require()
             if ctx.activation == "relu":
@@ -101,11 +107,14 @@ class _LinearFn(torch.autograd.Function):
                 ext.relu_bwd(dy, y, dz)
             else:
                 dz = dy
-            wt = transpose2d(weight)      # [K, N]
-            dzt = transpose2d(dz)         # [N, B]
-            xt = transpose2d(x)           # [K, B]
-            dx = gemm_nt(dz, wt)          # [B, K]
-            dw = gemm_nt(dzt, xt)         # [N, K]
+            dx = dw = None
+            if need_dx:
+                wt = transpose2d(weight)  # [K, N]
+                dx = gemm_nt(dz, wt)      # [B, K]
+            if need_dw:
+                dzt = transpose2d(dz)     # [N, B]
+                xt = transpose2d(x)       # [K, B]
+                dw = gemm_nt(dzt, xt)     # [N, K]
             db = None
             if ctx.has_bias:
                 db = torch.empty((dz.shape[1],), device=dz.device,
@@ -115,8 +124,8 @@ class _LinearFn(torch.autograd.Function):
             dzf = dy.float()
             if ctx.activation == "relu":
                 dzf = dzf * (y > 0).float() if y is not None else dzf
-            dx = (dzf @ weight.float()).to(x.dtype)
-            dw = (dzf.t() @ x.float()).to(weight.dtype)
+            dx = (dzf @ weight.float()).to(x.dtype) if need_dx else None
+            dw = (dzf.t() @ x.float()).to(weight.dtype) if need_dw else None
             db = dzf.sum(dim=0) if ctx.has_bias else None
         return dx, dw, db, None
 

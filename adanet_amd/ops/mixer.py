@@ -70,15 +70,27 @@ class _MixerFn(torch.autograd.Function):
 
 
 def weighted_sum_logits(logits: Sequence[torch.Tensor],
-                        weights: Sequence[torch.Tensor],
+                        weights,
                         bias: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """bias + sum_j w_j * logits_j with scalar or per-class vector weights."""
-    assert len(logits) == len(weights) and len(logits) > 0
-    vector_mode = weights[0].dim() >= 1 and weights[0].numel() > 1
+    """bias + sum_j w_j * logits_j with scalar or per-class vector weights.
+
+    ``weights`` is either the ensemble's FLAT parameter ([J] scalar mode /
+    [J, C] vector mode — the zero-copy fast path) or a list of per-member
+    weight tensors (stacked here; autograd un-stacks dw back to each).
+    """
+    assert len(logits) > 0
+    if torch.is_tensor(weights):
+        stacked = weights
+        assert stacked.shape[0] == len(logits)
+        vector_mode = stacked.dim() == 2 and stacked.shape[1] > 1
+    else:
+        assert len(logits) == len(weights)
+        vector_mode = weights[0].dim() >= 1 and weights[0].numel() > 1
+        stacked = torch.stack([w.reshape(-1) for w in weights])
+        if not vector_mode:
+            stacked = stacked.reshape(len(weights))
     if logits[0].is_cuda:
-        # Stack weights (differentiable) into the flat fp32 buffer the
-        # kernel reads; autograd un-stacks dw back to each parameter.
-        stacked = torch.stack([w.reshape(-1) for w in weights]).contiguous()
+        stacked = stacked.contiguous()
         logits = [l if (l.stride(-1) == 1 and l.dtype == torch.bfloat16
                         and l.stride(0) == logits[0].stride(0))
                   else l.to(torch.bfloat16).contiguous() for l in logits]
@@ -88,7 +100,8 @@ def weighted_sum_logits(logits: Sequence[torch.Tensor],
         return _MixerFn.apply(stacked, bias, vector_mode, *logits)
     # CPU reference path (fp32).
     total = None
-    for w, l in zip(weights, logits):
+    for j, l in enumerate(logits):
+        w = stacked[j]
         term = l.float() * w.float()
         total = term if total is None else total + term
     if bias is not None:

@@ -82,6 +82,20 @@ class FusedSGD(torch.optim.Optimizer):
                                   group["lr"], group["momentum"],
                                   group["dampening"], group["weight_decay"],
                                   group["nesterov"], 1.0)
+                elif p.is_cuda and p.dtype == torch.float32:
+                    ext = _extension.require()
+                    mom = None
+                    if group["momentum"]:
+                        mom = state.get("momentum_buffer")
+                        if mom is None:
+                            mom = torch.zeros_like(p.data)
+                            state["momentum_buffer"] = mom
+                    ext.fused_sgd_fp32(p.data,
+                                       p.grad.float().contiguous(), mom,
+                                       group["lr"], group["momentum"],
+                                       group["dampening"],
+                                       group["weight_decay"],
+                                       group["nesterov"], 1.0)
                 else:
                     _fp32_sgd_update(p, p.grad, state, group["lr"],
                                      group["momentum"], group["dampening"],
@@ -122,6 +136,16 @@ class FusedAdam(torch.optim.Optimizer):
                                    state["exp_avg"], state["exp_avg_sq"],
                                    group["lr"], b1, b2, group["eps"],
                                    group["weight_decay"], t, 1.0)
+                elif p.is_cuda and p.dtype == torch.float32:
+                    ext = _extension.require()
+                    if "exp_avg" not in state:
+                        state["exp_avg"] = torch.zeros_like(p.data)
+                        state["exp_avg_sq"] = torch.zeros_like(p.data)
+                    ext.fused_adam_fp32(p.data, p.grad.float().contiguous(),
+                                        state["exp_avg"],
+                                        state["exp_avg_sq"], group["lr"], b1,
+                                        b2, group["eps"],
+                                        group["weight_decay"], t, 1.0)
                 else:
                     master = state.get("master")
                     if master is None:

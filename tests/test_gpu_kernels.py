@@ -209,6 +209,47 @@ def test_fused_adam_vs_fp32():
     assert (master - p_ref.detach()).abs().max() < 1e-3
 
 
+def test_fused_sgd_fp32_matches_torch():
+    ext = _ext()
+    torch.manual_seed(0)
+    p = torch.randn(513, device=DEV, dtype=torch.float32)
+    p_ref = torch.nn.Parameter(p.clone())
+    opt = torch.optim.SGD([p_ref], lr=0.1, momentum=0.9, weight_decay=0.01)
+    mom = torch.zeros_like(p)
+    for _ in range(4):
+        g = torch.randn(513, device=DEV, dtype=torch.float32)
+        ext.fused_sgd_fp32(p, g, mom, 0.1, 0.9, 0.0, 0.01, False, 1.0)
+        p_ref.grad = g.clone()
+        opt.step()
+    assert (p - p_ref.detach()).abs().max() < 1e-5
+
+
+def test_fused_adam_fp32_matches_torch():
+    ext = _ext()
+    torch.manual_seed(0)
+    p = torch.randn(257, device=DEV, dtype=torch.float32)
+    p_ref = torch.nn.Parameter(p.clone())
+    opt = torch.optim.Adam([p_ref], lr=0.01)
+    m = torch.zeros_like(p)
+    v = torch.zeros_like(p)
+    for t in range(1, 5):
+        g = torch.randn(257, device=DEV, dtype=torch.float32)
+        ext.fused_adam_fp32(p, g, m, v, 0.01, 0.9, 0.999, 1e-8, 0.0, t, 1.0)
+        p_ref.grad = g.clone()
+        opt.step()
+    assert (p - p_ref.detach()).abs().max() < 1e-5
+
+
+def test_colsum_two_stage_large_B():
+    ext = _ext()
+    torch.manual_seed(1)
+    x = torch.randn(4096, 32, device=DEV).to(torch.bfloat16)
+    out = torch.empty(32, device=DEV, dtype=torch.float32)
+    ext.colsum_bf16(x, out)
+    ref = x.float().sum(dim=0)
+    assert (out - ref).abs().max() / (ref.abs().max() + 1e-3) < 0.01
+
+
 # ------------------------------------------------------------- layernorm K8
 def test_layernorm_fwd_bwd():
     from adanet_amd.ops.layernorm import HipLayerNorm

@@ -41,12 +41,23 @@ def test_hip_linear_relu_and_grads():
 
 
 def test_hip_linear_padding():
-    lin = HipLinear(8, 10, dtype=torch.float32)  # 10 -> padded 16
-    assert lin.padded_out == 16
-    assert lin.weight.shape == (16, 8)
+    lin = HipLinear(8, 10, dtype=torch.float32)  # 10 -> padded 32
+    assert lin.padded_out == 32
+    assert lin.weight.shape == (32, 8)
     assert (lin.weight[10:] == 0).all()
     y = lin(torch.randn(3, 8))
     assert y.shape == (3, 10)
+
+
+def test_weighted_sum_logits_flat_tensor():
+    a = torch.full((4, 3), 1.0)
+    b = torch.full((4, 3), 3.0)
+    flat = torch.tensor([0.5, 0.25])
+    out = weighted_sum_logits([a, b], flat)
+    assert torch.allclose(out, torch.full((4, 3), 1.25))
+    flat_v = torch.tensor([[1.0, 0.0, 1.0], [0.0, 1.0, 0.0]])
+    out_v = weighted_sum_logits([a, b], flat_v)
+    assert torch.allclose(out_v, torch.tensor([1.0, 3.0, 1.0]).expand(4, 3))
 
 
 def test_softmax_xent_matches_cross_entropy():
