@@ -171,6 +171,7 @@ class Estimator(object):
         self._placement: PlacementStrategy = placement or ReplicationStrategy()
         self._placement.config = self._config
         self._use_streams = kwargs.pop("use_streams", True)
+        self._use_hip_graphs = kwargs.pop("use_hip_graphs", True)
         if kwargs:
             raise ValueError("Unknown kwargs: %s" % sorted(kwargs))
 
@@ -622,7 +623,8 @@ class Estimator(object):
             placement=self._placement, use_streams=self._use_streams,
             replicate_ensemble_in_training=(
                 self._replicate_ensemble_in_training),
-            to_device=lambda f, l: _to_device(f, l, self._device))
+            to_device=lambda f, l: _to_device(f, l, self._device),
+            use_graphs=self._use_hip_graphs)
         self._restore_iteration_state(iteration)
         self._current_iteration = iteration
         return iteration

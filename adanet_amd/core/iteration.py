@@ -148,7 +148,8 @@ class _Iteration(object):
                  max_iteration_steps: Optional[int], adanet_loss_decay: float,
                  device: torch.device, placement, use_streams: bool = True,
                  replicate_ensemble_in_training: bool = False,
-                 to_device: Optional[Callable] = None):
+                 to_device: Optional[Callable] = None,
+                 use_graphs: bool = True):
         self.number = number
         self.head = head
         self.subnetwork_specs: List[_SubnetworkSpec] = list(subnetwork_specs)
@@ -175,9 +176,22 @@ class _Iteration(object):
         self._loss_buf = torch.full((_LOSS_FLUSH_STEPS, max(n, 1)),
                                     float("nan"), device=device,
                                     dtype=torch.float32)
+        # Device write cursor so the ring-buffer write is a pure device op
+        # (required for hipGraph capture: the row index can't be a host
+        # constant baked into the graph).
+        self._loss_counter = torch.zeros((1,), device=device,
+                                         dtype=torch.long)
         self._loss_buf_rows = 0
         self._row_recorded: List[set] = []  # names actually written per row
         self._frozen_event = None
+        # hipGraph state
+        self._use_graphs = use_graphs
+        self._graph = None
+        self._graph_sig = None
+        self._graph_ok = None
+        self._graph_warmups = 0
+        self._graph_recorded: set = set()
+        self._static_inputs = None
 
     # ------------------------------------------------------------------
     # training
@@ -209,7 +223,133 @@ class _Iteration(object):
     def train_step(self, features, labels) -> None:
         """One lockstep training step for every still-active spec
         (the reference's single session.run over all candidate train ops,
-        iteration.py:779-804 + hooks)."""
+        iteration.py:779-804 + hooks). The device work is hipGraph-captured
+        after two warmup steps when eligible (single process, shared input,
+        no LR schedule / dropout): the steady-state step then replays as
+        one graph launch, eliminating the python launch-gap overhead that
+        otherwise dominates (measured ~3.5 ms host vs ~1.1 ms GPU per step
+        on the CIFAR DNN bench before graphing)."""
+        tm = self.train_manager
+        if self._graph_eligible():
+            self._graphed_train_step(features, labels)
+        else:
+            losses_row = self._device_step(features, labels)
+            self._write_loss_row(losses_row)
+            self._finish_loss_row(set(losses_row.keys()))
+        self._post_step_bookkeeping()
+
+    def _post_step_bookkeeping(self):
+        tm = self.train_manager
+        for spec in self.subnetwork_specs:
+            if spec.subnetwork is None:
+                continue
+            if tm.should_train(spec.name):
+                if spec.optimizer is not None:
+                    sched = getattr(spec.optimizer, "_adanet_lr_sched", None)
+                    if sched is not None:
+                        sched.step()
+                spec.step += 1
+                if (self.max_iteration_steps is not None
+                        and spec.step >= self.max_iteration_steps):
+                    tm.request_stop(spec.name, "Training is over.")
+        for spec in self.ensemble_specs:
+            if spec.ensemble is None:
+                continue
+            spec.step += 1
+            if (self.max_iteration_steps is not None
+                    and spec.step >= self.max_iteration_steps):
+                tm.request_stop(spec.name, "Training is over.")
+        self.step += 1
+        if (self.max_iteration_steps is not None
+                and self.step >= self.max_iteration_steps):
+            for name in self.spec_names:
+                tm.request_stop(name, "Training is over.")
+
+    # ------------------------------------------------------------------
+    # hipGraph capture of the steady-state step
+    # ------------------------------------------------------------------
+
+    def _graph_eligible(self) -> bool:
+        if self._graph_ok is None:
+            ok = (self._use_graphs and self.device.type == "cuda"
+                  and not comm.is_initialized())
+            if ok:
+                for spec in self.subnetwork_specs:
+                    if spec.train_input_fn is not None:
+                        ok = False
+                    if spec.optimizer is not None and getattr(
+                            spec.optimizer, "_adanet_lr_sched", None):
+                        ok = False
+                    if spec.subnetwork is not None:
+                        from adanet_amd.ops.dropout import HipDropout
+                        for m in spec.subnetwork.module.modules():
+                            if isinstance(m, HipDropout) and m.p > 0:
+                                ok = False
+            self._graph_ok = ok
+        return bool(self._graph_ok)
+
+    def _copy_static_inputs(self, features, labels):
+        if self._static_inputs is None:
+            if isinstance(features, dict):
+                sf = {k: v.clone() for k, v in features.items()}
+            else:
+                sf = features.clone()
+            self._static_inputs = (sf, labels.clone())
+            return
+        sf, sl = self._static_inputs
+        if isinstance(features, dict):
+            for k, v in features.items():
+                sf[k].copy_(v, non_blocking=True)
+        else:
+            sf.copy_(features, non_blocking=True)
+        sl.copy_(labels, non_blocking=True)
+
+    def _active_signature(self):
+        return tuple(sorted(self.train_manager.stopped))
+
+    def _graphed_train_step(self, features, labels):
+        self._copy_static_inputs(features, labels)
+        sf, sl = self._static_inputs
+        sig = self._active_signature()
+        if sig != self._graph_sig:
+            self._graph = None
+            self._graph_sig = sig
+        if self._graph_warmups < 2 and self._graph is None:
+            # Warmup on a side stream (allocator pool priming) — these ARE
+            # real training steps.
+            s = torch.cuda.Stream(device=self.device)
+            s.wait_stream(torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(s):
+                row = self._device_step(sf, sl)
+                self._write_loss_row(row)
+            torch.cuda.current_stream(self.device).wait_stream(s)
+            self._graph_warmups += 1
+            self._graph_recorded = set(row.keys())
+            self._finish_loss_row(self._graph_recorded)
+            return
+        if self._graph is None:
+            g = torch.cuda.CUDAGraph()
+            try:
+                with torch.cuda.graph(g):
+                    row = self._device_step(sf, sl)
+                    self._write_loss_row(row)
+                self._graph = g
+                self._graph_recorded = set(row.keys())
+                log.info("hipGraph captured for iteration %s (%d specs)",
+                         self.number, len(self._graph_recorded))
+            except Exception as e:  # pragma: no cover - driver quirks
+                log.warning("hipGraph capture failed (%r); eager fallback", e)
+                self._use_graphs = False
+                self._graph_ok = False
+                row = self._device_step(sf, sl)
+                self._write_loss_row(row)
+                self._finish_loss_row(set(row.keys()))
+                return
+        self._graph.replay()
+        self._finish_loss_row(self._graph_recorded)
+
+    def _device_step(self, features, labels):
+        """All of one step's device work; returns name -> loss tensor."""
         tm = self.train_manager
         frozen_out = self.compute_frozen_outputs(features, training=True)
         if self._use_streams:
@@ -258,15 +398,7 @@ class _Iteration(object):
                             comm.allreduce_gradients(
                                 list(spec.subnetwork.module.parameters()))
                         spec.optimizer.step()
-                        sched = getattr(spec.optimizer, "_adanet_lr_sched",
-                                        None)
-                        if sched is not None:
-                            sched.step()
-                    spec.step += 1
                     losses_row[spec.name] = loss.detach()
-                    if (self.max_iteration_steps is not None
-                            and spec.step >= self.max_iteration_steps):
-                        tm.request_stop(spec.name, "Training is over.")
                 else:
                     with torch.no_grad():
                         spec.subnetwork.module.eval()
@@ -297,10 +429,6 @@ class _Iteration(object):
                                                     train=True)
                 if loss_t is not None:
                     losses_row[spec.name] = loss_t.detach()
-            spec.step += 1
-            if (self.max_iteration_steps is not None
-                    and spec.step >= self.max_iteration_steps):
-                tm.request_stop(spec.name, "Training is over.")
 
         if self._use_streams:
             # Device-side join: the default stream waits on every candidate
@@ -309,13 +437,7 @@ class _Iteration(object):
             for spec in self.subnetwork_specs:
                 if spec.stream is not None:
                     cur.wait_stream(spec.stream)
-
-        self._record_losses(losses_row)
-        self.step += 1
-        if (self.max_iteration_steps is not None
-                and self.step >= self.max_iteration_steps):
-            for name in self.spec_names:
-                tm.request_stop(name, "Training is over.")
+        return losses_row
 
     def _subnetwork_loss(self, spec, logits, labels, features, frozen_out):
         """Head loss, or the builder's custom loss hook when provided
@@ -415,14 +537,23 @@ class _Iteration(object):
     # loss ring buffer -> host EMAs
     # ------------------------------------------------------------------
 
-    def _record_losses(self, losses_row: Dict[str, torch.Tensor]):
-        row = self._loss_buf_rows
+    def _write_loss_row(self, losses_row: Dict[str, torch.Tensor]):
+        """Pure device ops: scatter this step's losses into the ring buffer
+        at the device write cursor (hipGraph-capturable)."""
         names = self.spec_names
+        vec = torch.full((len(names),), float("nan"), device=self.device,
+                         dtype=torch.float32)
         for i, name in enumerate(names):
             t = losses_row.get(name)
             if t is not None:
-                self._loss_buf[row, i] = t
-        self._row_recorded.append(set(losses_row.keys()))
+                vec[i] = t
+        idx = torch.remainder(self._loss_counter, _LOSS_FLUSH_STEPS)
+        self._loss_buf.index_copy_(0, idx, vec.unsqueeze(0))
+        self._loss_counter.add_(1)
+
+    def _finish_loss_row(self, recorded: set):
+        """Host bookkeeping for one written row + periodic flush."""
+        self._row_recorded.append(recorded)
         self._loss_buf_rows += 1
         if self._loss_buf_rows >= _LOSS_FLUSH_STEPS:
             self.flush_losses()
@@ -467,6 +598,7 @@ class _Iteration(object):
                 spec.summary.set_step(step)
                 spec.summary.scalar("adanet_loss", cand.adanet_loss)
         self._loss_buf.fill_(float("nan"))
+        self._loss_counter.fill_(0)
         self._loss_buf_rows = 0
         self._row_recorded = []
 
