@@ -12,11 +12,11 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
 void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
                       const at::Tensor& grad_rows, at::Tensor& dlogits,
                       double eps);
-void mixer_fwd(const at::Tensor& ptrs, const at::Tensor& weights,
+void mixer_fwd(const at::Tensor& stack, const at::Tensor& weights,
                const c10::optional<at::Tensor>& bias, at::Tensor& out,
-               int64_t B, int64_t C, int64_t ldl, int64_t vector_mode);
-void mixer_bwd_dw(const at::Tensor& ptrs, const at::Tensor& dY, at::Tensor& dw,
-                  int64_t B, int64_t C, int64_t ldl, int64_t vector_mode);
+               int64_t vector_mode);
+void mixer_bwd_dw(const at::Tensor& stack, const at::Tensor& dY,
+                  at::Tensor& dw, int64_t vector_mode);
 void mixer_bwd_dlogits(const at::Tensor& dY, const at::Tensor& weights,
                        at::Tensor& dL, int64_t j, int64_t vector_mode);
 void fused_sgd(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,

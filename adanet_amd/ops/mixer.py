@@ -23,35 +23,28 @@ class _MixerFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, stacked_w, bias, vector_mode, *logits):
         ext = _extension.require()
-        J = len(logits)
         B, C = logits[0].shape
-        ldl = logits[0].stride(0)
-        ptrs = torch.tensor([t.data_ptr() for t in logits],
-                            dtype=torch.int64).to(logits[0].device,
-                                                  non_blocking=True)
+        # Device-side stack ([J,B,C] bf16): hipGraph-capturable, unlike a
+        # host-built pointer table whose H2D copy would bake capture-time
+        # addresses into the graph.
+        stack = torch.stack([l.detach() for l in logits]).contiguous()
         out = torch.empty((B, C), device=logits[0].device,
                           dtype=torch.bfloat16)
-        ext.mixer_fwd(ptrs, stacked_w, bias, out, B, C, ldl,
-                      1 if vector_mode else 0)
+        ext.mixer_fwd(stack, stacked_w, bias, out, 1 if vector_mode else 0)
         ctx.vector_mode = vector_mode
         ctx.has_bias = bias is not None
         ctx.logit_requires = [t.requires_grad for t in logits]
-        # Save the member tensors too: ptrs holds raw device addresses, so
-        # the buffers must stay alive until backward runs.
-        ctx.save_for_backward(stacked_w, ptrs, *logits)
-        ctx.shape = (B, C, ldl)
+        ctx.save_for_backward(stacked_w, stack)
         return out
 
     @staticmethod
     def backward(ctx, dy):
         ext = _extension.require()
-        stacked_w, ptrs = ctx.saved_tensors[:2]
-        B, C, ldl = ctx.shape
+        stacked_w, stack = ctx.saved_tensors
+        J, B, C = stack.shape
         dy = dy.contiguous()
-        J = ptrs.numel()
         dw = torch.zeros_like(stacked_w)
-        ext.mixer_bwd_dw(ptrs, dy, dw, B, C, ldl,
-                         1 if ctx.vector_mode else 0)
+        ext.mixer_bwd_dw(stack, dy, dw, 1 if ctx.vector_mode else 0)
         dbias = None
         if ctx.has_bias:
             dbias = torch.empty((C,), device=dy.device, dtype=torch.float32)
@@ -91,12 +84,8 @@ def weighted_sum_logits(logits: Sequence[torch.Tensor],
             stacked = stacked.reshape(len(weights))
     if logits[0].is_cuda:
         stacked = stacked.contiguous()
-        logits = [l if (l.stride(-1) == 1 and l.dtype == torch.bfloat16
-                        and l.stride(0) == logits[0].stride(0))
-                  else l.to(torch.bfloat16).contiguous() for l in logits]
-        # All members must share a row stride for the pointer-table kernel.
-        if len({l.stride(0) for l in logits}) != 1:
-            logits = [l.contiguous() for l in logits]
+        logits = [l if l.dtype == torch.bfloat16 else l.to(torch.bfloat16)
+                  for l in logits]
         return _MixerFn.apply(stacked, bias, vector_mode, *logits)
     # CPU reference path (fp32).
     total = None
