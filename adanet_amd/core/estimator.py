@@ -188,6 +188,8 @@ class Estimator(object):
         self._best_ensemble_state: Optional[dict] = None
         self._replay_indices: List[int] = []
         self._current_iteration: Optional[_Iteration] = None
+        self._phase_secs = {"build": 0.0, "train": 0.0, "bookkeeping": 0.0,
+                            "checkpoint": 0.0}
         self._restore_checkpoint()
 
     # ------------------------------------------------------------------
@@ -240,9 +242,12 @@ class Estimator(object):
                 break
             t = self._iteration_number
             log.info("Beginning training AdaNet iteration %s", t)
+            t_build0 = time.perf_counter()
             iteration = self._get_or_build_iteration(input_fn)
+            self._phase_secs["build"] += time.perf_counter() - t_build0
             if input_iter is None:
                 input_iter = iter(input_fn())
+            t_train0 = time.perf_counter()
             iteration_ended = True
             while not iteration.is_over():
                 if self._global_step >= budget_end:
@@ -267,6 +272,7 @@ class Estimator(object):
                     log.info("global_step = %s (iteration %s, step %s)",
                              self._global_step, t, iteration.step)
             iteration.flush_losses()
+            self._phase_secs["train"] += time.perf_counter() - t_train0
             if not iteration.is_over():
                 # Budget or input ran out mid-iteration: checkpoint so a
                 # restart resumes this iteration in place (reference
@@ -276,10 +282,16 @@ class Estimator(object):
                          t)
                 break
             log.info("Finished training Adanet iteration %s", t)
+            t_book0 = time.perf_counter()
             self._execute_bookkeeping_phase(input_fn)
+            self._phase_secs["bookkeeping"] += time.perf_counter() - t_book0
             self._current_iteration = None
             self._iteration_number += 1
+            t_ckpt0 = time.perf_counter()
             self._save_checkpoint(mid_iteration=False)
+            self._phase_secs["checkpoint"] += time.perf_counter() - t_ckpt0
+            log.info("Phase seconds so far: %s",
+                     {k: round(v, 3) for k, v in self._phase_secs.items()})
             if input_iter is None:
                 break
         return self
