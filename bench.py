@@ -49,6 +49,9 @@ def parse_args():
 
 def main():
     args = parse_args()
+    if os.environ.get("ADANET_LOG"):
+        import logging
+        logging.basicConfig(level=os.environ["ADANET_LOG"])
     sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
     import functools
 
