@@ -436,9 +436,17 @@ class Estimator(object):
         0..t-1 from their architecture JSON + stored weights (the analog of
         reference _architecture_ensemble_spec, estimator.py:1785-1882;
         builders re-invoked in EVAL mode so dropout is off,
-        iteration.py:569-572)."""
+        iteration.py:569-572).
+
+        Fast path: when the previous iteration just finished IN THIS
+        process, its winning ensemble's modules are already live in HBM —
+        reuse them instead of re-instantiating the whole frozen chain
+        (measured ~0.2 s/iteration of rebuild on the CIFAR DNN bench)."""
         if t == 0:
             return None, {}
+        live = getattr(self, "_live_prev", None)
+        if live is not None and live[0] == t:
+            return live[1], live[2]
         module_cache: Dict[str, Subnetwork] = {}
         prev_ensemble = None
         for i in range(t):
@@ -552,11 +560,14 @@ class Estimator(object):
             summary = self._make_summary("subnetwork", b.name, t,
                                          self._enable_subnetwork_summaries)
             if build_here:
-                sub = b.build_subnetwork(
-                    features,
-                    logits_dimension=self._head.logits_dimension,
-                    training=True,
-                    previous_ensemble=prev_ensemble)
+                with torch.device(self._device):
+                    # Factory calls inside builders allocate directly on the
+                    # target device (no CPU init + transfer per iteration).
+                    sub = b.build_subnetwork(
+                        features,
+                        logits_dimension=self._head.logits_dimension,
+                        training=True,
+                        previous_ensemble=prev_ensemble)
                 sub.name = b.name
                 sub.module.to(self._device)
                 if self._device.type == "cuda":
@@ -774,6 +785,7 @@ class Estimator(object):
         arch.set_replay_indices(self._replay_indices)
         serialized = arch.serialize(t, self._global_step)
         self._architectures[t] = serialized
+        self._stash_live_prev(iteration, chosen, t)
         if comm.is_chief():
             with open(
                     os.path.join(self._model_dir,
@@ -803,6 +815,42 @@ class Estimator(object):
         if comm.is_chief():
             self._report_accessor.write_iteration_report(
                 iteration.number, list(seen.values()))
+
+    def _stash_live_prev(self, iteration, chosen: _EnsembleSpec, t: int):
+        """Keep the winner's live on-device modules as iteration t+1's
+        previous ensemble (skips the O(t) frozen-chain rebuild + HBM
+        reload; the checkpoint stays the source of truth for restarts)."""
+        self._live_prev = None
+        if chosen.ensemble is None:
+            return
+        frozen = {}
+        renamed = {}
+        for kind, name in chosen.members:
+            if kind == "frozen":
+                sub = iteration.frozen_subnetworks.get(name)
+                if sub is None:
+                    return
+                frozen[name] = sub
+            else:
+                try:
+                    spec = iteration._subnetwork_spec(name)
+                except KeyError:
+                    return
+                sub = spec.subnetwork
+                if sub is None:
+                    return  # round-robin non-owner: rebuild from states
+                new_name = "t{}_{}".format(t, name)
+                renamed[sub.name] = new_name
+                sub.name = new_name
+                for p in sub.module.parameters():
+                    p.requires_grad_(False)
+                sub.module.eval()
+                frozen[new_name] = sub
+        ens = chosen.ensemble
+        for ws in getattr(ens, "weighted_subnetworks", []):
+            if ws.builder_name in renamed:
+                ws.builder_name = renamed[ws.builder_name]
+        self._live_prev = (t + 1, ens, frozen)
 
     def _freeze_winner(self, iteration, chosen: _EnsembleSpec, t: int):
         """Persist the winning ensemble's member weights + mixture weights;
@@ -968,6 +1016,7 @@ class Estimator(object):
         if checkpoint_path:
             payload = torch.load(checkpoint_path, map_location="cpu",
                                  weights_only=False)
+            self._live_prev = None  # explicit load invalidates live modules
             self._iteration_number = payload[self._Keys.CURRENT_ITERATION]
             self._global_step = payload["global_step"]
             self._architectures = {

@@ -156,14 +156,12 @@ class HipLinear(nn.Module):
 
     def reset_parameters(self):
         with torch.no_grad():
-            w = torch.empty((self.padded_out, self.in_features),
-                            dtype=torch.float32)
-            # Kaiming-uniform as torch.nn.Linear default.
+            # Kaiming-uniform (torch.nn.Linear default), initialized
+            # directly on the parameter's device/dtype (no CPU round-trip).
             bound = 1.0 / math.sqrt(self.in_features)
-            w.uniform_(-bound, bound)
+            self.weight.uniform_(-bound, bound)
             if self.padded_out != self.out_features:
-                w[self.out_features:].zero_()
-            self.weight.copy_(w.to(self.weight.dtype))
+                self.weight[self.out_features:].zero_()
             if self.bias is not None:
                 self.bias.zero_()
 

@@ -122,6 +122,21 @@ def test_evaluate_predict_export(model_dir, synthetic_classification):
     assert os.path.exists(os.path.join(export_dir, "architecture.json"))
 
 
+def test_live_prev_matches_checkpoint_rebuild(model_dir,
+                                              synthetic_classification):
+    """The in-process live-module fast path and a fresh restore from the
+    checkpoint must describe the same frozen best ensemble."""
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn)
+    est.train(input_fn, max_steps=20)
+    live_res = est.evaluate(input_fn, steps=4)
+    est2 = _make_estimator(model_dir, input_fn)  # rebuilds from checkpoint
+    rebuilt_res = est2.evaluate(input_fn, steps=4)
+    assert live_res["loss"] == pytest.approx(rebuilt_res["loss"], abs=1e-5)
+    assert (live_res["architecture/adanet/ensembles"] ==
+            rebuilt_res["architecture/adanet/ensembles"])
+
+
 def test_evaluate_before_training_raises(model_dir, synthetic_classification):
     X, Y, input_fn = synthetic_classification
     est = _make_estimator(model_dir, input_fn)
