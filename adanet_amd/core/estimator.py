@@ -228,7 +228,11 @@ class Estimator(object):
         if self._global_step >= budget_end:
             return self
 
-        comm.maybe_init_process_group()
+        # The reference's workers wait up to worker_wait_timeout_secs for the
+        # chief (estimator.py:951-996, checkpoint polling); here the analog
+        # is the process-group timeout on the collective control plane.
+        comm.maybe_init_process_group(
+            timeout_secs=int(self._worker_wait_timeout_secs))
         if self._config.random_seed is not None:
             torch.manual_seed(self._config.random_seed)
 
@@ -770,6 +774,14 @@ class Estimator(object):
         best_index = comm.broadcast_object(best_index, src=0) if (
             comm.is_initialized()) else best_index
         chosen = iteration.ensemble_specs[best_index]
+        # Per-candidate eval summaries (the analog of _EvalMetricSaverHook's
+        # per-candidate eval dirs, reference estimator.py:150-233).
+        for i, spec in enumerate(iteration.ensemble_specs):
+            if spec.summary is not None and i < len(losses) and not (
+                    math.isnan(losses[i]) or math.isinf(losses[i])):
+                spec.summary.set_step(self._global_step)
+                spec.summary.scalar("adanet_loss", abs(losses[i]),
+                                    family="eval")
         log.info("Iteration %s: best ensemble is %r (index %d)", t,
                  chosen.name, best_index)
         self._replay_indices.append(best_index)
