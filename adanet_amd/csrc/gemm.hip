@@ -23,39 +23,43 @@
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
-#define BK 32
 #define THREADS 256
 
 typedef s16x8 frag_ab;
 
-// Stage a [ROWS x BK] bf16 tile into LDS: each wave issues ROWS/16/(4 waves)
-// (or strided) 1 KiB global_load_lds ops (64 lanes x 16 B, LDS-linear).
-template <int ROWS>
+// Stage a [ROWS x KSTEP] bf16 tile into LDS: each global_load_lds covers
+// 1 KiB (64 lanes x 16 B, LDS-linear = 512/KSTEP rows); waves stride the
+// segment list.
+template <int ROWS, int KSTEP>
 __device__ __forceinline__ void stage_tile_nt(
     const bf16_t* __restrict__ G, int ld, int tile_row0, int max_row, int k0,
     bf16_t* __restrict__ lds, int wid, int lane) {
-  constexpr int SEGMENTS = ROWS / 16;
+  constexpr int ROWS_PER_SEG = 512 / KSTEP;      // rows per 1 KiB op
+  constexpr int LANES_PER_ROW = KSTEP / 8;       // 16 B loads per row
+  constexpr int SEGMENTS = ROWS / ROWS_PER_SEG;
 #pragma unroll
   for (int seg = wid; seg < SEGMENTS; seg += 4) {
-    const int row_in_tile = seg * 16 + (lane >> 2);
+    const int row_in_tile = seg * ROWS_PER_SEG + lane / LANES_PER_ROW;
     int grow = tile_row0 + row_in_tile;
     grow = grow < max_row ? grow : max_row - 1;  // clamp; masked on C-store
-    const bf16_t* gp = G + (int64_t)grow * ld + k0 + (lane & 3) * 8;
-    bf16_t* lp = lds + seg * 16 * BK;  // wave-uniform base
+    const bf16_t* gp =
+        G + (int64_t)grow * ld + k0 + (lane % LANES_PER_ROW) * 8;
+    bf16_t* lp = lds + seg * ROWS_PER_SEG * KSTEP;  // wave-uniform base
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)gp,
         (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
   }
 }
 
-// BM x BN tile, 2x2 wave grid, FM x FN 16x16 MFMA fragments per wave.
-template <int BM, int BN, int FM, int FN, int MINWAVES>
+// BM x BN tile, 2x2 wave grid, FM x FN 16x16 MFMA fragments per wave,
+// KSTEP K-columns staged per barrier (KSTEP/32 MFMA K-slices).
+template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32>
 __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles) {
-  __shared__ bf16_t As[2][BM * BK];
-  __shared__ bf16_t Bs[2][BN * BK];
+  __shared__ bf16_t As[2][BM * KSTEP];
+  __shared__ bf16_t Bs[2][BN * KSTEP];
 
   // Bijective XCD-aware swizzle (guide m204): contiguous tile chunks/XCD.
   const int nwg = mtiles * ntiles;
@@ -79,35 +83,38 @@ __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int ktiles = K / BK;
-  stage_tile_nt<BM>(A, lda, row0, M, 0, As[0], wid, lane);
-  stage_tile_nt<BN>(B, ldb, col0, N, 0, Bs[0], wid, lane);
+  const int ktiles = K / KSTEP;
+  stage_tile_nt<BM, KSTEP>(A, lda, row0, M, 0, As[0], wid, lane);
+  stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, 0, Bs[0], wid, lane);
 
   int buf = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
     __syncthreads();  // staged tile `buf` visible; prior reads of buf^1 done
     if (kt + 1 < ktiles) {
-      const int k0 = (kt + 1) * BK;
-      stage_tile_nt<BM>(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
-      stage_tile_nt<BN>(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
+      const int k0 = (kt + 1) * KSTEP;
+      stage_tile_nt<BM, KSTEP>(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
+      stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
     }
-    // Fragment loads: 8 contiguous bf16 per lane -> ds_read_b128.
-    frag_ab a[FM], b[FN];
-    const int kofs = (lane >> 4) * 8;
-    const int arow = wm * (FM * 16) + (lane & 15);
-    const int brow = wn * (FN * 16) + (lane & 15);
 #pragma unroll
-    for (int f = 0; f < FM; ++f)
-      a[f] = *(const frag_ab*)&As[buf][(arow + f * 16) * BK + kofs];
+    for (int kk = 0; kk < KSTEP / 32; ++kk) {
+      // Fragment loads: 8 contiguous bf16 per lane -> ds_read_b128.
+      frag_ab a[FM], b[FN];
+      const int kofs = kk * 32 + (lane >> 4) * 8;
+      const int arow = wm * (FM * 16) + (lane & 15);
+      const int brow = wn * (FN * 16) + (lane & 15);
 #pragma unroll
-    for (int f = 0; f < FN; ++f)
-      b[f] = *(const frag_ab*)&Bs[buf][(brow + f * 16) * BK + kofs];
+      for (int f = 0; f < FM; ++f)
+        a[f] = *(const frag_ab*)&As[buf][(arow + f * 16) * KSTEP + kofs];
 #pragma unroll
-    for (int i = 0; i < FM; ++i)
+      for (int f = 0; f < FN; ++f)
+        b[f] = *(const frag_ab*)&Bs[buf][(brow + f * 16) * KSTEP + kofs];
 #pragma unroll
-      for (int j = 0; j < FN; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
-                                                            acc[i][j], 0, 0, 0);
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
     buf ^= 1;
   }
 
@@ -178,7 +185,7 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   const bf16_t* b = (const bf16_t*)B.data_ptr();
   bf16_t* c = (bf16_t*)C.data_ptr();
 
-  const bool fast = (K % BK == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
+  const bool fast = (K % 32 == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
   if (fast) {
     const int mt128 = (M + 127) / 128, nt128 = (N + 127) / 128;
     // 64-tile when the 128-tile grid underfills the chip (256 CUs want
@@ -203,5 +210,45 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
                        ldc, (int)act);
   }
+  HIP_CHECK_KERNEL();
+}
+
+// Tuning probe: dispatch a specific tile/occupancy/K-step variant so a
+// single GPU session can A/B the whole configuration grid
+// (benchmarks/gemm_variants.py).
+void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
+                        at::Tensor& C, int64_t variant) {
+  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
+  const int lda = (int)A.stride(0), ldb = (int)B.stride(0),
+            ldc = (int)C.stride(0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+
+#define LAUNCH_V(BM, BN, FM, FN, MW, KS)                                      \
+  do {                                                                        \
+    TORCH_CHECK(K % KS == 0, "probe: K %% KSTEP");                            \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, KS>),         \
+                       dim3(mt * nt), dim3(THREADS), 0, stream.stream(), a,  \
+                       b, c, nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);     \
+  } while (0)
+
+  switch (variant) {
+    case 0: LAUNCH_V(128, 128, 4, 4, 2, 32); break;
+    case 1: LAUNCH_V(128, 128, 4, 4, 3, 32); break;
+    case 2: LAUNCH_V(128, 128, 4, 4, 4, 32); break;
+    case 3: LAUNCH_V(64, 64, 2, 2, 4, 32); break;
+    case 4: LAUNCH_V(64, 64, 2, 2, 8, 32); break;
+    case 5: LAUNCH_V(128, 64, 4, 2, 4, 32); break;
+    case 6: LAUNCH_V(128, 64, 4, 2, 2, 32); break;
+    case 7: LAUNCH_V(128, 128, 4, 4, 2, 64); break;
+    case 8: LAUNCH_V(128, 64, 4, 2, 4, 64); break;
+    case 9: LAUNCH_V(64, 64, 2, 2, 6, 32); break;
+    case 10: LAUNCH_V(64, 128, 2, 4, 4, 32); break;
+    default: TORCH_CHECK(false, "unknown variant");
+  }
+#undef LAUNCH_V
   HIP_CHECK_KERNEL();
 }
