@@ -187,22 +187,30 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
 
   const bool fast = (K % 32 == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
   if (fast) {
+    // Tile dispatch tuned on MI355X (benchmarks/gemm_variants.py,
+    // profiles/gemm_variants_r01.json): occupancy rules until ~4 blocks/CU,
+    // then per-wave efficiency of the big tile wins; for asymmetric mid
+    // shapes put the 128 side on the SHORTER output dim.
     const int mt128 = (M + 127) / 128, nt128 = (N + 127) / 128;
-    // 64-tile when the 128-tile grid underfills the chip (256 CUs want
-    // >=2 blocks each before the big tile's per-wave efficiency wins).
-    const bool small = (int64_t)mt128 * nt128 < 384;
-    if (small) {
-      const int mt = (M + 63) / 64, nt = (N + 63) / 64;
-      hipLaunchKernelGGL((gemm_nt_bf16_kernel<64, 64, 2, 2, 4>),
-                         dim3(mt * nt), dim3(THREADS), 0, stream.stream(), a,
-                         b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt,
-                         nt);
+    const int64_t b128 = (int64_t)mt128 * nt128;
+#define LAUNCH_CFG(BM, BN, FM, FN, MW)                                        \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32>),         \
+                       dim3(mt * nt), dim3(THREADS), 0, stream.stream(), a,  \
+                       b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, \
+                       nt);                                                   \
+  } while (0)
+    if (b128 >= 1024) {
+      LAUNCH_CFG(128, 128, 4, 4, 4);
+    } else if (b128 >= 256 && M > N) {
+      LAUNCH_CFG(64, 128, 2, 4, 4);
+    } else if (b128 >= 256 && N > M) {
+      LAUNCH_CFG(128, 64, 4, 2, 2);
     } else {
-      hipLaunchKernelGGL((gemm_nt_bf16_kernel<128, 128, 4, 4, 2>),
-                         dim3(mt128 * nt128), dim3(THREADS), 0,
-                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,
-                         ldb, ldc, (int)act, mt128, nt128);
+      LAUNCH_CFG(64, 64, 2, 2, 6);
     }
+#undef LAUNCH_CFG
   } else {
     const int64_t total = (int64_t)M * N;
     const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
