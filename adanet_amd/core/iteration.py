@@ -183,6 +183,7 @@ class _Iteration(object):
                                          dtype=torch.long)
         self._loss_buf_rows = 0
         self._row_recorded: List[set] = []  # names actually written per row
+        self._nan_scalar = None
         self._frozen_event = None
         # hipGraph state
         self._use_graphs = use_graphs
@@ -539,14 +540,17 @@ class _Iteration(object):
 
     def _write_loss_row(self, losses_row: Dict[str, torch.Tensor]):
         """Pure device ops: scatter this step's losses into the ring buffer
-        at the device write cursor (hipGraph-capturable)."""
+        at the device write cursor (hipGraph-capturable). One stack kernel
+        builds the row (not a per-spec D2D copy each)."""
         names = self.spec_names
-        vec = torch.full((len(names),), float("nan"), device=self.device,
-                         dtype=torch.float32)
-        for i, name in enumerate(names):
-            t = losses_row.get(name)
-            if t is not None:
-                vec[i] = t
+        if self._nan_scalar is None:
+            self._nan_scalar = torch.full((), float("nan"),
+                                          device=self.device,
+                                          dtype=torch.float32)
+        vec = torch.stack([
+            losses_row.get(name, self._nan_scalar).float().reshape(())
+            for name in names
+        ])
         idx = torch.remainder(self._loss_counter, _LOSS_FLUSH_STEPS)
         self._loss_buf.index_copy_(0, idx, vec.unsqueeze(0))
         self._loss_counter.add_(1)

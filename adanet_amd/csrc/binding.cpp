@@ -10,10 +10,13 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
                         at::Tensor& C, int64_t variant);
 void transpose_bf16(const at::Tensor& in, at::Tensor& out);
 void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
-                      at::Tensor& loss, at::Tensor& probs, double eps);
+                      const c10::optional<at::Tensor>& loss,
+                      at::Tensor& probs, double eps,
+                      const c10::optional<at::Tensor>& mean_out);
 void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
-                      const at::Tensor& grad_rows, at::Tensor& dlogits,
-                      double eps);
+                      const c10::optional<at::Tensor>& grad_rows,
+                      at::Tensor& dlogits, double eps,
+                      const c10::optional<at::Tensor>& grad_scalar);
 void mixer_fwd(const at::Tensor& stack, const at::Tensor& weights,
                const c10::optional<at::Tensor>& bias, at::Tensor& out,
                int64_t vector_mode);

@@ -18,7 +18,7 @@
 __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
     const bf16_t* __restrict__ logits, const int64_t* __restrict__ labels,
     float* __restrict__ loss, bf16_t* __restrict__ probs, int B, int C,
-    int ldl, int ldp, float eps) {
+    int ldl, int ldp, float eps, float* __restrict__ mean_out) {
   const int wave_in_block = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int waves = (gridDim.x * blockDim.x) >> 6;
@@ -45,22 +45,29 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
     }
     sum_logits = wave_reduce_sum(sum_logits);
     ly = wave_reduce_sum(ly);  // only the label lane contributed
-    if (lane == 0)
-      loss[row] = lse - (1.f - eps) * ly - (eps / C) * sum_logits;
+    if (lane == 0) {
+      const float l = lse - (1.f - eps) * ly - (eps / C) * sum_logits;
+      if (loss) loss[row] = l;
+      // fused mean: one atomic per row (mean_out pre-zeroed).
+      if (mean_out) atomicAdd(mean_out, l / B);
+    }
   }
 }
 
 __global__ __launch_bounds__(256) void softmax_xent_bwd_kernel(
     const bf16_t* __restrict__ probs, const int64_t* __restrict__ labels,
     const float* __restrict__ grad_rows, bf16_t* __restrict__ dlogits, int B,
-    int C, int ldp, int ldd, float eps) {
+    int C, int ldp, int ldd, float eps,
+    const float* __restrict__ grad_scalar) {
   const int wave_in_block = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int waves = (gridDim.x * blockDim.x) >> 6;
+  // grad_scalar: upstream grad of the MEAN loss (device pointer, read once
+  // per row) -> per-row grad = *grad_scalar / B. Else per-row grad_rows.
   for (int row = blockIdx.x * (blockDim.x >> 6) + wave_in_block; row < B;
        row += waves) {
     const int64_t y = labels[row];
-    const float g = grad_rows[row];
+    const float g = grad_scalar ? (*grad_scalar) / B : grad_rows[row];
     for (int c = lane; c < C; c += 64) {
       float t = (c == (int)y ? 1.f - eps : 0.f) + eps / C;
       float p = bf2f(probs[(int64_t)row * ldp + c]);
@@ -70,7 +77,9 @@ __global__ __launch_bounds__(256) void softmax_xent_bwd_kernel(
 }
 
 void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
-                      at::Tensor& loss, at::Tensor& probs, double eps) {
+                      const c10::optional<at::Tensor>& loss,
+                      at::Tensor& probs, double eps,
+                      const c10::optional<at::Tensor>& mean_out) {
   TORCH_CHECK(logits.is_cuda() && logits.scalar_type() == at::kBFloat16,
               "xent: bf16 GPU logits required");
   TORCH_CHECK(labels.scalar_type() == at::kLong, "xent: int64 labels");
@@ -78,24 +87,38 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   const int B = (int)logits.size(0), C = (int)logits.size(1);
   auto stream = at::cuda::getCurrentCUDAStream();
   const int blocks = std::min((B + 3) / 4, 2048);
+  float* loss_ptr = (loss.has_value() && loss->defined())
+                        ? loss->data_ptr<float>()
+                        : nullptr;
+  float* mean_ptr = (mean_out.has_value() && mean_out->defined())
+                        ? mean_out->data_ptr<float>()
+                        : nullptr;
   hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const bf16_t*)logits.data_ptr(),
-                     labels.data_ptr<int64_t>(), loss.data_ptr<float>(),
+                     labels.data_ptr<int64_t>(), loss_ptr,
                      (bf16_t*)probs.data_ptr(), B, C, (int)logits.stride(0),
-                     (int)probs.stride(0), (float)eps);
+                     (int)probs.stride(0), (float)eps, mean_ptr);
   HIP_CHECK_KERNEL();
 }
 
 void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
-                      const at::Tensor& grad_rows, at::Tensor& dlogits,
-                      double eps) {
+                      const c10::optional<at::Tensor>& grad_rows,
+                      at::Tensor& dlogits, double eps,
+                      const c10::optional<at::Tensor>& grad_scalar) {
   const int B = (int)probs.size(0), C = (int)probs.size(1);
   auto stream = at::cuda::getCurrentCUDAStream();
   const int blocks = std::min((B + 3) / 4, 2048);
+  const float* gr = (grad_rows.has_value() && grad_rows->defined())
+                        ? grad_rows->data_ptr<float>()
+                        : nullptr;
+  const float* gs = (grad_scalar.has_value() && grad_scalar->defined())
+                        ? grad_scalar->data_ptr<float>()
+                        : nullptr;
+  TORCH_CHECK(gr || gs, "xent bwd: need grad_rows or grad_scalar");
   hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const bf16_t*)probs.data_ptr(),
-                     labels.data_ptr<int64_t>(), grad_rows.data_ptr<float>(),
+                     labels.data_ptr<int64_t>(), gr,
                      (bf16_t*)dlogits.data_ptr(), B, C, (int)probs.stride(0),
-                     (int)dlogits.stride(0), (float)eps);
+                     (int)dlogits.stride(0), (float)eps, gs);
   HIP_CHECK_KERNEL();
 }

@@ -18,6 +18,7 @@ from adanet_amd.ops import _extension
 
 
 class _SoftmaxXentFn(torch.autograd.Function):
+    """Per-row loss vector variant (reduction='none'/'sum')."""
 
     @staticmethod
     def forward(ctx, logits, labels, label_smoothing):
@@ -27,7 +28,7 @@ class _SoftmaxXentFn(torch.autograd.Function):
         probs = torch.empty((B, C), device=logits.device,
                             dtype=torch.bfloat16)
         ext.softmax_xent_fwd(logits, labels, loss, probs,
-                             float(label_smoothing))
+                             float(label_smoothing), None)
         ctx.save_for_backward(probs, labels)
         ctx.label_smoothing = float(label_smoothing)
         return loss
@@ -38,7 +39,36 @@ class _SoftmaxXentFn(torch.autograd.Function):
         ext = _extension.require()
         dlogits = torch.empty_like(probs)
         ext.softmax_xent_bwd(probs, labels, grad_rows.contiguous().float(),
-                             dlogits, ctx.label_smoothing)
+                             dlogits, ctx.label_smoothing, None)
+        return dlogits, None, None
+
+
+class _SoftmaxXentMeanFn(torch.autograd.Function):
+    """Fused-mean variant: forward returns the scalar mean loss directly
+    (one kernel, no torch reduce), backward reads the upstream scalar grad
+    from device memory (hipGraph-safe, no host scalars)."""
+
+    @staticmethod
+    def forward(ctx, logits, labels, label_smoothing):
+        ext = _extension.require()
+        B, C = logits.shape
+        mean = torch.zeros((), device=logits.device, dtype=torch.float32)
+        probs = torch.empty((B, C), device=logits.device,
+                            dtype=torch.bfloat16)
+        ext.softmax_xent_fwd(logits, labels, None, probs,
+                             float(label_smoothing), mean)
+        ctx.save_for_backward(probs, labels)
+        ctx.label_smoothing = float(label_smoothing)
+        return mean
+
+    @staticmethod
+    def backward(ctx, grad):
+        probs, labels = ctx.saved_tensors
+        ext = _extension.require()
+        dlogits = torch.empty_like(probs)
+        ext.softmax_xent_bwd(probs, labels, None, dlogits,
+                             ctx.label_smoothing,
+                             grad.contiguous().float())
         return dlogits, None, None
 
 
@@ -49,13 +79,15 @@ def softmax_xent(logits: torch.Tensor, labels: torch.Tensor,
     if logits.is_cuda:
         if logits.stride(1) != 1:
             logits = logits.contiguous()
+        if reduction == "mean":
+            return _SoftmaxXentMeanFn.apply(logits, labels, label_smoothing)
         per_row = _SoftmaxXentFn.apply(logits, labels, label_smoothing)
     else:
         per_row = F.cross_entropy(logits.float(), labels,
                                   label_smoothing=label_smoothing,
                                   reduction="none")
-    if reduction == "mean":
-        return per_row.mean()
+        if reduction == "mean":
+            return per_row.mean()
     if reduction == "sum":
         return per_row.sum()
     return per_row

@@ -102,7 +102,7 @@ def test_softmax_xent_fwd_bwd(B, C, eps):
     labels = torch.randint(0, C, (B,), device=DEV)
     loss = torch.empty(B, device=DEV, dtype=torch.float32)
     probs = torch.empty(B, C, device=DEV, dtype=torch.bfloat16)
-    ext.softmax_xent_fwd(logits, labels, loss, probs, eps)
+    ext.softmax_xent_fwd(logits, labels, loss, probs, eps, None)
     ref = torch.nn.functional.cross_entropy(logits.float(), labels,
                                             label_smoothing=eps,
                                             reduction="none")
@@ -112,11 +112,27 @@ def test_softmax_xent_fwd_bwd(B, C, eps):
 
     grad_rows = torch.full((B,), 1.0 / B, device=DEV, dtype=torch.float32)
     dlogits = torch.empty(B, C, device=DEV, dtype=torch.bfloat16)
-    ext.softmax_xent_bwd(probs, labels, grad_rows, dlogits, eps)
+    ext.softmax_xent_bwd(probs, labels, grad_rows, dlogits, eps, None)
     lf = logits.float().requires_grad_(True)
     torch.nn.functional.cross_entropy(lf, labels, label_smoothing=eps,
                                       reduction="mean").backward()
     assert (dlogits.float() - lf.grad).abs().max() < 0.01
+
+
+def test_softmax_xent_fused_mean_autograd():
+    """The scalar-mean fused path vs torch CE (fwd + bwd)."""
+    from adanet_amd.ops.xent import softmax_xent
+    torch.manual_seed(2)
+    logits = torch.randn(128, 10, device=DEV).to(
+        torch.bfloat16).requires_grad_(True)
+    labels = torch.randint(0, 10, (128,), device=DEV)
+    loss = softmax_xent(logits, labels, reduction="mean")
+    lf = logits.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(lf, labels)
+    assert abs(float(loss) - float(ref)) < 0.02
+    loss.backward()
+    ref.backward()
+    assert (logits.grad.float() - lf.grad).abs().max() < 0.01
 
 
 def test_softmax_xent_strided_view():
@@ -128,7 +144,7 @@ def test_softmax_xent_strided_view():
     labels = torch.randint(0, 10, (64,), device=DEV)
     loss = torch.empty(64, device=DEV, dtype=torch.float32)
     probs = torch.empty(64, 16, device=DEV, dtype=torch.bfloat16)[:, :10]
-    ext.softmax_xent_fwd(logits, labels, loss, probs, 0.0)
+    ext.softmax_xent_fwd(logits, labels, loss, probs, 0.0, None)
     ref = torch.nn.functional.cross_entropy(logits.float(), labels,
                                             reduction="none")
     assert (loss - ref).abs().max() < 0.02
