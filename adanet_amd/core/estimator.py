@@ -69,7 +69,11 @@ def _to_device(features, labels, device, dtype=torch.bfloat16):
     else:
         features = conv(features)
     if labels is not None:
-        labels = labels.to(device, non_blocking=True)
+        if isinstance(labels, dict):  # multi-head labels
+            labels = {k: v.to(device, non_blocking=True)
+                      for k, v in labels.items()}
+        else:
+            labels = labels.to(device, non_blocking=True)
     return features, labels
 
 

@@ -13,6 +13,8 @@
 #define TDIM 64
 #define TPAD 8  // pad leading dim: 64+8 halfwords -> read stride 144 B, bank-spread
 
+typedef __attribute__((ext_vector_type(4))) unsigned short us16x4;
+
 __global__ __launch_bounds__(256) void transpose_bf16_kernel(
     const bf16_t* __restrict__ in, bf16_t* __restrict__ out, int M, int N,
     int ldi, int ldo, int mtiles, int ntiles) {
@@ -22,14 +24,25 @@ __global__ __launch_bounds__(256) void transpose_bf16_kernel(
   const int t = threadIdx.x;
   const int r4 = t >> 4;        // 0..15
   const int c4 = (t & 15) * 4;  // column group of 4
+  // Interior tiles take the fully-vectorized path (ushort4 both sides).
+  const bool interior = (tm * TDIM + TDIM <= M) && (tn * TDIM + TDIM <= N) &&
+                        (ldi % 4 == 0) && (ldo % 4 == 0);
 
-  // Load: rows of `in` coalesced (ushort4 = 4 bf16 per thread, 4 phases).
+  // Load: rows of `in` coalesced.
 #pragma unroll
   for (int p = 0; p < 4; ++p) {
     const int row = p * 16 + r4;
     const int gr = tm * TDIM + row;
     const int gc = tn * TDIM + c4;
-    if (gr < M) {
+    if (interior) {
+      const us16x4 v =
+          *(const us16x4*)&in[(int64_t)gr * ldi + gc];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        const unsigned short s = v[u];
+        tile[row][c4 + u] = *(const bf16_t*)&s;
+      }
+    } else if (gr < M) {
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int c = gc + u;
@@ -38,13 +51,19 @@ __global__ __launch_bounds__(256) void transpose_bf16_kernel(
     }
   }
   __syncthreads();
-  // Store: rows of `out` (columns of `in`) coalesced.
+  // Store: rows of `out` (columns of `in`) coalesced, packed ushort4.
 #pragma unroll
   for (int p = 0; p < 4; ++p) {
     const int row = p * 16 + r4;       // row of out == col of in
     const int gr = tn * TDIM + row;
     const int gc = tm * TDIM + c4;
-    if (gr < N) {
+    if (interior) {
+      us16x4 v;
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        v[u] = *(const unsigned short*)&tile[c4 + u][row];
+      *(us16x4*)&out[(int64_t)gr * ldo + gc] = v;
+    } else if (gr < N) {
 #pragma unroll
       for (int u = 0; u < 4; ++u) {
         const int c = gc + u;

@@ -117,6 +117,57 @@ class BinaryClassHead(Head):
                     "average_loss": float(self.loss(logits, labels).cpu())}
 
 
+class MultiHead(Head):
+    """Composite head over named sub-heads (the reference's multi-head
+    support, adanet/core/estimator_test.py:1517: logits split across heads,
+    labels a dict keyed by head name, losses summed)."""
+
+    def __init__(self, heads: dict):
+        if not heads:
+            raise ValueError("heads must not be empty")
+        self._heads = dict(heads)
+        self._order = sorted(heads.keys())
+
+    @property
+    def logits_dimension(self) -> int:
+        return sum(self._heads[k].logits_dimension for k in self._order)
+
+    def _split(self, logits):
+        out = {}
+        off = 0
+        for k in self._order:
+            d = self._heads[k].logits_dimension
+            out[k] = logits[:, off:off + d]
+            off += d
+        return out
+
+    def loss(self, logits, labels):
+        parts = self._split(logits)
+        total = None
+        for k in self._order:
+            term = self._heads[k].loss(parts[k], labels[k])
+            total = term if total is None else total + term
+        return total
+
+    def predictions(self, logits):
+        parts = self._split(logits)
+        out = {}
+        for k in self._order:
+            for name, v in self._heads[k].predictions(parts[k]).items():
+                out["%s/%s" % (k, name)] = v
+        return out
+
+    def metrics(self, logits, labels):
+        parts = self._split(logits)
+        out = {}
+        for k in self._order:
+            for name, v in self._heads[k].metrics(parts[k],
+                                                  labels[k]).items():
+                out["%s/%s" % (k, name)] = v
+        out["average_loss"] = float(self.loss(logits, labels).detach().cpu())
+        return out
+
+
 class RegressionHead(Head):
     """Mean-squared-error head (the reference tests' regression head,
     adanet/core/testing_utils.py:236)."""

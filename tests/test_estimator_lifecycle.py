@@ -260,3 +260,33 @@ def test_summaries_written(model_dir, synthetic_classification):
     est = _make_estimator(model_dir, input_fn)
     est.train(input_fn, max_steps=10)
     assert os.path.isdir(os.path.join(model_dir, "summaries"))
+
+
+def test_multi_head_lifecycle(model_dir):
+    """Multi-head estimator (reference estimator_test.py:1517)."""
+    from adanet_amd.head import MultiClassHead, MultiHead
+
+    torch.manual_seed(0)
+    N, D = 256, 16
+    X = torch.randn(N, D)
+    W1, W2 = torch.randn(D, 3), torch.randn(D, 4)
+    Y = {"a": (X @ W1).argmax(1), "b": (X @ W2).argmax(1)}
+
+    def input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(5)
+            while True:
+                idx = torch.randint(0, N, (64,), generator=g)
+                yield X[idx], {k: v[idx] for k, v in Y.items()}
+
+        return gen()
+
+    head = MultiHead({"a": MultiClassHead(3), "b": MultiClassHead(4)})
+    est = adanet_amd.Estimator(
+        head=head,
+        subnetwork_generator=simple_dnn.Generator(layer_size=16),
+        max_iteration_steps=10, model_dir=model_dir,
+        config=adanet_amd.RunConfig(tf_random_seed=3))
+    est.train(input_fn, max_steps=20)
+    res = est.evaluate(input_fn, steps=4)
+    assert "a/accuracy" in res and "b/accuracy" in res
