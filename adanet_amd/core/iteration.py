@@ -396,8 +396,14 @@ class _Iteration(object):
                         if (self.placement is not None
                                 and self.placement.data_parallel
                                 and comm.is_initialized()):
-                            comm.allreduce_gradients(
-                                list(spec.subnetwork.module.parameters()))
+                            arenas = getattr(spec.optimizer, "_arenas", None)
+                            if arenas:
+                                # Zero-copy flat buckets over xGMI.
+                                comm.allreduce_buffers(
+                                    spec.optimizer.flat_grad_buffers())
+                            else:
+                                comm.allreduce_gradients(
+                                    list(spec.subnetwork.module.parameters()))
                         spec.optimizer.step()
                     losses_row[spec.name] = loss.detach()
                 else:

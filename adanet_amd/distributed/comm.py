@@ -80,6 +80,17 @@ def allreduce_mean_(t: torch.Tensor):
     return t
 
 
+def allreduce_buffers(buffers: Sequence[torch.Tensor]):
+    """Mean all-reduce of pre-flattened gradient buckets (zero-copy: the
+    fused optimizers expose their parameter-arena grad buffers)."""
+    if not is_initialized():
+        return
+    w = world_size()
+    for b in buffers:
+        dist.all_reduce(b, op=dist.ReduceOp.SUM)
+        b.div_(w)
+
+
 def allreduce_gradients(params: Sequence[torch.nn.Parameter]):
     """Flat-bucket gradient all-reduce (mean) over all ranks.
 

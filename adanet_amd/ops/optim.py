@@ -23,6 +23,146 @@ import torch
 from adanet_amd.ops import _extension
 
 
+class _ArenaMixin(object):
+    """Transparent parameter-arena flattening for the fused optimizers.
+
+    On the first GPU step, each param group's CUDA parameters coalesce by
+    dtype into ONE contiguous working buffer (plus flat master / momentum /
+    Adam state); `p.data` and `p.grad` become views, so:
+
+      * the whole update is one fused kernel per (group, dtype) instead of
+        one per tensor,
+      * data-parallel gradient all-reduce is ONE zero-copy xGMI bucket per
+        candidate (`flat_grad_buffers()` — used by distributed/comm.py),
+      * `zero_grad` is one memset per buffer (and re-pins the grad views,
+        so the engine's `set_to_none=True` is safely ignored).
+
+    The arena is built lazily after the first backward (grads must exist to
+    seed the views), which is also before any hipGraph capture (warmup
+    steps), so the captured graph sees only stable flat buffers.
+    """
+
+    def _arena_init(self):
+        self._arenas = None  # list of dicts per (group, dtype)
+
+    def _maybe_flatten(self):
+        if self._arenas is not None:
+            return
+        arenas = []
+        for gi, group in enumerate(self.param_groups):
+            by_dtype = {}
+            for p in group["params"]:
+                if not p.is_cuda or p.grad is None:
+                    continue
+                by_dtype.setdefault(p.dtype, []).append(p)
+            for dtype, params in by_dtype.items():
+                if len(params) < 2:
+                    continue
+                n = sum(p.numel() for p in params)
+                device = params[0].device
+                flat_p = torch.empty(n, device=device, dtype=dtype)
+                flat_g = torch.zeros(n, device=device, dtype=dtype)
+                off = 0
+                views = []
+                for p in params:
+                    k = p.numel()
+                    flat_p[off:off + k] = p.data.reshape(-1)
+                    flat_g[off:off + k] = p.grad.reshape(-1).to(dtype)
+                    p.data = flat_p[off:off + k].view_as(p.data)
+                    p.grad = flat_g[off:off + k].view_as(p.data)
+                    views.append((p, off, k))
+                    # migrate any per-param state already created
+                    self.state.pop(p, None)
+                    off += k
+                arenas.append({
+                    "group": gi, "dtype": dtype, "params": params,
+                    "flat_p": flat_p, "flat_g": flat_g, "views": views,
+                    "state": {},
+                })
+        self._arenas = arenas
+
+    def _arena_params(self):
+        covered = set()
+        for a in self._arenas or []:
+            for p in a["params"]:
+                covered.add(id(p))
+        return covered
+
+    def flat_grad_buffers(self):
+        """Flat gradient buckets for zero-copy all-reduce (+ leftover
+        per-param grads)."""
+        bufs = []
+        if self._arenas:
+            for a in self._arenas:
+                bufs.append(a["flat_g"])
+        covered = self._arena_params()
+        for group in self.param_groups:
+            for p in group["params"]:
+                if id(p) not in covered and p.grad is not None:
+                    bufs.append(p.grad)
+        return bufs
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self._arenas:
+            for a in self._arenas:
+                a["flat_g"].zero_()
+                for p, off, k in a["views"]:
+                    if p.grad is None or p.grad.data_ptr() != a[
+                            "flat_g"][off:off + k].data_ptr():
+                        p.grad = a["flat_g"][off:off + k].view_as(p.data)
+            covered = self._arena_params()
+            for group in self.param_groups:
+                for p in group["params"]:
+                    if id(p) not in covered and p.grad is not None:
+                        if set_to_none:
+                            p.grad = None
+                        else:
+                            p.grad.zero_()
+            return
+        super().zero_grad(set_to_none=set_to_none)
+
+    def _arena_state_dict(self):
+        out = []
+        for a in self._arenas or []:
+            out.append({
+                "group": a["group"], "dtype": str(a["dtype"]),
+                "flat_p": a["flat_p"].detach().cpu(),
+                "state": {k: (v.detach().cpu() if torch.is_tensor(v) else v)
+                          for k, v in a["state"].items()},
+            })
+        return out
+
+    def _arena_load_state_dict(self, blobs):
+        if not blobs or self._arenas is None:
+            return False
+        if len(blobs) != len(self._arenas):
+            return False
+        for a, blob in zip(self._arenas, blobs):
+            a["flat_p"].copy_(blob["flat_p"].to(a["flat_p"].device))
+            for k, v in blob["state"].items():
+                a["state"][k] = (v.to(a["flat_p"].device)
+                                 if torch.is_tensor(v) else v)
+        return True
+
+    def state_dict(self):
+        sd = super().state_dict()
+        if self._arenas:
+            sd["_arenas"] = self._arena_state_dict()
+        return sd
+
+    def load_state_dict(self, sd):
+        blobs = sd.pop("_arenas", None)
+        try:
+            super().load_state_dict(sd)
+        except Exception:
+            pass  # per-param state may not round-trip across arena builds
+        if blobs is not None:
+            if self._arenas is None:
+                self._pending_arena_state = blobs
+            else:
+                self._arena_load_state_dict(blobs)
+
+
 def _fp32_sgd_update(p, g, state, lr, momentum, dampening, weight_decay,
                      nesterov):
     master = state.get("master")
@@ -43,8 +183,9 @@ def _fp32_sgd_update(p, g, state, lr, momentum, dampening, weight_decay,
     p.data.copy_(master.to(p.dtype))
 
 
-class FusedSGD(torch.optim.Optimizer):
-    """SGD with momentum/nesterov; fused single-pass kernel for bf16 params."""
+class FusedSGD(_ArenaMixin, torch.optim.Optimizer):
+    """SGD with momentum/nesterov; fused single-pass kernel for bf16 params
+    with automatic parameter-arena flattening (see _ArenaMixin)."""
 
     def __init__(self, params, lr: float = 0.01, momentum: float = 0.0,
                  dampening: float = 0.0, weight_decay: float = 0.0,
@@ -54,13 +195,48 @@ class FusedSGD(torch.optim.Optimizer):
         defaults = dict(lr=lr, momentum=momentum, dampening=dampening,
                         weight_decay=weight_decay, nesterov=nesterov)
         super().__init__(params, defaults)
+        self._arena_init()
+        self._pending_arena_state = None
+
+    def _arena_step(self, a):
+        ext = _extension.require()
+        group = self.param_groups[a["group"]]
+        state = a["state"]
+        mom = None
+        if group["momentum"]:
+            mom = state.get("momentum_buffer")
+            if mom is None:
+                mom = torch.zeros(a["flat_p"].numel(),
+                                  device=a["flat_p"].device,
+                                  dtype=torch.float32)
+                state["momentum_buffer"] = mom
+        if a["dtype"] == torch.bfloat16:
+            master = state.get("master")
+            if master is None:
+                master = a["flat_p"].float()
+                state["master"] = master
+            ext.fused_sgd(master, a["flat_p"], a["flat_g"], mom,
+                          group["lr"], group["momentum"],
+                          group["dampening"], group["weight_decay"],
+                          group["nesterov"], 1.0)
+        else:
+            ext.fused_sgd_fp32(a["flat_p"], a["flat_g"], mom, group["lr"],
+                               group["momentum"], group["dampening"],
+                               group["weight_decay"], group["nesterov"], 1.0)
 
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
+        self._maybe_flatten()
+        if self._pending_arena_state is not None and self._arenas:
+            self._arena_load_state_dict(self._pending_arena_state)
+            self._pending_arena_state = None
+        covered = self._arena_params()
+        for a in self._arenas or []:
+            self._arena_step(a)
         for group in self.param_groups:
             for p in group["params"]:
-                if p.grad is None:
+                if p.grad is None or id(p) in covered:
                     continue
                 state = self.state[p]
                 if p.is_cuda and p.dtype == torch.bfloat16:
@@ -103,22 +279,60 @@ class FusedSGD(torch.optim.Optimizer):
         return loss
 
 
-class FusedAdam(torch.optim.Optimizer):
-    """Adam; fused single-pass kernel for bf16 params (fp32 master + m/v)."""
+class FusedAdam(_ArenaMixin, torch.optim.Optimizer):
+    """Adam; fused single-pass kernel for bf16 params (fp32 master + m/v)
+    with automatic parameter-arena flattening (see _ArenaMixin)."""
 
     def __init__(self, params, lr: float = 1e-3, betas=(0.9, 0.999),
                  eps: float = 1e-8, weight_decay: float = 0.0):
         defaults = dict(lr=lr, betas=betas, eps=eps,
                         weight_decay=weight_decay)
         super().__init__(params, defaults)
+        self._arena_init()
+        self._pending_arena_state = None
+
+    def _arena_step(self, a):
+        ext = _extension.require()
+        group = self.param_groups[a["group"]]
+        b1, b2 = group["betas"]
+        state = a["state"]
+        state["step"] = state.get("step", 0) + 1
+        if "exp_avg" not in state:
+            n = a["flat_p"].numel()
+            dev = a["flat_p"].device
+            state["exp_avg"] = torch.zeros(n, device=dev,
+                                           dtype=torch.float32)
+            state["exp_avg_sq"] = torch.zeros(n, device=dev,
+                                              dtype=torch.float32)
+        if a["dtype"] == torch.bfloat16:
+            master = state.get("master")
+            if master is None:
+                master = a["flat_p"].float()
+                state["master"] = master
+            ext.fused_adam(master, a["flat_p"], a["flat_g"],
+                           state["exp_avg"], state["exp_avg_sq"],
+                           group["lr"], b1, b2, group["eps"],
+                           group["weight_decay"], state["step"], 1.0)
+        else:
+            ext.fused_adam_fp32(a["flat_p"], a["flat_g"], state["exp_avg"],
+                                state["exp_avg_sq"], group["lr"], b1, b2,
+                                group["eps"], group["weight_decay"],
+                                state["step"], 1.0)
 
     @torch.no_grad()
     def step(self, closure=None):
         loss = closure() if closure is not None else None
+        self._maybe_flatten()
+        if self._pending_arena_state is not None and self._arenas:
+            self._arena_load_state_dict(self._pending_arena_state)
+            self._pending_arena_state = None
+        covered = self._arena_params()
+        for a in self._arenas or []:
+            self._arena_step(a)
         for group in self.param_groups:
             b1, b2 = group["betas"]
             for p in group["params"]:
-                if p.grad is None:
+                if p.grad is None or id(p) in covered:
                     continue
                 state = self.state[p]
                 state["step"] = state.get("step", 0) + 1

@@ -73,7 +73,7 @@ def main():
     loss = torch.empty(B, device=dev, dtype=torch.float32)
     probs = torch.empty(B, C, device=dev, dtype=torch.bfloat16)
     sec = bench_kernel(
-        lambda: ext.softmax_xent_fwd(logits, labels, loss, probs, 0.0))
+        lambda: ext.softmax_xent_fwd(logits, labels, loss, probs, 0.0, None))
     out["xent fwd 16k x 10 us"] = round(sec * 1e6, 1)
 
     print(json.dumps(out, indent=1))
