@@ -22,6 +22,7 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
   const int wave_in_block = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int waves = (gridDim.x * blockDim.x) >> 6;
+  float block_sum = 0.f;  // per-wave running sum for the fused mean
   for (int row = blockIdx.x * (blockDim.x >> 6) + wave_in_block; row < B;
        row += waves) {
     const bf16_t* lrow = logits + (int64_t)row * ldl;
@@ -48,8 +49,20 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
     if (lane == 0) {
       const float l = lse - (1.f - eps) * ly - (eps / C) * sum_logits;
       if (loss) loss[row] = l;
-      // fused mean: one atomic per row (mean_out pre-zeroed).
-      if (mean_out) atomicAdd(mean_out, l / B);
+      block_sum += l;
+    }
+  }
+  // Fused mean: per-block LDS reduction -> ONE global atomic per block
+  // (a per-row atomic to a single address serialized ~6x the kernel).
+  if (mean_out) {
+    __shared__ float partial[4];
+    if (lane == 0) partial[wave_in_block] = block_sum;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      float s = 0.f;
+      const int nwaves = blockDim.x >> 6;
+      for (int w = 0; w < nwaves; ++w) s += partial[w];
+      atomicAdd(mean_out, s / B);
     }
   }
 }
@@ -86,7 +99,10 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   TORCH_CHECK(logits.stride(1) == 1, "xent: contiguous class dim");
   const int B = (int)logits.size(0), C = (int)logits.size(1);
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int blocks = std::min((B + 3) / 4, 2048);
+  const bool fused_mean = mean_out.has_value() && mean_out->defined();
+  // With the fused mean, cap blocks so the final atomics stay few; each
+  // wave loops multiple rows instead.
+  const int blocks = std::min((B + 3) / 4, fused_mean ? 256 : 2048);
   float* loss_ptr = (loss.has_value() && loss->defined())
                         ? loss->data_ptr<float>()
                         : nullptr;
