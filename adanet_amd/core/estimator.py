@@ -217,8 +217,12 @@ class Estimator(object):
         return self._global_step
 
     def train(self, input_fn, steps: Optional[int] = None,
-              max_steps: Optional[int] = None):
-        """The AdaNet while-loop (reference estimator.py:809-999)."""
+              max_steps: Optional[int] = None, hooks=None):
+        """The AdaNet while-loop (reference estimator.py:809-999).
+
+        ``hooks``: optional adanet_amd.hooks.TrainHook objects invoked
+        around every lockstep step (plus any hooks Builders attach via
+        TrainOpSpec)."""
         if steps is not None and max_steps is not None:
             raise ValueError("Can not provide both steps and max_steps.")
         if (steps is not None and steps <= 0) or (max_steps is not None
@@ -231,6 +235,7 @@ class Estimator(object):
             budget_end = max_steps
         if self._global_step >= budget_end:
             return self
+        user_hooks = list(hooks or [])
 
         # The reference's workers wait up to worker_wait_timeout_secs for the
         # chief (estimator.py:951-996, checkpoint polling); here the analog
@@ -253,12 +258,16 @@ class Estimator(object):
             t_build0 = time.perf_counter()
             iteration = self._get_or_build_iteration(input_fn)
             self._phase_secs["build"] += time.perf_counter() - t_build0
+            active_hooks = user_hooks + list(iteration.builder_hooks)
+            for h in active_hooks:
+                h.begin(estimator=self, iteration=iteration)
             if input_iter is None:
                 input_iter = iter(input_fn())
             t_train0 = time.perf_counter()
             iteration_ended = True
+            stop_requested = False
             while not iteration.is_over():
-                if self._global_step >= budget_end:
+                if self._global_step >= budget_end or stop_requested:
                     iteration_ended = False
                     break
                 try:
@@ -271,8 +280,14 @@ class Estimator(object):
                 features, labels = _to_device(features, labels, self._device)
                 if self._debug:
                     self._check_finite(features, labels)
+                for h in active_hooks:
+                    h.before_step(self._global_step)
                 iteration.train_step(features, labels)
                 self._global_step += 1
+                for h in active_hooks:
+                    h.after_step(self._global_step)
+                    if getattr(h, "should_stop", False):
+                        stop_requested = True
                 if (self._config.save_checkpoints_steps and self._global_step %
                         self._config.save_checkpoints_steps == 0):
                     self._save_checkpoint(mid_iteration=True)
@@ -281,6 +296,11 @@ class Estimator(object):
                              self._global_step, t, iteration.step)
             iteration.flush_losses()
             self._phase_secs["train"] += time.perf_counter() - t_train0
+            for h in active_hooks:
+                h.end(estimator=self)
+            if stop_requested and not iteration.is_over():
+                self._save_checkpoint(mid_iteration=True)
+                break
             if not iteration.is_over():
                 # Budget or input ran out mid-iteration: checkpoint so a
                 # restart resumes this iteration in place (reference
@@ -560,6 +580,7 @@ class Estimator(object):
                                       is_chief=comm.is_chief())
         # --- subnetwork specs (placement-gated: reference iteration.py:629) ---
         sub_specs: List[_SubnetworkSpec] = []
+        builder_hooks = []
         n = len(builders)
         for i, b in enumerate(builders):
             name = "t{}_{}".format(t, b.name)
@@ -584,7 +605,12 @@ class Estimator(object):
                 if (self._placement.data_parallel and comm.is_initialized()):
                     comm.broadcast_state_dict(sub.module, src=0)
                 opt = b.build_optimizer(sub.module.parameters(), iteration=t)
-                opt = getattr(opt, "optimizer", opt)  # TrainOpSpec unwrap
+                if hasattr(opt, "optimizer"):  # TrainOpSpec: collect hooks
+                    builder_hooks.extend(getattr(opt, "hooks", ()) or ())
+                    if comm.is_chief():
+                        builder_hooks.extend(
+                            getattr(opt, "chief_hooks", ()) or ())
+                    opt = opt.optimizer
                 sub_specs.append(
                     _SubnetworkSpec(
                         name=name, builder=b, subnetwork=sub, optimizer=opt,
@@ -656,6 +682,7 @@ class Estimator(object):
                 self._replicate_ensemble_in_training),
             to_device=lambda f, l: _to_device(f, l, self._device),
             use_graphs=self._use_hip_graphs)
+        iteration.builder_hooks = builder_hooks
         self._restore_iteration_state(iteration)
         self._current_iteration = iteration
         return iteration

@@ -185,6 +185,7 @@ class _Iteration(object):
         self._row_recorded: List[set] = []  # names actually written per row
         self._nan_scalar = None
         self._frozen_event = None
+        self.builder_hooks = []  # TrainOpSpec hooks collected by the engine
         # hipGraph state
         self._use_graphs = use_graphs
         self._graph = None

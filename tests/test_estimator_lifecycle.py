@@ -290,3 +290,54 @@ def test_multi_head_lifecycle(model_dir):
     est.train(input_fn, max_steps=20)
     res = est.evaluate(input_fn, steps=4)
     assert "a/accuracy" in res and "b/accuracy" in res
+
+
+def test_train_hooks_and_stop(model_dir, synthetic_classification):
+    """Hook callbacks + StopAfterSteps early stop (reference
+    _StopAfterTrainingHook, estimator.py:50-87) and TrainOpSpec hooks."""
+    from adanet_amd import hooks as hooks_lib
+    from adanet_amd.subnetwork import TrainOpSpec
+    X, Y, input_fn = synthetic_classification
+    calls = {"begin": 0, "before": 0, "after": 0, "end": 0}
+
+    class _H(hooks_lib.TrainHook):
+
+        def begin(self, estimator=None, iteration=None):
+            calls["begin"] += 1
+
+        def before_step(self, gs):
+            calls["before"] += 1
+
+        def after_step(self, gs):
+            calls["after"] += 1
+
+        def end(self, estimator=None):
+            calls["end"] += 1
+
+    est = _make_estimator(model_dir, input_fn)
+    stopper = hooks_lib.StopAfterSteps(4)
+    est.train(input_fn, max_steps=30, hooks=[_H(), stopper])
+    assert est.global_step == 4  # stopped mid-iteration 0
+    assert calls["begin"] == 1 and calls["end"] == 1
+    assert calls["before"] == 4 and calls["after"] == 4
+
+    # TrainOpSpec-attached hooks reach the loop too.
+    spec_calls = []
+
+    class _SpecHook(hooks_lib.TrainHook):
+
+        def after_step(self, gs):
+            spec_calls.append(gs)
+
+    class _HookedBuilder(_GoodBuilder):
+
+        def build_optimizer(self, params, iteration=0):
+            from adanet_amd.ops.optim import FusedSGD
+            return TrainOpSpec(FusedSGD(params, lr=0.01),
+                               hooks=(_SpecHook(),))
+
+    est2 = _make_estimator(str(model_dir) + "2", input_fn,
+                           subnetwork_generator=SimpleGenerator(
+                               [_HookedBuilder()]))
+    est2.train(input_fn, steps=3)
+    assert len(spec_calls) == 3
