@@ -273,8 +273,14 @@ class _Iteration(object):
 
     def _graph_eligible(self) -> bool:
         if self._graph_ok is None:
+            # Graphs need a collective-free step: single process, or any
+            # world size under RoundRobin (candidates are rank-local; the
+            # only collectives run at iteration end, outside the graph).
+            no_step_collectives = (comm.world_size() <= 1
+                                   or (self.placement is not None
+                                       and not self.placement.data_parallel))
             ok = (self._use_graphs and self.device.type == "cuda"
-                  and not comm.is_initialized())
+                  and no_step_collectives)
             if ok:
                 for spec in self.subnetwork_specs:
                     if spec.train_input_fn is not None:
@@ -396,7 +402,7 @@ class _Iteration(object):
                         loss.backward()
                         if (self.placement is not None
                                 and self.placement.data_parallel
-                                and comm.is_initialized()):
+                                and comm.world_size() > 1):
                             arenas = getattr(spec.optimizer, "_arenas", None)
                             if arenas:
                                 # Zero-copy flat buckets over xGMI.
