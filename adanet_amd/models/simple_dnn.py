@@ -63,25 +63,31 @@ class _SimpleDNNBuilder(Builder):
     """Reference adanet/examples/simple_dnn.py:26-133."""
 
     def __init__(self, optimizer_fn, layer_size: int, num_layers: int,
-                 learn_mixture_weights: bool, dropout: float, seed=None):
+                 learn_mixture_weights: bool, dropout: float, seed=None,
+                 name_suffix: str = ""):
         self._optimizer_fn = optimizer_fn
         self._layer_size = layer_size
         self._num_layers = num_layers
         self._learn_mixture_weights = learn_mixture_weights
         self._dropout = dropout
         self._seed = seed
+        self._name_suffix = name_suffix
 
     @property
     def name(self) -> str:
         # Reference naming: "linear" for 0 layers else "<n>_layer_dnn" (:129).
-        if self._num_layers == 0:
-            return "linear"
-        return "{}_layer_dnn".format(self._num_layers)
+        base = ("linear" if self._num_layers == 0 else
+                "{}_layer_dnn".format(self._num_layers))
+        return base + self._name_suffix
 
     def build_subnetwork(self, features, logits_dimension, training,
                          previous_ensemble=None) -> Subnetwork:
         if self._seed is not None:
-            torch.manual_seed(self._seed + self._num_layers)
+            # stable across processes (python str hash is randomized)
+            suffix_code = sum(ord(c) * (i + 1)
+                              for i, c in enumerate(self._name_suffix))
+            torch.manual_seed(self._seed + self._num_layers * 7919 +
+                              suffix_code)
         module = _DNNModule(_feature_dim(features), self._num_layers,
                             self._layer_size, logits_dimension,
                             self._dropout if training else 0.0)
@@ -111,12 +117,15 @@ class _SimpleDNNBuilder(Builder):
 
 class Generator(Generator):
     """Two candidates per iteration: same depth + one deeper
-    (reference simple_dnn.py:134-213)."""
+    (reference simple_dnn.py:134-213). ``num_restarts`` widens the search
+    with independently-initialized copies of each candidate ("_r<i>"
+    suffixes) — the pool a round-robin placement spreads over the node's
+    GPUs (2*restarts candidates per iteration)."""
 
     def __init__(self, optimizer_fn=None, layer_size: int = 32,
                  initial_num_layers: int = 0,
                  learn_mixture_weights: bool = False, dropout: float = 0.0,
-                 seed: Optional[int] = None):
+                 seed: Optional[int] = None, num_restarts: int = 1):
         if optimizer_fn is None:
             optimizer_fn = functools.partial(FusedSGD, lr=0.01)
         self._builder_fn = functools.partial(
@@ -127,6 +136,7 @@ class Generator(Generator):
             dropout=dropout,
             seed=seed)
         self._initial_num_layers = initial_num_layers
+        self._num_restarts = max(1, int(num_restarts))
 
     def generate_candidates(self, previous_ensemble, iteration_number,
                             previous_ensemble_reports, all_reports,
@@ -137,7 +147,11 @@ class Generator(Generator):
             shared = last.shared or {}
             num_layers = int(shared.get("num_layers",
                                         self._initial_num_layers))
-        return [
-            self._builder_fn(num_layers=num_layers),
-            self._builder_fn(num_layers=num_layers + 1),
-        ]
+        out = []
+        for r in range(self._num_restarts):
+            suffix = "" if r == 0 else "_r{}".format(r)
+            out.append(self._builder_fn(num_layers=num_layers,
+                                        name_suffix=suffix))
+            out.append(self._builder_fn(num_layers=num_layers + 1,
+                                        name_suffix=suffix))
+        return out

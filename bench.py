@@ -37,10 +37,11 @@ def parse_args():
     p.add_argument("--hidden", type=int, default=2048)
     p.add_argument("--train-steps-per-iter", type=int, default=150)
     p.add_argument("--eval-batches", type=int, default=8)
-    p.add_argument("--candidates", type=int, default=2,
-                   help="candidates per iteration (paper: 2)")
+    p.add_argument("--restarts", type=int, default=0,
+                   help="random restarts per candidate depth (0 = one per "
+                        "GPU so round-robin owns 2 candidates per rank)")
     p.add_argument("--placement", choices=["replication", "round_robin"],
-                   default="replication")
+                   default="round_robin")
     p.add_argument("--lr", type=float, default=0.05)
     p.add_argument("--model-dir", default=None)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
@@ -114,13 +115,19 @@ def main():
                  else RoundRobinStrategy())
 
     # Paper-style CIFAR DNN search space: candidates at the current depth
-    # and one deeper, width --hidden, fused momentum-SGD.
+    # and one deeper, width --hidden, fused momentum-SGD. The pool widens
+    # with the GPU count (independent restarts) so round-robin placement
+    # keeps 2 candidates per GPU at every N — the north star's
+    # candidate-per-GPU scaling axis with constant per-GPU work.
+    restarts = args.restarts if args.restarts > 0 else world
     generator = simple_dnn.Generator(
         optimizer_fn=functools.partial(FusedSGD, lr=args.lr, momentum=0.9),
         layer_size=args.hidden,
         initial_num_layers=1,
         learn_mixture_weights=True,
-        dropout=0.0)
+        dropout=0.0,
+        seed=77,
+        num_restarts=restarts)
 
     est = adanet_amd.Estimator(
         head=MultiClassHead(C),
@@ -189,7 +196,7 @@ def main():
                 "input_dim": D,
                 "n_classes": C,
                 "hidden": args.hidden,
-                "candidates_per_iter": 2,
+                "candidates_per_iter": 2 * restarts,
                 "train_steps_per_iter": args.train_steps_per_iter,
                 "eval_batches": args.eval_batches,
                 "parallelism": ("dp%d" % world
