@@ -55,6 +55,28 @@ from adanet_amd.subnetwork.generator import Generator, Subnetwork
 log = logging.getLogger("adanet_amd")
 
 
+class _roctx(object):
+    """rocTX range markers around engine phases (the reference
+    only had ProfilerHook in a test; here ranges are first-class, shown in
+    rocprofv3 --sys-trace). Enabled with ADANET_ROCTX=1; torch.cuda.nvtx
+    maps to rocTX on ROCm."""
+
+    enabled = os.environ.get("ADANET_ROCTX", "") not in ("", "0")
+
+    def __init__(self, name):
+        self._name = name
+
+    def __enter__(self):
+        if self.enabled and torch.cuda.is_available():
+            torch.cuda.nvtx.range_push(self._name)
+        return self
+
+    def __exit__(self, *a):
+        if self.enabled and torch.cuda.is_available():
+            torch.cuda.nvtx.range_pop()
+        return False
+
+
 def _to_device(features, labels, device, dtype=torch.bfloat16):
     def conv(t):
         if not torch.is_tensor(t):
@@ -282,7 +304,8 @@ class Estimator(object):
                     self._check_finite(features, labels)
                 for h in active_hooks:
                     h.before_step(self._global_step)
-                iteration.train_step(features, labels)
+                with _roctx("adanet/train_step"):
+                    iteration.train_step(features, labels)
                 self._global_step += 1
                 for h in active_hooks:
                     h.after_step(self._global_step)
@@ -311,7 +334,8 @@ class Estimator(object):
                 break
             log.info("Finished training Adanet iteration %s", t)
             t_book0 = time.perf_counter()
-            self._execute_bookkeeping_phase(input_fn)
+            with _roctx("adanet/bookkeeping"):
+                self._execute_bookkeeping_phase(input_fn)
             self._phase_secs["bookkeeping"] += time.perf_counter() - t_book0
             self._current_iteration = None
             self._iteration_number += 1
@@ -323,6 +347,15 @@ class Estimator(object):
             if input_iter is None:
                 break
         return self
+
+    def train_and_evaluate(self, train_input_fn, eval_input_fn,
+                           max_steps: Optional[int] = None,
+                           eval_steps: Optional[int] = None,
+                           hooks=None) -> Dict[str, float]:
+        """tf.estimator.train_and_evaluate analog: train to max_steps, then
+        evaluate the frozen best ensemble."""
+        self.train(train_input_fn, max_steps=max_steps, hooks=hooks)
+        return self.evaluate(eval_input_fn, steps=eval_steps)
 
     def evaluate(self, input_fn, steps: Optional[int] = None,
                  checkpoint_path: Optional[str] = None) -> Dict[str, float]:
