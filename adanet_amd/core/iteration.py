@@ -186,6 +186,10 @@ class _Iteration(object):
         self._nan_scalar = None
         self._frozen_event = None
         self.builder_hooks = []  # TrainOpSpec hooks collected by the engine
+        # HBM frozen-logit cache (per-iteration lifetime).
+        self._frozen_cache: Dict = {}
+        self._frozen_static = None
+        self._need_last = None
         # hipGraph state
         self._use_graphs = use_graphs
         self._graph = None
@@ -222,6 +226,67 @@ class _Iteration(object):
                 outputs[name] = (last, logits)
         return outputs
 
+    def _need_frozen_last_layer(self) -> bool:
+        if self._need_last is None:
+            need = False
+            for spec in self.ensemble_specs:
+                ens = spec.ensemble
+                if ens is None:
+                    continue
+                if getattr(ens, "mixture_weights", "x") is None:
+                    need = True  # MATRIX mixtures consume last_layer
+                if getattr(ens, "add_mean_last_layer_predictions", False):
+                    need = True
+            self._need_last = need
+        return self._need_last
+
+    def _prepare_frozen(self, features, training: bool = False):
+        """Frozen outputs for this batch, staged into STATIC buffers.
+
+        The HBM logit cache (north star): inputs carrying an
+        ``adanet_cache_key`` attribute (resident-dataset batches — e.g. the
+        whole CIFAR-sized pool fits trivially in 288 GB) have their frozen
+        members' outputs computed ONCE per iteration and replayed from HBM
+        afterwards, so the per-step cost of the growing ensemble stays
+        flat across epochs. Results land in static buffers so the
+        hipGraph-captured step can consume them as plain graph inputs.
+        """
+        if not self.frozen_subnetworks:
+            return {}
+        key = getattr(features, "adanet_cache_key", None) if torch.is_tensor(
+            features) else None
+        cacheable = (key is not None
+                     and not (training
+                              and self.replicate_ensemble_in_training))
+        keep_last = self._need_frozen_last_layer()
+        if cacheable and key in self._frozen_cache:
+            cached = self._frozen_cache[key]
+            for name, (last_c, logits_c) in cached.items():
+                last_s, logits_s = self._frozen_static[name]
+                logits_s.copy_(logits_c)
+                if keep_last and last_c is not None:
+                    last_s.copy_(last_c)
+            return self._frozen_static
+        fresh = self.compute_frozen_outputs(features, training=training)
+        if self._frozen_static is None or any(
+                self._frozen_static[n][1].shape != o[1].shape
+                for n, o in fresh.items()):
+            self._frozen_static = {
+                n: (last.clone(), logits.clone())
+                for n, (last, logits) in fresh.items()
+            }
+        else:
+            for n, (last, logits) in fresh.items():
+                last_s, logits_s = self._frozen_static[n]
+                logits_s.copy_(logits)
+                last_s.copy_(last)
+        if cacheable:
+            self._frozen_cache[key] = {
+                n: ((last.clone() if keep_last else None), logits.clone())
+                for n, (last, logits) in fresh.items()
+            }
+        return self._frozen_static
+
     def train_step(self, features, labels) -> None:
         """One lockstep training step for every still-active spec
         (the reference's single session.run over all candidate train ops,
@@ -232,6 +297,9 @@ class _Iteration(object):
         otherwise dominates (measured ~3.5 ms host vs ~1.1 ms GPU per step
         on the CIFAR DNN bench before graphing)."""
         tm = self.train_manager
+        # Frozen members run (or replay from the HBM cache) OUTSIDE the
+        # captured graph; the step consumes their static buffers.
+        self._prepare_frozen(features, training=True)
         if self._graph_eligible():
             self._graphed_train_step(features, labels)
         else:
@@ -357,9 +425,11 @@ class _Iteration(object):
         self._finish_loss_row(self._graph_recorded)
 
     def _device_step(self, features, labels):
-        """All of one step's device work; returns name -> loss tensor."""
+        """All of one step's device work; returns name -> loss tensor.
+        Frozen outputs come from the static buffers prepared by
+        train_step (outside any graph capture)."""
         tm = self.train_manager
-        frozen_out = self.compute_frozen_outputs(features, training=True)
+        frozen_out = self._frozen_static or {}
         if self._use_streams:
             self._frozen_event = torch.cuda.Event()
             self._frozen_event.record()
@@ -678,7 +748,7 @@ class _Iteration(object):
             except StopIteration:
                 break
             features, labels = to_device(features, labels)
-            frozen_out = self.compute_frozen_outputs(features)
+            frozen_out = self._prepare_frozen(features)
             with torch.no_grad():
                 for spec in self.subnetwork_specs:
                     if spec.subnetwork is None:

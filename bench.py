@@ -81,21 +81,25 @@ def main():
     # HBM-resident synthetic dataset: a pool of batches cycled forever.
     n_pool = 32
     pool = []
-    for _ in range(n_pool):
+    for i in range(n_pool):
         x = torch.randn(args.batch, D)
         y = (x @ teacher).argmax(dim=1)
         if use_gpu:
             x = x.to(device).to(torch.bfloat16)
             y = y.to(device)
+        # Resident-dataset batch: lets the engine replay frozen-member
+        # logits from the HBM cache instead of recomputing per step.
+        x.adanet_cache_key = ("train", rank, i)
         pool.append((x, y))
     eval_pool = []
     torch.manual_seed(5000 + rank)
-    for _ in range(args.eval_batches):
+    for i in range(args.eval_batches):
         x = torch.randn(args.batch, D)
         y = (x @ teacher).argmax(dim=1)
         if use_gpu:
             x = x.to(device).to(torch.bfloat16)
             y = y.to(device)
+        x.adanet_cache_key = ("eval", rank, i)
         eval_pool.append((x, y))
 
     def input_fn():

@@ -341,3 +341,45 @@ def test_train_hooks_and_stop(model_dir, synthetic_classification):
                                [_HookedBuilder()]))
     est2.train(input_fn, steps=3)
     assert len(spec_calls) == 3
+
+
+def test_frozen_logit_cache_correctness(model_dir, synthetic_classification):
+    """Cached frozen logits must give the same losses as recomputation:
+    run two iterations with cache keys vs without — identical winners and
+    eval losses."""
+    X, Y, input_fn = synthetic_classification
+
+    def keyed_input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(7)
+            batches = []
+            for i in range(8):
+                idx = torch.randint(0, X.shape[0], (64,), generator=g)
+                xb = X[idx].clone()
+                xb.adanet_cache_key = ("t", i)
+                batches.append((xb, Y[idx]))
+            i = 0
+            while True:
+                yield batches[i % 8]
+                i += 1
+
+        return gen()
+
+    est1 = _make_estimator(str(model_dir) + "_keyed", keyed_input_fn)
+    est1.train(keyed_input_fn, max_steps=20)
+    r1 = est1.evaluate(keyed_input_fn, steps=4)
+
+    def unkeyed_input_fn():
+        def gen():
+            for f, l in keyed_input_fn():
+                f2 = f.clone()  # drops the cache key attribute
+                yield f2, l
+
+        return gen()
+
+    est2 = _make_estimator(str(model_dir) + "_unkeyed", unkeyed_input_fn)
+    est2.train(unkeyed_input_fn, max_steps=20)
+    r2 = est2.evaluate(unkeyed_input_fn, steps=4)
+    assert r1["loss"] == pytest.approx(r2["loss"], abs=1e-4)
+    assert (r1["architecture/adanet/ensembles"] ==
+            r2["architecture/adanet/ensembles"])
