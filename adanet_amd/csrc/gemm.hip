@@ -53,21 +53,30 @@ __device__ __forceinline__ void stage_tile_nt(
 
 // BM x BN tile, 2x2 wave grid, FM x FN 16x16 MFMA fragments per wave,
 // KSTEP K-columns staged per barrier (KSTEP/32 MFMA K-slices).
-template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32>
+// SPLITK > 1: the grid carries SPLITK K-slabs per output tile; each slab
+// atomically accumulates fp32 partials into C32 (no bias/act), and a
+// separate epilogue kernel finishes bias+ReLU+bf16 — trades a little
+// output traffic for filling all 256 CUs with the efficient big tile.
+template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32,
+          bool SPLITK = false>
 __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
-    int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles) {
+    int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles,
+    float* __restrict__ C32 = nullptr, int ksplit = 1) {
   __shared__ bf16_t As[2][BM * KSTEP];
   __shared__ bf16_t Bs[2][BN * KSTEP];
 
   // Bijective XCD-aware swizzle (guide m204): contiguous tile chunks/XCD.
-  const int nwg = mtiles * ntiles;
+  const int nwg = mtiles * ntiles * (SPLITK ? ksplit : 1);
   const int orig = blockIdx.x;
   const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
   const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
                  (orig >> 3);
-  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  const int tiles = mtiles * ntiles;
+  const int split = SPLITK ? (wg / tiles) : 0;
+  const int tile_lin = SPLITK ? (wg % tiles) : wg;
+  const int tile_m = tile_lin / ntiles, tile_n = tile_lin % ntiles;
 
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
@@ -83,14 +92,20 @@ __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int ktiles = K / KSTEP;
-  stage_tile_nt<BM, KSTEP>(A, lda, row0, M, 0, As[0], wid, lane);
-  stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, 0, Bs[0], wid, lane);
+  int kt0 = 0, ktend = K / KSTEP;
+  if (SPLITK) {
+    const int per = (ktend + ksplit - 1) / ksplit;
+    kt0 = split * per;
+    ktend = min(ktend, kt0 + per);
+    if (kt0 >= ktend) return;
+  }
+  stage_tile_nt<BM, KSTEP>(A, lda, row0, M, kt0 * KSTEP, As[0], wid, lane);
+  stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, kt0 * KSTEP, Bs[0], wid, lane);
 
   int buf = 0;
-  for (int kt = 0; kt < ktiles; ++kt) {
+  for (int kt = kt0; kt < ktend; ++kt) {
     __syncthreads();  // staged tile `buf` visible; prior reads of buf^1 done
-    if (kt + 1 < ktiles) {
+    if (kt + 1 < ktend) {
       const int k0 = (kt + 1) * KSTEP;
       stage_tile_nt<BM, KSTEP>(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
       stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
@@ -127,16 +142,34 @@ __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
     for (int j = 0; j < FN; ++j) {
       const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
       if (col >= N) continue;
-      const float bv = bias ? bias[col] : 0.f;
+      const float bv = (!SPLITK && bias) ? bias[col] : 0.f;
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
         if (row >= M) continue;
-        float v = acc[i][j][rr] + bv;
-        if (act == 1) v = v > 0.f ? v : 0.f;
-        C[(int64_t)row * ldc + col] = f2bf(v);
+        if (SPLITK) {
+          atomicAdd(&C32[(int64_t)row * N + col], acc[i][j][rr]);
+        } else {
+          float v = acc[i][j][rr] + bv;
+          if (act == 1) v = v > 0.f ? v : 0.f;
+          C[(int64_t)row * ldc + col] = f2bf(v);
+        }
       }
     }
+  }
+}
+
+// Finishes a split-K accumulation: bf16 C = act(C32 + bias).
+__global__ __launch_bounds__(256) void gemm_splitk_epilogue_kernel(
+    const float* __restrict__ C32, const float* __restrict__ bias,
+    bf16_t* __restrict__ C, int M, int N, int ldc, int act) {
+  const int64_t total = (int64_t)M * N;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += (int64_t)gridDim.x * blockDim.x) {
+    const int m = (int)(p / N), n = (int)(p % N);
+    float v = C32[p] + (bias ? bias[n] : 0.f);
+    if (act == 1) v = v > 0.f ? v : 0.f;
+    C[(int64_t)m * ldc + n] = f2bf(v);
   }
 }
 
@@ -243,6 +276,23 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
                        b, c, nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);     \
   } while (0)
 
+#define LAUNCH_SK(BM, BN, FM, FN, MW, KS, SPLIT)                              \
+  do {                                                                        \
+    TORCH_CHECK(K % KS == 0, "probe: K %% KSTEP");                            \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    auto C32 = at::zeros({(int64_t)M, (int64_t)N},                            \
+                         A.options().dtype(at::kFloat));                      \
+    hipLaunchKernelGGL((gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, KS, true>),   \
+                       dim3(mt * nt * SPLIT), dim3(THREADS), 0,              \
+                       stream.stream(), a, b, c, nullptr, M, N, K, lda, ldb, \
+                       ldc, 0, mt, nt, C32.data_ptr<float>(), SPLIT);         \
+    const int64_t tot = (int64_t)M * N;                                       \
+    hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,                           \
+                       dim3((int)std::min<int64_t>((tot + 255) / 256, 2048)), \
+                       dim3(256), 0, stream.stream(),                         \
+                       C32.data_ptr<float>(), nullptr, c, M, N, ldc, 0);      \
+  } while (0)
+
   switch (variant) {
     case 0: LAUNCH_V(128, 128, 4, 4, 2, 32); break;
     case 1: LAUNCH_V(128, 128, 4, 4, 3, 32); break;
@@ -255,8 +305,14 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     case 8: LAUNCH_V(128, 64, 4, 2, 4, 64); break;
     case 9: LAUNCH_V(64, 64, 2, 2, 6, 32); break;
     case 10: LAUNCH_V(64, 128, 2, 4, 4, 32); break;
+    case 11: LAUNCH_SK(128, 128, 4, 4, 2, 32, 2); break;
+    case 12: LAUNCH_SK(128, 128, 4, 4, 2, 32, 4); break;
+    case 13: LAUNCH_SK(128, 128, 4, 4, 4, 32, 4); break;
+    case 14: LAUNCH_SK(128, 128, 4, 4, 4, 32, 2); break;
+    case 15: LAUNCH_SK(64, 64, 2, 2, 6, 32, 2); break;
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V
+#undef LAUNCH_SK
   HIP_CHECK_KERNEL();
 }

@@ -22,6 +22,11 @@ VARIANTS = {
     8: "128x64 f4x2 mw4 k64",
     9: "64x64 f2x2 mw6 k32",
     10: "64x128 f2x4 mw4 k32",
+    11: "128x128 mw2 splitk2",
+    12: "128x128 mw2 splitk4",
+    13: "128x128 mw4 splitk4",
+    14: "128x128 mw4 splitk2",
+    15: "64x64 mw6 splitk2",
 }
 
 SHAPES = [(2048, 2048, 3072), (4096, 2048, 3072), (2048, 3072, 2048),
