@@ -1,0 +1,138 @@
+"""AutoEnsemble integration (reference adanet/autoensemble/estimator_test.py:
+345-510): pools, bagging via private input pipelines, prediction_only."""
+
+import json
+import os
+
+import pytest
+import torch
+
+import adanet_amd
+from adanet_amd import AutoEnsembleEstimator, AutoEnsembleSubestimator
+from adanet_amd.autoensemble.common import _GeneratorFromCandidatePool
+from adanet_amd.head import MultiClassHead
+from adanet_amd.models.canned import DNNEstimator, LinearEstimator
+
+
+def _data(seed=0):
+    torch.manual_seed(seed)
+    N, D, C = 256, 12, 3
+    X = torch.randn(N, D)
+    Y = (X @ torch.randn(D, C)).argmax(dim=1)
+    return X, Y
+
+
+def _input_fn_factory(X, Y, seed=7, batch=32):
+    def input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(seed)
+            while True:
+                idx = torch.randint(0, X.shape[0], (batch,), generator=g)
+                yield X[idx], Y[idx]
+
+        return gen()
+
+    return input_fn
+
+
+def test_pool_dict_sorted_for_determinism():
+    head = MultiClassHead(3)
+    gen = _GeneratorFromCandidatePool({
+        "zeta": LinearEstimator(head),
+        "alpha": DNNEstimator(head, hidden_units=[8]),
+    })
+    builders = gen.generate_candidates(None, 0, [], [])
+    assert [b.name for b in builders] == ["alpha", "zeta"]
+
+
+def test_pool_callable_with_iteration():
+    head = MultiClassHead(3)
+
+    def pool(config, iteration_number):
+        return {"lin%d" % iteration_number: LinearEstimator(head)}
+
+    gen = _GeneratorFromCandidatePool(pool)
+    assert [b.name for b in gen.generate_candidates(None, 2, [], [])] == [
+        "lin2"
+    ]
+
+
+def test_autoensemble_lifecycle(tmp_path):
+    X, Y = _data()
+    input_fn = _input_fn_factory(X, Y)
+    head = MultiClassHead(3)
+    est = AutoEnsembleEstimator(
+        head=head,
+        candidate_pool={
+            "linear": LinearEstimator(head),
+            "dnn": DNNEstimator(head, hidden_units=[16]),
+        },
+        max_iteration_steps=8,
+        model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=11),
+    )
+    est.train(input_fn, max_steps=16)
+    assert est.iteration_number == 2
+    res = est.evaluate(input_fn, steps=4)
+    assert "accuracy" in res
+    arch = json.loads(res["architecture/adanet/ensembles"])
+    assert arch["subnetworks"][0]["builder_name"] in ("linear", "dnn")
+
+
+def test_bagging_private_input_fn(tmp_path):
+    """AutoEnsembleSubestimator.train_input_fn: the candidate trains on its
+    own batches (reference common.py:43-56 _SecondaryTrainOpRunnerHook)."""
+    X, Y = _data()
+    shared_input = _input_fn_factory(X, Y, seed=1)
+    private_pulls = []
+
+    def private_input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(99)
+            while True:
+                idx = torch.randint(0, X.shape[0], (16,), generator=g)
+                private_pulls.append(1)
+                yield X[idx], Y[idx]
+
+        return gen()
+
+    head = MultiClassHead(3)
+    est = AutoEnsembleEstimator(
+        head=head,
+        candidate_pool={
+            "bagged": AutoEnsembleSubestimator(
+                LinearEstimator(head), train_input_fn=private_input_fn),
+            "plain": LinearEstimator(head),
+        },
+        max_iteration_steps=5,
+        model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=2),
+    )
+    est.train(shared_input, max_steps=5)
+    assert len(private_pulls) == 5  # one private batch per step
+
+
+def test_prediction_only_candidate_does_not_train(tmp_path):
+    X, Y = _data()
+    input_fn = _input_fn_factory(X, Y)
+    head = MultiClassHead(3)
+    frozen_lin = LinearEstimator(head, seed=5)
+    est = AutoEnsembleEstimator(
+        head=head,
+        candidate_pool={
+            "fixed": AutoEnsembleSubestimator(frozen_lin,
+                                              prediction_only=True),
+            "train": DNNEstimator(head, hidden_units=[8]),
+        },
+        max_iteration_steps=5,
+        model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=3),
+    )
+    it = est._get_or_build_iteration(input_fn)
+    fixed_spec = [s for s in it.subnetwork_specs if "fixed" in s.name][0]
+    before = {k: v.clone() for k, v in
+              fixed_spec.subnetwork.module.state_dict().items()}
+    est.train(input_fn, max_steps=5)
+    after = fixed_spec.subnetwork.module.state_dict()
+    for k in before:
+        assert torch.equal(before[k], after[k]), k
