@@ -983,11 +983,25 @@ class Estimator(object):
         if not comm.is_chief():
             comm.barrier() if comm.is_initialized() else None
             return
+        # Frozen member weights are immutable once frozen: write each ONCE
+        # to <model_dir>/frozen/<key>.pt and reference by path (the
+        # monolithic checkpoint grew O(ensemble^2) bytes over a run).
+        frozen_dir = os.path.join(self._model_dir, "frozen")
+        os.makedirs(frozen_dir, exist_ok=True)
+        frozen_refs = {}
+        for key, sd in self._frozen_states.items():
+            fname = key.replace("|", "_") + ".pt"
+            path = os.path.join(frozen_dir, fname)
+            if not os.path.exists(path) and sd is not None:
+                tmp_f = path + ".tmp"
+                torch.save(sd, tmp_f)
+                os.replace(tmp_f, path)
+            frozen_refs[key] = fname
         payload = {
             self._Keys.CURRENT_ITERATION: self._iteration_number,
             "global_step": self._global_step,
             "architectures": dict(self._architectures),
-            "frozen_states": self._frozen_states,
+            "frozen_refs": frozen_refs,
             "ensemble_state": self._best_ensemble_state,
             "replay_indices": self._replay_indices,
             "iteration_state": None,
@@ -1077,12 +1091,26 @@ class Estimator(object):
         self._architectures = {
             int(k): v for k, v in payload["architectures"].items()
         }
-        self._frozen_states = payload["frozen_states"]
+        self._frozen_states = self._load_frozen_states(payload)
         self._best_ensemble_state = payload["ensemble_state"]
         self._replay_indices = list(payload.get("replay_indices", []))
         self._pending_iteration_state = payload.get("iteration_state")
         log.info("Restored checkpoint %s (iteration %s, global step %s)",
                  path, self._iteration_number, self._global_step)
+
+    def _load_frozen_states(self, payload):
+        if "frozen_states" in payload:  # legacy monolithic layout
+            return payload["frozen_states"]
+        out = {}
+        frozen_dir = os.path.join(self._model_dir, "frozen")
+        for key, fname in payload.get("frozen_refs", {}).items():
+            fpath = os.path.join(frozen_dir, fname)
+            if os.path.exists(fpath):
+                out[key] = torch.load(fpath, map_location="cpu",
+                                      weights_only=False)
+            else:
+                log.warning("Missing frozen member file %s", fpath)
+        return out
 
     # ------------------------------------------------------------------
     # serving helpers
@@ -1098,7 +1126,7 @@ class Estimator(object):
             self._architectures = {
                 int(k): v for k, v in payload["architectures"].items()
             }
-            self._frozen_states = payload["frozen_states"]
+            self._frozen_states = self._load_frozen_states(payload)
             self._best_ensemble_state = payload["ensemble_state"]
             self._replay_indices = list(payload.get("replay_indices", []))
         t = self._iteration_number
