@@ -310,6 +310,10 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     case 13: LAUNCH_SK(128, 128, 4, 4, 4, 32, 4); break;
     case 14: LAUNCH_SK(128, 128, 4, 4, 4, 32, 2); break;
     case 15: LAUNCH_SK(64, 64, 2, 2, 6, 32, 2); break;
+    case 16: LAUNCH_V(64, 64, 2, 2, 6, 64); break;
+    case 17: LAUNCH_V(64, 64, 2, 2, 4, 64); break;
+    case 18: LAUNCH_V(64, 128, 2, 4, 4, 64); break;
+    case 19: LAUNCH_V(64, 128, 2, 4, 2, 64); break;
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V

@@ -27,6 +27,10 @@ VARIANTS = {
     13: "128x128 mw4 splitk4",
     14: "128x128 mw4 splitk2",
     15: "64x64 mw6 splitk2",
+    16: "64x64 f2x2 mw6 k64",
+    17: "64x64 f2x2 mw4 k64",
+    18: "64x128 f2x4 mw4 k64",
+    19: "64x128 f2x4 mw2 k64",
 }
 
 SHAPES = [(2048, 2048, 3072), (4096, 2048, 3072), (2048, 3072, 2048),

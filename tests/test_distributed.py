@@ -84,8 +84,12 @@ def _worker(rank, world, port, model_dir, placement_name, q):
         q.put((rank, traceback.format_exc(), None, None, None, None))
 
 
-@pytest.mark.parametrize("placement_name", ["replication", "round_robin"])
-def test_two_rank_agreement(tmp_path, placement_name):
+@pytest.mark.parametrize("world,placement_name", [
+    (2, "replication"),
+    (2, "round_robin"),
+    (3, "round_robin"),  # more ranks than per-iteration candidates at t=0
+])
+def test_multi_rank_agreement(tmp_path, world, placement_name):
     model_dir = str(tmp_path / "model")
     os.makedirs(model_dir, exist_ok=True)
     port = _free_port()
@@ -93,13 +97,13 @@ def test_two_rank_agreement(tmp_path, placement_name):
     q = ctx.SimpleQueue()
     procs = [
         ctx.Process(target=_worker,
-                    args=(r, 2, port, model_dir, placement_name, q))
-        for r in range(2)
+                    args=(r, world, port, model_dir, placement_name, q))
+        for r in range(world)
     ]
     for p in procs:
         p.start()
     results = {}
-    for _ in range(2):
+    for _ in range(world):
         rank, err, archs, state_sum, frozen_sum, it = _get(q)
         assert err is None, "rank %s failed:\n%s" % (rank, err)
         results[rank] = (archs, state_sum, frozen_sum, it)
@@ -107,11 +111,12 @@ def test_two_rank_agreement(tmp_path, placement_name):
         p.join(timeout=60)
         assert p.exitcode == 0
     a0, s0, f0, it0 = results[0]
-    a1, s1, f1, it1 = results[1]
-    assert it0 == it1 == 2
-    assert a0 == a1, "architectures diverged between ranks"
-    assert s0 == pytest.approx(s1), "ensemble weights diverged"
-    assert f0 == pytest.approx(f1), "frozen member weights diverged"
+    for r in range(1, world):
+        ar, sr, fr, itr = results[r]
+        assert it0 == itr == 2
+        assert a0 == ar, "architectures diverged between ranks"
+        assert s0 == pytest.approx(sr), "ensemble weights diverged"
+        assert f0 == pytest.approx(fr), "frozen member weights diverged"
     # architecture files exist (written by chief only)
     assert os.path.exists(os.path.join(model_dir, "architecture-0.json"))
     assert os.path.exists(os.path.join(model_dir, "architecture-1.json"))
