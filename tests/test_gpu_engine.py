@@ -89,3 +89,61 @@ def test_gpu_autoensemble(tmp_path):
     assert est.iteration_number == 2
     res = est.evaluate(input_fn, steps=5)
     assert "accuracy" in res
+
+
+def test_gpu_improve_nas_conv_search(tmp_path):
+    """NASNet conv candidates on MI355X: bf16 convs (MIOpen), BN fp32
+    stats, our classifier/loss/optimizer kernels, KD hook."""
+    import adanet_amd
+    from adanet_amd.head import MultiClassHead
+    from adanet_amd.models import improve_nas
+    from adanet_amd.models.cifar import FakeImageProvider
+
+    hp = improve_nas.Hparams(num_cells=3, num_conv_filters=8,
+                             train_steps=6, drop_path_keep=1.0)
+    provider = FakeImageProvider(n_examples=128, batch_size=32, seed=3,
+                                 augment=False)
+    input_fn = provider.get_input_fn()
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(10, label_smoothing=hp.label_smoothing),
+        subnetwork_generator=improve_nas.DynamicGenerator(hp, seed=0),
+        max_iteration_steps=6,
+        force_grow=True,
+        model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=1),
+    )
+    est.train(input_fn, max_steps=12)
+    assert est.iteration_number == 2
+    res = est.evaluate(input_fn, steps=2)
+    assert res["loss"] == res["loss"]  # finite
+
+
+def test_gpu_modelflow(tmp_path):
+    from adanet_amd.experimental import (AutoEnsemblePhase, GrowStrategy,
+                                         InputPhase, MeanEnsemble,
+                                         ModelSearch, SequentialController,
+                                         TrainerPhase)
+    from adanet_amd.ops.linear import HipLinear
+    from torch import nn
+
+    torch.manual_seed(0)
+    W = torch.randn(16, 3)
+    train = []
+    g = torch.Generator().manual_seed(0)
+    for _ in range(4):
+        x = torch.randn(64, 16, generator=g)
+        train.append((x, (x @ W).argmax(1)))
+    phases = [
+        InputPhase(train, train),
+        TrainerPhase([
+            nn.Sequential(HipLinear(16, 32, activation="relu"),
+                          HipLinear(32, 3)) for _ in range(2)
+        ], epochs=2),
+        AutoEnsemblePhase(ensemblers=[MeanEnsemble],
+                          ensemble_strategies=[GrowStrategy()],
+                          num_candidates=2),
+    ]
+    search = ModelSearch(SequentialController(phases))
+    search.run()
+    best = search.get_best_models(1)[0]
+    assert isinstance(best.module, MeanEnsemble)
