@@ -53,6 +53,15 @@ def restore_fp32_params(module) -> None:
         if isinstance(m, HipLayerNorm) and m.weight is not None:
             m.weight.data = m.weight.data.float()
             m.bias.data = m.bias.data.float()
+        if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+            # BN statistics/affine stay fp32 under bf16 activations
+            # (models/nasnet.py BNfp32 computes in fp32).
+            if m.weight is not None:
+                m.weight.data = m.weight.data.float()
+                m.bias.data = m.bias.data.float()
+            if m.running_mean is not None:
+                m.running_mean = m.running_mean.float()
+                m.running_var = m.running_var.float()
 
 
 def gemm_nt(a: torch.Tensor, b: torch.Tensor,
