@@ -729,6 +729,23 @@ class Estimator(object):
         arch = _Architecture(cand.name, ensembler.name)
         members = []
         prev_handles = list(cand.previous_ensemble_subnetwork_builders or ())
+        # Legacy Builder.prune_previous_ensemble (reference honors it at
+        # ensemble_builder.py:371-395): a lone new builder may drop previous
+        # members from ITS candidate.
+        new_builders_for_prune = list(cand.subnetwork_builders)
+        if prev_handles and len(new_builders_for_prune) == 1 and (
+                prev_ensemble is not None):
+            b0 = new_builders_for_prune[0]
+            try:
+                keep = b0.prune_previous_ensemble(prev_ensemble)
+            except TypeError:
+                keep = None
+            if keep is not None:
+                keep = set(int(i) for i in keep)
+                if keep != set(range(len(prev_handles))):
+                    prev_handles = [
+                        h for i, h in enumerate(prev_handles) if i in keep
+                    ]
         prev_names = [h.name for h in prev_handles]
         if prev_ensemble is not None:
             prev_by_name = {s.name: s for s in prev_ensemble.subnetworks}

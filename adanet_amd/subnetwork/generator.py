@@ -129,7 +129,7 @@ class Builder(abc.ABC):
     # ensemble_builder.py:371-395.
     def prune_previous_ensemble(self, previous_ensemble) -> List[int]:
         """Returns indices of previous-ensemble subnetworks to keep."""
-        return list(range(len(previous_ensemble.weighted_subnetworks))) if (
+        return list(range(len(previous_ensemble.subnetworks))) if (
             previous_ensemble is not None) else []
 
 

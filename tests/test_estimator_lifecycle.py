@@ -383,3 +383,62 @@ def test_frozen_logit_cache_correctness(model_dir, synthetic_classification):
     assert r1["loss"] == pytest.approx(r2["loss"], abs=1e-4)
     assert (r1["architecture/adanet/ensembles"] ==
             r2["architecture/adanet/ensembles"])
+
+
+def test_builder_prune_previous_ensemble(model_dir,
+                                         synthetic_classification):
+    """Legacy pruning hook (reference ensemble_builder.py:371-395): a
+    builder that keeps no previous members yields a 1-member architecture
+    even at iteration 1."""
+    X, Y, input_fn = synthetic_classification
+
+    class _PruningBuilder(_GoodBuilder):
+
+        def prune_previous_ensemble(self, previous_ensemble):
+            return []  # drop everything
+
+    est = _make_estimator(
+        model_dir, input_fn, force_grow=True,
+        subnetwork_generator=SimpleGenerator([_PruningBuilder("pruner")]))
+    est.train(input_fn, max_steps=20)
+    arch = json.loads(
+        open(os.path.join(model_dir, "architecture-1.json")).read())
+    assert len(arch["subnetworks"]) == 1
+    assert arch["subnetworks"][0]["iteration_number"] == 1
+
+
+@pytest.mark.parametrize("mixture_type", ["scalar", "vector", "matrix"])
+def test_mixture_types_end_to_end(model_dir, synthetic_classification,
+                                  mixture_type):
+    """Full lifecycle per MixtureWeightType incl. mixture-weight training
+    (reference weighted_test.py parameter grid :85-483)."""
+    import functools
+
+    from adanet_amd.ensemble import ComplexityRegularizedEnsembler
+    from adanet_amd.ops.optim import FusedSGD
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        ensemblers=[
+            ComplexityRegularizedEnsembler(
+                optimizer=functools.partial(FusedSGD, lr=0.05),
+                mixture_weight_type=mixture_type,
+                warm_start_mixture_weights=True,
+                model_dir=model_dir,
+                adanet_lambda=1e-4,
+                use_bias=True)
+        ])
+    est.train(input_fn, max_steps=20)
+    assert est.iteration_number == 2
+    res = est.evaluate(input_fn, steps=4)
+    assert res["loss"] == res["loss"]
+
+
+def test_mean_ensembler_end_to_end(model_dir, synthetic_classification):
+    from adanet_amd.ensemble import MeanEnsembler
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn,
+                          ensemblers=[MeanEnsembler()])
+    est.train(input_fn, max_steps=20)
+    res = est.evaluate(input_fn, steps=4)
+    assert "accuracy" in res
