@@ -442,3 +442,53 @@ def test_mean_ensembler_end_to_end(model_dir, synthetic_classification):
     est.train(input_fn, max_steps=20)
     res = est.evaluate(input_fn, steps=4)
     assert "accuracy" in res
+
+
+def test_multiple_ensemblers_and_strategies(model_dir,
+                                            synthetic_classification):
+    """candidates x ensemblers grid (reference iteration.py:683-740) with
+    Solo+Grow strategies and ComplexityRegularized+Mean ensemblers."""
+    from adanet_amd.ensemble import (AllStrategy, ComplexityRegularizedEnsembler,
+                                     MeanEnsembler, SoloStrategy)
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        ensemblers=[ComplexityRegularizedEnsembler(), MeanEnsembler()],
+        ensemble_strategies=[SoloStrategy(), AllStrategy()])
+    est.train(input_fn, max_steps=10)
+    it = est._current_iteration
+    # strategies: solo -> 2 candidates, all -> 1; x2 ensemblers = 6 specs
+    # (iteration 0: no previous-best candidate).
+    assert est.iteration_number == 1
+    arch = json.loads(
+        open(os.path.join(model_dir, "architecture-0.json")).read())
+    assert arch["ensembler_name"] in ("complexity_regularized", "mean")
+
+
+def test_duplicate_ensembler_names_raise(model_dir):
+    from adanet_amd.ensemble import MeanEnsembler
+    with pytest.raises(ValueError):
+        _make_estimator(model_dir, None,
+                        ensemblers=[MeanEnsembler(), MeanEnsembler()])
+
+
+def test_regression_head_end_to_end(model_dir):
+    from adanet_amd.head import RegressionHead
+    torch.manual_seed(0)
+    N, D = 256, 8
+    X = torch.randn(N, D)
+    Y = X @ torch.randn(D)
+
+    def input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(5)
+            while True:
+                idx = torch.randint(0, N, (64,), generator=g)
+                yield X[idx], Y[idx]
+
+        return gen()
+
+    est = _make_estimator(model_dir, input_fn, head=RegressionHead())
+    est.train(input_fn, max_steps=20)
+    res = est.evaluate(input_fn, steps=4)
+    assert res["average_loss"] == res["average_loss"]
