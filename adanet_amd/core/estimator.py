@@ -997,6 +997,21 @@ class Estimator(object):
         return cands[-1] if cands else None
 
     def _save_checkpoint(self, mid_iteration: bool):
+        # Mid-iteration state is gathered from ALL ranks first (round-robin
+        # placement: each rank owns different specs; a chief-only capture
+        # would reset non-owned candidates on restart). Collective, so every
+        # rank must reach this call in the same order.
+        iteration_state = None
+        if mid_iteration and self._current_iteration is not None:
+            iteration_state = self._capture_iteration_state()
+            if comm.is_initialized() and not self._placement.data_parallel:
+                gathered = comm.all_gather_objects(iteration_state)
+                merged = gathered[0]
+                for other in gathered[1:]:
+                    merged["subnetworks"].update(other["subnetworks"])
+                    merged["ensembles"].update(other["ensembles"])
+                    merged["emas"].update(other["emas"])
+                iteration_state = merged
         if not comm.is_chief():
             comm.barrier() if comm.is_initialized() else None
             return
@@ -1021,10 +1036,8 @@ class Estimator(object):
             "frozen_refs": frozen_refs,
             "ensemble_state": self._best_ensemble_state,
             "replay_indices": self._replay_indices,
-            "iteration_state": None,
+            "iteration_state": iteration_state,
         }
-        if mid_iteration and self._current_iteration is not None:
-            payload["iteration_state"] = self._capture_iteration_state()
         path = os.path.join(
             self._model_dir,
             "increment.ckpt-{}.pt".format(self._iteration_number))

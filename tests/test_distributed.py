@@ -24,7 +24,8 @@ def _free_port():
     return port
 
 
-def _worker(rank, world, port, model_dir, placement_name, q):
+def _worker(rank, world, port, model_dir, placement_name, q,
+            resume_mid_iteration=False):
     os.environ["RANK"] = str(rank)
     os.environ["WORLD_SIZE"] = str(world)
     os.environ["LOCAL_RANK"] = str(rank)
@@ -52,18 +53,32 @@ def _worker(rank, world, port, model_dir, placement_name, q):
 
             return gen()
 
-        placement = (ReplicationStrategy() if placement_name == "replication"
-                     else RoundRobinStrategy())
-        est = adanet_amd.Estimator(
-            head=MultiClassHead(C),
-            subnetwork_generator=simple_dnn.Generator(layer_size=8),
-            max_iteration_steps=6,
-            model_dir=model_dir,
-            config=adanet_amd.RunConfig(tf_random_seed=42),
-            experimental_placement_strategy=placement,
-            use_streams=False,
-        )
-        est.train(input_fn, max_steps=12)  # 2 iterations
+        def make_est():
+            placement = (ReplicationStrategy()
+                         if placement_name == "replication" else
+                         RoundRobinStrategy())
+            return adanet_amd.Estimator(
+                head=MultiClassHead(C),
+                subnetwork_generator=simple_dnn.Generator(layer_size=8),
+                max_iteration_steps=6,
+                model_dir=model_dir,
+                config=adanet_amd.RunConfig(tf_random_seed=42),
+                experimental_placement_strategy=placement,
+                use_streams=False,
+            )
+
+        est = make_est()
+        if resume_mid_iteration:
+            # Stop mid-iteration 1, then resume with a FRESH estimator
+            # (per-rank spec state must round-trip through the all-rank
+            # checkpoint gather).
+            est.train(input_fn, steps=9)
+            assert est.iteration_number == 1
+            est = make_est()
+            assert est.global_step == 9
+            est.train(input_fn, steps=3)
+        else:
+            est.train(input_fn, max_steps=12)  # 2 iterations
         # Every rank must agree on the winning architectures.
         archs = {t: est._architectures[t] for t in sorted(est._architectures)}
         # Replication: winner weights must be identical across ranks.
@@ -84,12 +99,13 @@ def _worker(rank, world, port, model_dir, placement_name, q):
         q.put((rank, traceback.format_exc(), None, None, None, None))
 
 
-@pytest.mark.parametrize("world,placement_name", [
-    (2, "replication"),
-    (2, "round_robin"),
-    (3, "round_robin"),  # more ranks than per-iteration candidates at t=0
+@pytest.mark.parametrize("world,placement_name,resume", [
+    (2, "replication", False),
+    (2, "round_robin", False),
+    (3, "round_robin", False),  # more ranks than candidates at t=0
+    (2, "round_robin", True),   # mid-iteration checkpoint + resume
 ])
-def test_multi_rank_agreement(tmp_path, world, placement_name):
+def test_multi_rank_agreement(tmp_path, world, placement_name, resume):
     model_dir = str(tmp_path / "model")
     os.makedirs(model_dir, exist_ok=True)
     port = _free_port()
@@ -97,7 +113,8 @@ def test_multi_rank_agreement(tmp_path, world, placement_name):
     q = ctx.SimpleQueue()
     procs = [
         ctx.Process(target=_worker,
-                    args=(r, world, port, model_dir, placement_name, q))
+                    args=(r, world, port, model_dir, placement_name, q,
+                          resume))
         for r in range(world)
     ]
     for p in procs:
