@@ -52,6 +52,7 @@ static int ew_grid(int64_t n) {
 
 void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed) {
   const int64_t n = x.numel();
+  if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(dropout_fwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)x.data_ptr(),
@@ -61,6 +62,7 @@ void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed) {
 
 void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed) {
   const int64_t n = dy.numel();
+  if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(dropout_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)dy.data_ptr(),
@@ -70,6 +72,7 @@ void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed) {
 
 void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx) {
   const int64_t n = dy.numel();
+  if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(relu_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)dy.data_ptr(),

@@ -207,6 +207,8 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   TORCH_CHECK(C.size(0) == M && C.size(1) == N, "gemm: C shape mismatch");
   const int lda = (int)A.stride(0), ldb = (int)B.stride(0),
             ldc = (int)C.stride(0);
+  if (M == 0 || N == 0) return;
+  if (K == 0) { C.zero_(); return; }
   const float* bias_ptr = nullptr;
   if (bias.has_value() && bias->defined()) {
     TORCH_CHECK(bias->scalar_type() == at::kFloat && bias->numel() == N,

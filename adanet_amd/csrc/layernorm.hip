@@ -81,6 +81,7 @@ void layernorm_fwd(const at::Tensor& x, const c10::optional<at::Tensor>& gamma,
                    const c10::optional<at::Tensor>& beta, at::Tensor& y,
                    at::Tensor& mean, at::Tensor& rstd, double eps) {
   const int B = (int)x.size(0), D = (int)x.size(1);
+  if (B == 0 || D == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   const float* g = (gamma && gamma->defined()) ? gamma->data_ptr<float>() : nullptr;
   const float* b = (beta && beta->defined()) ? beta->data_ptr<float>() : nullptr;
@@ -98,6 +99,7 @@ void layernorm_bwd(const at::Tensor& x, const at::Tensor& dy,
                    at::Tensor& dx, const c10::optional<at::Tensor>& dgamma,
                    const c10::optional<at::Tensor>& dbeta) {
   const int B = (int)x.size(0), D = (int)x.size(1);
+  if (B == 0 || D == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   const float* g = (gamma && gamma->defined()) ? gamma->data_ptr<float>() : nullptr;
   float* dg = (dgamma && dgamma->defined()) ? dgamma->data_ptr<float>() : nullptr;

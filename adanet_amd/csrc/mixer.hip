@@ -93,6 +93,7 @@ void mixer_fwd(const at::Tensor& stack, const at::Tensor& weights,
               "mixer: stacked [J,B,C] contiguous bf16 required");
   const int J = (int)stack.size(0), B = (int)stack.size(1),
             C = (int)stack.size(2);
+  if (J == 0 || (int64_t)B * C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   const float* bias_ptr =
       (bias.has_value() && bias->defined()) ? bias->data_ptr<float>() : nullptr;

@@ -141,6 +141,7 @@ void fused_sgd(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,
                double momentum, double dampening, double weight_decay,
                bool nesterov, double grad_scale) {
   const int64_t n = master.numel();
+  if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   float* mom = (momentum_buf.has_value() && momentum_buf->defined())
                    ? momentum_buf->data_ptr<float>()

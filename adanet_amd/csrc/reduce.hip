@@ -53,6 +53,8 @@ __global__ __launch_bounds__(256) void argmax_correct_kernel(
 
 void colsum_bf16(const at::Tensor& x, at::Tensor& out) {
   const int B = (int)x.size(0), C = (int)x.size(1);
+  if (C == 0) return;
+  if (B == 0) { out.zero_(); return; }
   auto stream = at::cuda::getCurrentCUDAStream();
   const int stripes = (C + 255) / 256;
   // Fill the chip: aim for ~1024 blocks, at least 8 rows per chunk.
@@ -73,6 +75,7 @@ void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred,
                     at::Tensor& correct) {
   const int B = (int)logits.size(0), C = (int)logits.size(1);
+  if (B == 0 || C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   int64_t* pred_ptr =
       (pred && pred->defined()) ? pred->data_ptr<int64_t>() : nullptr;

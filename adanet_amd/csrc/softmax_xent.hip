@@ -98,6 +98,7 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   TORCH_CHECK(labels.scalar_type() == at::kLong, "xent: int64 labels");
   TORCH_CHECK(logits.stride(1) == 1, "xent: contiguous class dim");
   const int B = (int)logits.size(0), C = (int)logits.size(1);
+  if (B == 0 || C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   const bool fused_mean = mean_out.has_value() && mean_out->defined();
   // With the fused mean, cap blocks so the final atomics stay few; each
@@ -122,6 +123,7 @@ void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
                       at::Tensor& dlogits, double eps,
                       const c10::optional<at::Tensor>& grad_scalar) {
   const int B = (int)probs.size(0), C = (int)probs.size(1);
+  if (B == 0 || C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   const int blocks = std::min((B + 3) / 4, 2048);
   const float* gr = (grad_rows.has_value() && grad_rows->defined())

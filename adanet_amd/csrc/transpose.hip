@@ -80,6 +80,7 @@ void transpose_bf16(const at::Tensor& in, at::Tensor& out) {
   TORCH_CHECK(in.stride(1) == 1 && out.stride(1) == 1, "transpose: row-major");
   const int M = (int)in.size(0), N = (int)in.size(1);
   TORCH_CHECK(out.size(0) == N && out.size(1) == M, "transpose: shape");
+  if (M == 0 || N == 0) return;
   const int mtiles = (M + TDIM - 1) / TDIM, ntiles = (N + TDIM - 1) / TDIM;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(transpose_bf16_kernel, dim3(mtiles * ntiles), dim3(256),

@@ -1,0 +1,151 @@
+"""Iteration-engine unit tests with fake specs (reference model:
+adanet/core/iteration_test.py:199-362 _FakeBuilder/_FakeEnsembler layer
+cutouts)."""
+
+import math
+import os
+
+import pytest
+import torch
+from torch import nn
+
+from adanet_amd.core.architecture import _Architecture
+from adanet_amd.core.candidate import _Candidate
+from adanet_amd.core.iteration import (_EnsembleSpec, _Iteration,
+                                       _SubnetworkSpec, _TrainManager)
+from adanet_amd.ensemble.weighted import (ComplexityRegularizedEnsembler)
+from adanet_amd.head import MultiClassHead
+from adanet_amd.ops.optim import FusedSGD
+from adanet_amd.subnetwork import Builder, Subnetwork
+
+
+class _FakeModule(nn.Module):
+
+    def __init__(self, d=6, c=3):
+        super().__init__()
+        self.lin = nn.Linear(d, c)
+        self.last_layer_dim = d
+
+    def forward(self, x):
+        return x, self.lin(x)
+
+
+class _FakeBuilder(Builder):
+
+    def __init__(self, name="fake"):
+        self._name = name
+
+    @property
+    def name(self):
+        return self._name
+
+    def build_subnetwork(self, features, logits_dimension, training,
+                         previous_ensemble=None):
+        return Subnetwork(module=_FakeModule(features.shape[1],
+                                             logits_dimension),
+                          complexity=1.0, name=self._name)
+
+
+def _make_iteration(tmp_path, n_builders=2, max_steps=5):
+    torch.manual_seed(0)
+    head = MultiClassHead(3)
+    x = torch.randn(8, 6)
+    sub_specs, ens_specs = [], []
+    ensembler = ComplexityRegularizedEnsembler()
+    for i in range(n_builders):
+        b = _FakeBuilder("b%d" % i)
+        sub = b.build_subnetwork(x, 3, True)
+        opt = FusedSGD(sub.module.parameters(), lr=0.1)
+        sub_specs.append(
+            _SubnetworkSpec(name="t0_b%d" % i, builder=b, subnetwork=sub,
+                            optimizer=opt))
+        ens = ensembler.build_ensemble([sub], None, x, None, 3, True, None)
+        ens_specs.append(
+            _EnsembleSpec(name="t0_b%d_grow" % i, candidate=None,
+                          ensemble=ens, ensembler_name=ensembler.name,
+                          architecture=_Architecture("b%d" % i,
+                                                     ensembler.name),
+                          optimizer=None, members=(("new", "b%d" % i),)))
+    tm = _TrainManager(str(tmp_path), 0)
+    it = _Iteration(number=0, head=head, subnetwork_specs=sub_specs,
+                    ensemble_specs=ens_specs, frozen_subnetworks={},
+                    train_manager=tm, max_iteration_steps=max_steps,
+                    adanet_loss_decay=0.5, device=torch.device("cpu"),
+                    placement=None, use_streams=False, use_graphs=False)
+    return it, x
+
+
+def test_train_step_updates_emas_and_steps(tmp_path):
+    it, x = _make_iteration(tmp_path)
+    y = torch.randint(0, 3, (8,))
+    for _ in range(3):
+        it.train_step(x, y)
+    it.flush_losses()
+    assert it.step == 3
+    for spec in it.subnetwork_specs:
+        assert spec.step == 3
+        assert not math.isnan(spec.last_loss)
+    for cand in it.candidates:
+        assert cand.adanet_loss < float("inf")
+
+
+def test_iteration_stops_at_max_steps(tmp_path):
+    it, x = _make_iteration(tmp_path, max_steps=2)
+    y = torch.randint(0, 3, (8,))
+    it.train_step(x, y)
+    assert not it.is_over()
+    it.train_step(x, y)
+    assert it.is_over()
+    # further steps are no-ops for subnetwork training
+    steps_before = [s.step for s in it.subnetwork_specs]
+    it.train_step(x, y)
+    assert [s.step for s in it.subnetwork_specs] == steps_before
+
+
+def test_best_candidate_nan_to_neg_inf(tmp_path):
+    it, x = _make_iteration(tmp_path, n_builders=3)
+    it.candidates[0].update(0.5)
+    it.candidates[1].update(float("nan"))
+    it.candidates[2].update(0.1)
+    # NaN -> -inf wins argmin (reference iteration.py:1040-1046)
+    assert it.best_candidate_index(losses=[0.5, float("nan"), 0.1]) == 1
+    assert it.best_candidate_index(losses=[0.5, 0.3, 0.1]) == 2
+    assert it.best_candidate_index(override=0) == 0
+
+
+def test_train_manager_persistence(tmp_path):
+    tm = _TrainManager(str(tmp_path), 3)
+    assert tm.should_train("a")
+    tm.request_stop("a", "done")
+    assert not tm.should_train("a")
+    assert not tm.is_over(["a", "b"])
+    tm.request_stop("b", "done")
+    assert tm.is_over(["a", "b"])
+    # persisted on disk, picked up by a fresh manager (restart semantics)
+    tm2 = _TrainManager(str(tmp_path), 3)
+    assert not tm2.should_train("a")
+    assert os.path.exists(
+        os.path.join(str(tmp_path), "train_manager", "t3", "a.json"))
+
+
+def test_loss_ring_buffer_flush_cadence(tmp_path):
+    it, x = _make_iteration(tmp_path, max_steps=100)
+    y = torch.randint(0, 3, (8,))
+    from adanet_amd.core import iteration as it_mod
+    for i in range(it_mod._LOSS_FLUSH_STEPS + 2):
+        it.train_step(x, y)
+    # flush happened automatically at the cadence boundary
+    assert it._loss_buf_rows == 2
+    it.flush_losses()
+    assert it._loss_buf_rows == 0
+
+
+def test_evaluate_candidates_shared_batches(tmp_path):
+    it, x = _make_iteration(tmp_path)
+    y = torch.randint(0, 3, (8,))
+    it.train_step(x, y)
+    data = [(x, y), (x, y)]
+    vals = it.evaluate_candidates(iter(data), steps=2,
+                                  to_device=lambda f, l: (f, l))
+    assert len(vals) == 2
+    assert all(v == v for v in vals)  # finite
