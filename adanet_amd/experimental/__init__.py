@@ -11,7 +11,9 @@ from adanet_amd.experimental.phases import (AllStrategy, AutoEnsemblePhase,
                                             GrowStrategy, InputPhase, Phase,
                                             RandomKStrategy, RepeatPhase,
                                             TrainerPhase, TunerPhase)
-from adanet_amd.experimental.schedulers import InProcessScheduler, Scheduler
+from adanet_amd.experimental.schedulers import (InProcessScheduler,
+                                                Scheduler,
+                                                ThreadedScheduler)
 from adanet_amd.experimental.storages import (InMemoryStorage,
                                               ModelContainer, Storage)
 from adanet_amd.experimental.work_units import TrainerWorkUnit, WorkUnit
@@ -21,6 +23,6 @@ __all__ = [
     "WeightedEnsemble", "CompiledModel", "ModelSearch", "AllStrategy",
     "AutoEnsemblePhase", "GrowStrategy", "InputPhase", "Phase",
     "RandomKStrategy", "RepeatPhase", "TrainerPhase", "TunerPhase",
-    "InProcessScheduler", "Scheduler", "InMemoryStorage", "ModelContainer",
+    "InProcessScheduler", "Scheduler", "ThreadedScheduler", "InMemoryStorage", "ModelContainer",
     "Storage", "TrainerWorkUnit", "WorkUnit",
 ]

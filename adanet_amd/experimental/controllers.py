@@ -40,5 +40,12 @@ class SequentialController(Controller):
             for wu in phase.work_units():
                 yield wu
 
+    def phased_work_units(self):
+        """Yields (work_units, parallel_ok) per phase: a phase's units are
+        mutually independent unless the phase says otherwise (RepeatPhase
+        chains repetitions); phases are barriers."""
+        for phase in self._phases:
+            yield phase.work_units(), getattr(phase, "parallel_ok", True)
+
     def get_best_models(self, num_models: int = 1):
         return self._phases[-1].get_best_models(num_models)

@@ -19,6 +19,11 @@ class ModelSearch(object):
         self._scheduler = scheduler or InProcessScheduler()
 
     def run(self):
+        if (hasattr(self._scheduler, "schedule_phased")
+                and hasattr(self._controller, "phased_work_units")):
+            self._scheduler.schedule_phased(
+                self._controller.phased_work_units())
+            return
         self._scheduler.schedule(self._controller.work_units())
 
     def get_best_models(self, num_models: int = 1) -> List:

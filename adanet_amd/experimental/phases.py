@@ -27,6 +27,10 @@ class Phase(abc.ABC):
     yields WorkUnits; DatasetProvider/ModelProvider mixins collapse to the
     get_train/eval_dataset + get_models accessors."""
 
+    #: a phase's work units are mutually independent (ThreadedScheduler may
+    #: run them concurrently); RepeatPhase overrides.
+    parallel_ok = True
+
     def __init__(self):
         self._previous: Optional[Phase] = None
 
@@ -73,23 +77,32 @@ class InputPhase(Phase):
 
 
 class TrainerPhase(Phase):
-    """Trains a fixed list of models (reference keras_trainer_phase.py:28)."""
+    """Trains a fixed list of models (reference keras_trainer_phase.py:28).
+
+    ``devices``: optional device list round-robined over the models (e.g.
+    ["cuda:0", ..., "cuda:7"]) so a ThreadedScheduler trains candidates on
+    the node's GPUs concurrently."""
 
     def __init__(self, models: Sequence, epochs: int = 1,
                  steps_per_epoch: Optional[int] = None,
                  eval_steps: Optional[int] = None,
-                 storage: Optional[Storage] = None):
+                 storage: Optional[Storage] = None,
+                 devices: Optional[Sequence] = None):
         super().__init__()
         self._models = list(models)
         self._epochs = epochs
         self._steps_per_epoch = steps_per_epoch
         self._eval_steps = eval_steps
         self._storage = storage or InMemoryStorage()
+        self._devices = list(devices) if devices else None
 
     def work_units(self):
-        for model in self._models:
+        import torch
+        for i, model in enumerate(self._models):
             if not isinstance(model, CompiledModel):
-                model = CompiledModel(model)
+                device = (torch.device(self._devices[i % len(self._devices)])
+                          if self._devices else None)
+                model = CompiledModel(model, device=device)
             yield TrainerWorkUnit(model, self.get_train_dataset(),
                                   self.get_eval_dataset(), self._storage,
                                   epochs=self._epochs,
@@ -203,6 +216,8 @@ class AutoEnsemblePhase(Phase):
 class RepeatPhase(Phase):
     """Repeats an inner phase pipeline n times (reference
     repeat_phase.py:26): phase factories are re-invoked per repetition."""
+
+    parallel_ok = False  # repetitions chain on each other
 
     def __init__(self, phase_factories: Sequence[Callable[[], Phase]],
                  repetitions: int):

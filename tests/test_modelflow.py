@@ -112,3 +112,47 @@ def test_repeat_phase():
     search = ModelSearch(SequentialController(phases))
     search.run()
     assert len(storage.get_models()) == 3
+
+
+def test_threaded_scheduler_phase_barriers():
+    """ThreadedScheduler: phase-parallel units with phase barriers
+    (AutoEnsemblePhase must see TrainerPhase's completed storage)."""
+    from adanet_amd.experimental import ThreadedScheduler
+    train, ev = _dataset(seed=0), _dataset(seed=1)
+    phases = [
+        InputPhase(train, ev),
+        TrainerPhase([_mlp(), _mlp(), _mlp()], epochs=1,
+                     devices=["cpu", "cpu"]),
+        AutoEnsemblePhase(ensemblers=[MeanEnsemble],
+                          ensemble_strategies=[AllStrategy()],
+                          num_candidates=3),
+    ]
+    search = ModelSearch(SequentialController(phases),
+                         scheduler=ThreadedScheduler(max_workers=3))
+    search.run()
+    best = search.get_best_models(1)[0]
+    assert isinstance(best.module, MeanEnsemble)
+    assert len(best.module.submodels) == 3
+
+
+def test_threaded_scheduler_repeat_phase_serial():
+    from adanet_amd.experimental import ThreadedScheduler
+    train, ev = _dataset(seed=0), _dataset(seed=1)
+    order = []
+
+    class _Tracking(TrainerPhase):
+
+        def work_units(self):
+            for wu in super().work_units():
+                order.append("unit")
+                yield wu
+
+    storage = InMemoryStorage()
+    phases = [
+        InputPhase(train, ev),
+        RepeatPhase([functools.partial(_Tracking, [_mlp()], epochs=1,
+                                       storage=storage)], repetitions=3),
+    ]
+    ModelSearch(SequentialController(phases),
+                scheduler=ThreadedScheduler()).run()
+    assert len(storage.get_models()) == 3
