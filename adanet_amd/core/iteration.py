@@ -275,6 +275,9 @@ class _Iteration(object):
                 n: (last.clone(), logits.clone())
                 for n, (last, logits) in fresh.items()
             }
+            # A captured graph reads the OLD static buffers — invalidate it
+            # (batch-shape change mid-iteration).
+            self._graph = None
         else:
             for n, (last, logits) in fresh.items():
                 last_s, logits_s = self._frozen_static[n]
@@ -365,20 +368,34 @@ class _Iteration(object):
         return bool(self._graph_ok)
 
     def _copy_static_inputs(self, features, labels):
-        if self._static_inputs is None:
-            if isinstance(features, dict):
-                sf = {k: v.clone() for k, v in features.items()}
+        if self._static_inputs is not None:
+            # Batch-shape change: rebuild statics and recapture.
+            sf0, sl0 = self._static_inputs
+            f0 = sf0[sorted(sf0)[0]] if isinstance(sf0, dict) else sf0
+            f1 = (features[sorted(features)[0]]
+                  if isinstance(features, dict) else features)
+            if f0.shape != f1.shape or (
+                    not isinstance(labels, dict)
+                    and sl0.shape != labels.shape):
+                self._static_inputs = None
+                self._graph = None
+        def _clone(x):
+            return ({k: v.clone() for k, v in x.items()}
+                    if isinstance(x, dict) else x.clone())
+
+        def _copy(dst, src):
+            if isinstance(dst, dict):
+                for k in dst:
+                    dst[k].copy_(src[k], non_blocking=True)
             else:
-                sf = features.clone()
-            self._static_inputs = (sf, labels.clone())
+                dst.copy_(src, non_blocking=True)
+
+        if self._static_inputs is None:
+            self._static_inputs = (_clone(features), _clone(labels))
             return
         sf, sl = self._static_inputs
-        if isinstance(features, dict):
-            for k, v in features.items():
-                sf[k].copy_(v, non_blocking=True)
-        else:
-            sf.copy_(features, non_blocking=True)
-        sl.copy_(labels, non_blocking=True)
+        _copy(sf, features)
+        _copy(sl, labels)
 
     def _active_signature(self):
         return tuple(sorted(self.train_manager.stopped))
