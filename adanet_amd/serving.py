@@ -15,6 +15,113 @@ from typing import Optional
 import torch
 
 
+def to_portable_module(module: torch.nn.Module) -> torch.nn.Module:
+    """Converts adanet_amd device modules (HipLinear/HipLayerNorm/
+    HipDropout) into pure-torch equivalents with identical weights, so the
+    frozen ensemble can be TorchScript-traced and served WITHOUT
+    adanet_amd or a GPU (the SavedModel-portability analog: reference
+    exports are runnable under stock TF serving)."""
+    import copy
+
+    from torch import nn
+
+    from adanet_amd.ops.dropout import HipDropout
+    from adanet_amd.ops.layernorm import HipLayerNorm
+    from adanet_amd.ops.linear import HipLinear
+
+    seen = []
+
+    def convert(m):
+        seen.append(m)
+        # Member subnetworks hang off `subnetwork` dataclass attributes
+        # (WeightedSubnetwork / MeanEnsemble), invisible to named_children.
+        sub = getattr(m, "subnetwork", None)
+        if sub is not None and isinstance(getattr(sub, "module", None),
+                                          nn.Module):
+            convert(sub.module)
+        subs = getattr(m, "_subnetworks", None)
+        if isinstance(subs, (list, tuple)):
+            for s in subs:
+                if isinstance(getattr(s, "module", None), nn.Module):
+                    convert(s.module)
+        for name, child in list(m.named_children()):
+            if isinstance(child, HipLinear):
+                lin = nn.Linear(child.in_features, child.out_features)
+                with torch.no_grad():
+                    lin.weight.copy_(
+                        child.weight[:child.out_features].float())
+                    if child.bias is not None:
+                        lin.bias.copy_(
+                            child.bias[:child.out_features].float())
+                    else:
+                        lin.bias.zero_()
+                repl = [nn.Flatten(start_dim=1), lin]
+                if child.activation == "relu":
+                    repl.append(nn.ReLU())
+                setattr(m, name, nn.Sequential(*repl))
+            elif isinstance(child, HipLayerNorm):
+                ln = nn.LayerNorm(child.dim, eps=child.eps,
+                                  elementwise_affine=child.weight is not None)
+                with torch.no_grad():
+                    if child.weight is not None:
+                        ln.weight.copy_(child.weight.float())
+                        ln.bias.copy_(child.bias.float())
+                setattr(m, name, ln)
+            elif isinstance(child, HipDropout):
+                setattr(m, name, nn.Dropout(child.p))
+            else:
+                convert(child)
+
+    portable = copy.deepcopy(module).cpu().float()
+    convert(portable)
+    # Dataclass-held member modules are free tensors to torch.jit.trace;
+    # inference artifact -> no grads anywhere.
+    for m in seen:
+        for p in m.parameters():
+            p.requires_grad_(False)
+    portable.eval()
+    return portable
+
+
+class _TracedEnsembleWrapper(torch.nn.Module):
+    """Adapter so tracing sees a plain logits forward."""
+
+    def __init__(self, ensemble):
+        super().__init__()
+        self.ensemble = ensemble
+
+    def forward(self, features):
+        return self.ensemble(features)
+
+
+def export_torchscript(estimator, example_features, path: str):
+    """Traces the frozen best ensemble into a standalone TorchScript file
+    loadable with plain `torch.jit.load` (no adanet_amd dependency).
+    """
+    ensemble, _ = estimator._load_frozen_best()
+    # Materialize the lazy ensemble with the example batch first.
+    with torch.no_grad():
+        ensemble(example_features)
+    live = ensemble
+    if hasattr(live, "_est"):  # ServableEnsemble
+        live = live._est
+    # _load_frozen_best returns the lazy wrapper; dig out the built module.
+    built = None
+    if hasattr(ensemble, "forward"):
+        # rebuild directly for conversion
+        t = estimator._iteration_number
+        built, _ = estimator._rebuild_previous_ensemble(t, example_features)
+    portable = to_portable_module(built)
+    wrapper = _TracedEnsembleWrapper(portable)
+    ex = example_features
+    if torch.is_tensor(ex):
+        ex = ex.detach().cpu().float()
+    with torch.no_grad():
+        traced = torch.jit.trace(wrapper, ex)
+    traced.save(path)
+    return path
+
+
 class ServableEnsemble(torch.nn.Module):
     """Frozen best ensemble + prediction helpers."""
 

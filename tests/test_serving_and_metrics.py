@@ -58,3 +58,27 @@ def test_export_and_serving_roundtrip(tmp_path, synthetic_classification):
     assert torch.allclose(out.float(), ref.float(), atol=1e-4)
     preds = servable.predict(X[:4])
     assert "probabilities" in preds
+
+
+def test_torchscript_export_standalone(tmp_path, synthetic_classification):
+    """Traced frozen ensemble: loadable with plain torch.jit.load, outputs
+    match the live ensemble."""
+    X, Y, input_fn = synthetic_classification
+    gen = simple_dnn.Generator(layer_size=8)
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(4), subnetwork_generator=gen,
+        max_iteration_steps=10, model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=42))
+    est.train(input_fn, max_steps=20)
+    path = str(tmp_path / "ensemble_ts.pt")
+    serving.export_torchscript(est, X[:8], path)
+    loaded = torch.jit.load(path)
+    with torch.no_grad():
+        out = loaded(X[:8].float())
+    live, _ = est._load_frozen_best()
+    with torch.no_grad():
+        ref = live(X[:8])
+    assert out.shape == ref.shape
+    # live path rounds through bf16; the portable artifact is full fp32.
+    assert torch.allclose(out, ref.float(), atol=1e-2), (
+        (out - ref.float()).abs().max())
