@@ -98,19 +98,10 @@ def export_torchscript(estimator, example_features, path: str):
     """Traces the frozen best ensemble into a standalone TorchScript file
     loadable with plain `torch.jit.load` (no adanet_amd dependency).
     """
-    ensemble, _ = estimator._load_frozen_best()
-    # Materialize the lazy ensemble with the example batch first.
-    with torch.no_grad():
-        ensemble(example_features)
-    live = ensemble
-    if hasattr(live, "_est"):  # ServableEnsemble
-        live = live._est
-    # _load_frozen_best returns the lazy wrapper; dig out the built module.
-    built = None
-    if hasattr(ensemble, "forward"):
-        # rebuild directly for conversion
-        t = estimator._iteration_number
-        built, _ = estimator._rebuild_previous_ensemble(t, example_features)
+    t = estimator._iteration_number
+    built, _ = estimator._rebuild_previous_ensemble(t, example_features)
+    if built is None:
+        raise ValueError("No trained ensemble to export — train() first.")
     portable = to_portable_module(built)
     wrapper = _TracedEnsembleWrapper(portable)
     ex = example_features
