@@ -30,7 +30,7 @@ typedef s16x8 frag_ab;
 // Stage a [ROWS x KSTEP] bf16 tile into LDS: each global_load_lds covers
 // 1 KiB (64 lanes x 16 B, LDS-linear = 512/KSTEP rows); waves stride the
 // segment list.
-template <int ROWS, int KSTEP>
+template <int ROWS, int KSTEP, int NWAVES = 4>
 __device__ __forceinline__ void stage_tile_nt(
     const bf16_t* __restrict__ G, int ld, int tile_row0, int max_row, int k0,
     bf16_t* __restrict__ lds, int wid, int lane) {
@@ -38,7 +38,7 @@ __device__ __forceinline__ void stage_tile_nt(
   constexpr int LANES_PER_ROW = KSTEP / 8;       // 16 B loads per row
   constexpr int SEGMENTS = ROWS / ROWS_PER_SEG;
 #pragma unroll
-  for (int seg = wid; seg < SEGMENTS; seg += 4) {
+  for (int seg = wid; seg < SEGMENTS; seg += NWAVES) {
     const int row_in_tile = seg * ROWS_PER_SEG + lane / LANES_PER_ROW;
     int grow = tile_row0 + row_in_tile;
     grow = grow < max_row ? grow : max_row - 1;  // clamp; masked on C-store
@@ -58,8 +58,8 @@ __device__ __forceinline__ void stage_tile_nt(
 // separate epilogue kernel finishes bias+ReLU+bf16 — trades a little
 // output traffic for filling all 256 CUs with the efficient big tile.
 template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32,
-          bool SPLITK = false>
-__global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
+          bool SPLITK = false, int WGM = 2, int WGN = 2>
+__global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles,
@@ -78,10 +78,13 @@ __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
   const int tile_lin = SPLITK ? (wg % tiles) : wg;
   const int tile_m = tile_lin / ntiles, tile_n = tile_lin % ntiles;
 
+  static_assert(BM == WGM * FM * 16 && BN == WGN * FN * 16,
+                "tile must equal wave grid x fragments");
+  constexpr int NWAVES = WGM * WGN;
   const int tid = threadIdx.x;
   const int wid = tid >> 6;
   const int lane = tid & 63;
-  const int wm = wid >> 1, wn = wid & 1;  // 2x2 wave grid
+  const int wm = wid / WGN, wn = wid % WGN;  // WGM x WGN wave grid
 
   const int row0 = tile_m * BM;
   const int col0 = tile_n * BN;
@@ -99,16 +102,20 @@ __global__ __launch_bounds__(THREADS, MINWAVES) void gemm_nt_bf16_kernel(
     ktend = min(ktend, kt0 + per);
     if (kt0 >= ktend) return;
   }
-  stage_tile_nt<BM, KSTEP>(A, lda, row0, M, kt0 * KSTEP, As[0], wid, lane);
-  stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, kt0 * KSTEP, Bs[0], wid, lane);
+  stage_tile_nt<BM, KSTEP, NWAVES>(A, lda, row0, M, kt0 * KSTEP, As[0], wid,
+                                   lane);
+  stage_tile_nt<BN, KSTEP, NWAVES>(B, ldb, col0, N, kt0 * KSTEP, Bs[0], wid,
+                                   lane);
 
   int buf = 0;
   for (int kt = kt0; kt < ktend; ++kt) {
     __syncthreads();  // staged tile `buf` visible; prior reads of buf^1 done
     if (kt + 1 < ktend) {
       const int k0 = (kt + 1) * KSTEP;
-      stage_tile_nt<BM, KSTEP>(A, lda, row0, M, k0, As[buf ^ 1], wid, lane);
-      stage_tile_nt<BN, KSTEP>(B, ldb, col0, N, k0, Bs[buf ^ 1], wid, lane);
+      stage_tile_nt<BM, KSTEP, NWAVES>(A, lda, row0, M, k0, As[buf ^ 1],
+                                       wid, lane);
+      stage_tile_nt<BN, KSTEP, NWAVES>(B, ldb, col0, N, k0, Bs[buf ^ 1],
+                                       wid, lane);
     }
 #pragma unroll
     for (int kk = 0; kk < KSTEP / 32; ++kk) {
@@ -236,6 +243,14 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                        b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, \
                        nt);                                                   \
   } while (0)
+#define LAUNCH_CFG_W(BM, BN, FM, FN, MW, WGM, WGN)                            \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL(                                                       \
+        (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN>),       \
+        dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
+        bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                  \
+  } while (0)
     if (b128 >= 1024) {
       LAUNCH_CFG(128, 128, 4, 4, 4);
     } else if (b128 >= 256 && M > N) {
@@ -313,9 +328,24 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     case 14: LAUNCH_SK(128, 128, 4, 4, 4, 32, 2); break;
     case 15: LAUNCH_SK(64, 64, 2, 2, 6, 32, 2); break;
     case 16: LAUNCH_V(64, 64, 2, 2, 6, 64); break;
+#define LAUNCH_VW(BM, BN, FM, FN, MW, WGM, WGN)                               \
+  do {                                                                        \
+    TORCH_CHECK(K % 32 == 0, "probe: K %% 32");                               \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL(                                                       \
+        (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN>),       \
+        dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
+        nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);                          \
+  } while (0)
     case 17: LAUNCH_V(64, 64, 2, 2, 4, 64); break;
     case 18: LAUNCH_V(64, 128, 2, 4, 4, 64); break;
     case 19: LAUNCH_V(64, 128, 2, 4, 2, 64); break;
+    case 20: LAUNCH_VW(128, 128, 2, 2, 2, 4, 4); break;   // 16 waves
+    case 21: LAUNCH_VW(128, 128, 4, 2, 2, 2, 4); break;   // 8 waves 64x32
+    case 22: LAUNCH_VW(128, 128, 2, 4, 2, 4, 2); break;   // 8 waves 32x64
+    case 23: LAUNCH_VW(256, 128, 4, 4, 2, 4, 2); break;   // 8 waves 64x64
+    case 24: LAUNCH_VW(128, 256, 4, 4, 2, 2, 4); break;
+    case 25: LAUNCH_VW(128, 128, 2, 2, 4, 4, 4); break;   // 16 waves mw4
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V

@@ -31,6 +31,12 @@ VARIANTS = {
     17: "64x64 f2x2 mw4 k64",
     18: "64x128 f2x4 mw4 k64",
     19: "64x128 f2x4 mw2 k64",
+    20: "128x128 16w (4x4 grid, 32x32/wave) mw2",
+    21: "128x128 8w (2x4 grid, 64x32/wave) mw2",
+    22: "128x128 8w (4x2 grid, 32x64/wave) mw2",
+    23: "256x128 8w (4x2 grid, 64x64/wave) mw2",
+    24: "128x256 8w (2x4 grid, 64x64/wave) mw2",
+    25: "128x128 16w mw4",
 }
 
 SHAPES = [(2048, 2048, 3072), (4096, 2048, 3072), (2048, 3072, 2048),
