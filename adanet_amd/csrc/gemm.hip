@@ -251,16 +251,25 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
         dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
         bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                  \
   } while (0)
-    if (b128 >= 1024) {
-      LAUNCH_CFG(128, 128, 4, 4, 4);
-    } else if (b128 >= 256 && M > N) {
-      LAUNCH_CFG(64, 128, 2, 4, 4);
-    } else if (b128 >= 256 && N > M) {
-      LAUNCH_CFG(128, 64, 4, 2, 2);
+    const int64_t t_256x128 =
+        (int64_t)((M + 255) / 256) * ((N + 127) / 128);
+    const int64_t t_128x256 =
+        (int64_t)((M + 127) / 128) * ((N + 255) / 256);
+    if (M >= N && t_256x128 >= 512) {
+      // 256x128, 8 waves (4x2), 64x64/wave: big-tile L2/LLC traffic with
+      // 2 waves/SIMD (980 TF @4096^3, profiles/gemm_variants_r01c).
+      LAUNCH_CFG_W(256, 128, 4, 4, 2, 4, 2);
+    } else if (N > M && t_128x256 >= 512) {
+      LAUNCH_CFG_W(128, 256, 4, 4, 2, 2, 4);
+    } else if (b128 >= 160) {
+      // 128^2, 16 waves (4x4), 32x32/wave: half the 64^2 tile's LLC
+      // re-reads at full occupancy even at 1 block/CU.
+      LAUNCH_CFG_W(128, 128, 2, 2, 4, 4, 4);
     } else {
       LAUNCH_CFG(64, 64, 2, 2, 6);
     }
 #undef LAUNCH_CFG
+#undef LAUNCH_CFG_W
   } else {
     const int64_t total = (int64_t)M * N;
     const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
