@@ -355,6 +355,27 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     case 23: LAUNCH_VW(256, 128, 4, 4, 2, 4, 2); break;   // 8 waves 64x64
     case 24: LAUNCH_VW(128, 256, 4, 4, 2, 2, 4); break;
     case 25: LAUNCH_VW(128, 128, 2, 2, 4, 4, 4); break;   // 16 waves mw4
+    case 26: LAUNCH_VW(256, 128, 4, 2, 2, 4, 4); break;   // 16w, 64x32/wave
+    case 27: LAUNCH_VW(256, 256, 4, 4, 2, 4, 4); break;   // 16w, 64x64/wave
+    case 28: LAUNCH_VW(256, 64, 4, 2, 2, 4, 2); break;    // 8w, 64x32/wave
+    case 29: {  // 256x128 8w with KSTEP=64
+      TORCH_CHECK(K % 64 == 0, "probe: K %% 64");
+      const int mt = (M + 255) / 256, nt = (N + 127) / 128;
+      hipLaunchKernelGGL(
+          (gemm_nt_bf16_kernel<256, 128, 4, 4, 2, 64, false, 4, 2>),
+          dim3(mt * nt), dim3(512), 0, stream.stream(), a, b, c, nullptr, M,
+          N, K, lda, ldb, ldc, 0, mt, nt);
+      break;
+    }
+    case 30: {  // 128x128 16w with KSTEP=64
+      TORCH_CHECK(K % 64 == 0, "probe: K %% 64");
+      const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+      hipLaunchKernelGGL(
+          (gemm_nt_bf16_kernel<128, 128, 2, 2, 4, 64, false, 4, 4>),
+          dim3(mt * nt), dim3(1024), 0, stream.stream(), a, b, c, nullptr, M,
+          N, K, lda, ldb, ldc, 0, mt, nt);
+      break;
+    }
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V

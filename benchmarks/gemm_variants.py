@@ -37,6 +37,11 @@ VARIANTS = {
     23: "256x128 8w (4x2 grid, 64x64/wave) mw2",
     24: "128x256 8w (2x4 grid, 64x64/wave) mw2",
     25: "128x128 16w mw4",
+    26: "256x128 16w (64x32/wave) mw2",
+    27: "256x256 16w (64x64/wave) mw2",
+    28: "256x64 8w (64x32/wave) mw2",
+    29: "256x128 8w k64",
+    30: "128x128 16w k64",
 }
 
 SHAPES = [(2048, 2048, 3072), (4096, 2048, 3072), (2048, 3072, 2048),
