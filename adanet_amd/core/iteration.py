@@ -208,10 +208,6 @@ class _Iteration(object):
         return ([s.name for s in self.subnetwork_specs] +
                 [s.name for s in self.ensemble_specs])
 
-    def _owned(self, spec) -> bool:
-        return spec.owner_rank == comm.rank() or (
-            self.placement is not None and self.placement.data_parallel)
-
     def compute_frozen_outputs(self, features, training: bool = False):
         """Runs every frozen previous-iteration subnetwork once per batch
         under no_grad; the resulting HBM-resident logits feed every

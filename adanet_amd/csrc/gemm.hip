@@ -5,10 +5,11 @@
 // tf.layers.dense / tf.matmul in adanet/examples/simple_dnn.py:74-86 and
 // the MATRIX mixture weights in adanet/ensemble/weighted.py:449) —
 // re-designed CDNA4-native rather than ported:
-//   * templated tile config: 128x128 (4 waves, 64x64/wave, 4x4 MFMA
-//     fragments) for large problems; 64x64 (4 waves, 32x32/wave) when the
-//     128-tile grid would underfill the 256-CU chip — the launch needs
-//     well over 256 workgroups before the big tile pays.
+//   * templated tile/wave-grid config (measured grid in
+//     profiles/gemm_variants_r01*.json): 256x128 with 8 waves (64x64/wave)
+//     for large problems, 128x128 with 16 waves (32x32/wave) for mid
+//     shapes (half the LLC re-read traffic of a 64 tile at full occupancy
+//     even at 1 block/CU), 64x64 for small/skinny shapes.
 //   * BK=32 K-steps staged double-buffered with global_load_lds (16 B per
 //     lane, wave-uniform LDS base: the direct HBM->LDS path).
 //   * one s_barrier per K-tile; the compiler's vmcnt drain at the barrier

@@ -12,9 +12,10 @@ column-sum for the bias gradient:
 The ReLU mask is recovered from the saved forward OUTPUT (y > 0), so the
 fused epilogue never materializes a mask tensor.
 
-Out-features are padded up to a multiple of 8 internally so every GEMM and
-logit buffer keeps an 8-element-aligned row stride (the gemm/global_load_lds
-fast path needs 16 B alignment); callers see the narrow [B, N] view.
+Out-features are padded up to a multiple of 32 internally so every GEMM and
+logit buffer keeps a 16 B-aligned row stride (global_load_lds fast path) AND
+stays a valid K%32 reduction dim for the backward dX GEMM; callers see the
+narrow [B, N] view.
 
 Reference dependency being replaced: tf.layers.dense in subnetworks
 (adanet/examples/simple_dnn.py:74-86).
