@@ -10,8 +10,12 @@ replay. Compute path: PyTorch-ROCm + hand-written HIP/CDNA4 kernels
 
 from adanet_amd import distributed
 from adanet_amd import ensemble
+from adanet_amd import head
+from adanet_amd import hooks
+from adanet_amd import models
 from adanet_amd import ops
 from adanet_amd import replay
+from adanet_amd import serving
 from adanet_amd import subnetwork
 from adanet_amd.autoensemble.common import AutoEnsembleSubestimator
 from adanet_amd.autoensemble.estimator import AutoEnsembleEstimator
@@ -40,8 +44,12 @@ __all__ = [
     "WeightedSubnetwork",
     "distributed",
     "ensemble",
+    "head",
+    "hooks",
+    "models",
     "ops",
     "replay",
+    "serving",
     "subnetwork",
     "__version__",
 ]
