@@ -148,3 +148,16 @@ def test_countdown_timer():
     assert 9.0 < t.secs_remaining() <= 10.0
     t2 = _CountDownTimer(0.0)
     assert t2.secs_remaining() == 0.0
+
+
+def test_device_loader_cpu_passthrough():
+    from adanet_amd.data import DeviceLoader
+    data = [(torch.ones(2, 3), torch.zeros(2))] * 3
+
+    def input_fn():
+        return iter(list(data))
+
+    loader = DeviceLoader(input_fn, device="cpu")
+    out = list(loader())
+    assert len(out) == 3
+    assert torch.equal(out[0][0], torch.ones(2, 3))

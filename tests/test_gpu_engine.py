@@ -147,3 +147,38 @@ def test_gpu_modelflow(tmp_path):
     search.run()
     best = search.get_best_models(1)[0]
     assert isinstance(best.module, MeanEnsemble)
+
+
+def test_device_loader_prefetch(tmp_path):
+    """DeviceLoader: overlapped pinned H2D feeding a short training run."""
+    import adanet_amd
+    from adanet_amd.data import DeviceLoader
+    from adanet_amd.head import MultiClassHead
+    from adanet_amd.models import simple_dnn
+
+    torch.manual_seed(0)
+    N, D, C = 1024, 64, 4
+    X = torch.randn(N, D)
+    Y = (X @ torch.randn(D, C)).argmax(dim=1)
+
+    def cpu_input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(3)
+            while True:
+                idx = torch.randint(0, N, (128,), generator=g)
+                yield X[idx], Y[idx]
+
+        return gen()
+
+    loader = DeviceLoader(cpu_input_fn, device="cuda:0")
+    f, l = next(loader())
+    assert f.is_cuda and f.dtype == torch.bfloat16
+    assert l.is_cuda
+
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(C),
+        subnetwork_generator=simple_dnn.Generator(layer_size=32),
+        max_iteration_steps=10, model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=1))
+    est.train(loader, max_steps=20)
+    assert est.iteration_number == 2
