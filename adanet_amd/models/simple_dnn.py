@@ -64,8 +64,9 @@ class _SimpleDNNBuilder(Builder):
 
     def __init__(self, optimizer_fn, layer_size: int, num_layers: int,
                  learn_mixture_weights: bool, dropout: float, seed=None,
-                 name_suffix: str = ""):
+                 name_suffix: str = "", mixture_optimizer_fn=None):
         self._optimizer_fn = optimizer_fn
+        self._mixture_optimizer_fn = mixture_optimizer_fn or optimizer_fn
         self._layer_size = layer_size
         self._num_layers = num_layers
         self._learn_mixture_weights = learn_mixture_weights
@@ -106,7 +107,7 @@ class _SimpleDNNBuilder(Builder):
         params = list(params)
         if not params:
             return None
-        return self._optimizer_fn(params)
+        return self._mixture_optimizer_fn(params)
 
     def build_subnetwork_report(self) -> Report:
         return Report(hparams={"layer_size": self._layer_size,
@@ -125,12 +126,14 @@ class Generator(Generator):
     def __init__(self, optimizer_fn=None, layer_size: int = 32,
                  initial_num_layers: int = 0,
                  learn_mixture_weights: bool = False, dropout: float = 0.0,
-                 seed: Optional[int] = None, num_restarts: int = 1):
+                 seed: Optional[int] = None, num_restarts: int = 1,
+                 mixture_optimizer_fn=None):
         if optimizer_fn is None:
             optimizer_fn = functools.partial(FusedSGD, lr=0.01)
         self._builder_fn = functools.partial(
             _SimpleDNNBuilder,
             optimizer_fn=optimizer_fn,
+            mixture_optimizer_fn=mixture_optimizer_fn,
             layer_size=layer_size,
             learn_mixture_weights=learn_mixture_weights,
             dropout=dropout,

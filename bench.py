@@ -126,6 +126,7 @@ def main():
     restarts = args.restarts if args.restarts > 0 else world
     generator = simple_dnn.Generator(
         optimizer_fn=functools.partial(FusedSGD, lr=args.lr, momentum=0.9),
+        mixture_optimizer_fn=functools.partial(FusedSGD, lr=args.lr / 10),
         layer_size=args.hidden,
         initial_num_layers=1,
         learn_mixture_weights=True,

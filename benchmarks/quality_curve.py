@@ -62,7 +62,8 @@ def main():
     est = adanet_amd.Estimator(
         head=MultiClassHead(C),
         subnetwork_generator=simple_dnn.Generator(
-            optimizer_fn=functools.partial(FusedSGD, lr=0.05, momentum=0.9),
+            optimizer_fn=functools.partial(FusedSGD, lr=0.02, momentum=0.9),
+            mixture_optimizer_fn=functools.partial(FusedSGD, lr=0.005),
             layer_size=args.hidden, initial_num_layers=1,
             learn_mixture_weights=True, seed=7),
         max_iteration_steps=args.steps_per_iter,
