@@ -9,10 +9,13 @@ winner on the complexity-regularized objective, freeze it, checkpoint, and
 grow. Data is synthetic CIFAR-10-shaped (3072-dim inputs, 10 classes,
 teacher-generated labels), weights random-init, compute dtype bf16.
 
-Multi-GPU (launched by the driver via torch.distributed.run): synchronous
-data-parallel replication over RCCL/xGMI with a fixed per-GPU batch (weak
-scaling — per-GPU work constant; the whole job still completes the same
-number of AdaNet iterations, so `value` is the job's iterations/hour).
+Multi-GPU (launched by the driver via torch.distributed.run): round-robin
+candidate-per-GPU placement (the north star's scaling axis) with the
+candidate pool widened by independent restarts so every GPU owns 2
+candidates at every N — per-GPU work constant (weak scaling), hipGraphs
+stay enabled at all N (no in-step collectives), and the winner broadcasts
+over RCCL/xGMI at each iteration end. --placement replication selects
+synchronous data-parallel instead (flat per-candidate gradient buckets).
 """
 
 import argparse
