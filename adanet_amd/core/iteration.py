@@ -355,11 +355,8 @@ class _Iteration(object):
                     if spec.optimizer is not None and getattr(
                             spec.optimizer, "_adanet_lr_sched", None):
                         ok = False
-                    if spec.subnetwork is not None:
-                        from adanet_amd.ops.dropout import HipDropout
-                        for m in spec.subnetwork.module.modules():
-                            if isinstance(m, HipDropout) and m.p > 0:
-                                ok = False
+                    # HipDropout is graph-safe: its seed is a device
+                    # counter snapshot, so each replay draws a fresh mask.
             self._graph_ok = ok
         return bool(self._graph_ok)
 

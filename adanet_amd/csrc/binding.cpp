@@ -48,8 +48,10 @@ void layernorm_bwd(const at::Tensor& x, const at::Tensor& dy,
                    const at::Tensor& mean, const at::Tensor& rstd,
                    at::Tensor& dx, const c10::optional<at::Tensor>& dgamma,
                    const c10::optional<at::Tensor>& dbeta);
-void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed);
-void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed);
+void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p,
+                 const at::Tensor& seed);
+void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
+                 const at::Tensor& seed);
 void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
 void colsum_bf16(const at::Tensor& x, at::Tensor& out);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,

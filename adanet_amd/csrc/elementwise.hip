@@ -9,11 +9,16 @@
 #include <ATen/cuda/CUDAContext.h>
 #include "common.h"
 
+// The seed arrives via DEVICE memory (seed_ptr): a per-module device
+// counter advances each forward, so the op is hipGraph-capturable with a
+// fresh mask per replay (a host-scalar seed would bake one mask into the
+// graph). Backward reads the forward's snapshot tensor.
 __global__ __launch_bounds__(256) void dropout_fwd_kernel(
     const bf16_t* __restrict__ x, bf16_t* __restrict__ y, int64_t n, float p,
-    uint64_t seed) {
+    const uint64_t* __restrict__ seed_ptr) {
   const float scale = 1.f / (1.f - p);
   const uint32_t thresh = (uint32_t)(p * 4294967296.0f);
+  const uint64_t seed = *seed_ptr;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -24,9 +29,10 @@ __global__ __launch_bounds__(256) void dropout_fwd_kernel(
 
 __global__ __launch_bounds__(256) void dropout_bwd_kernel(
     const bf16_t* __restrict__ dy, bf16_t* __restrict__ dx, int64_t n, float p,
-    uint64_t seed) {
+    const uint64_t* __restrict__ seed_ptr) {
   const float scale = 1.f / (1.f - p);
   const uint32_t thresh = (uint32_t)(p * 4294967296.0f);
+  const uint64_t seed = *seed_ptr;
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -50,23 +56,29 @@ static int ew_grid(int64_t n) {
   return (int)std::min<int64_t>((n + 255) / 256, 2048);
 }
 
-void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p, int64_t seed) {
+void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p,
+                 const at::Tensor& seed) {
   const int64_t n = x.numel();
   if (n == 0) return;
+  TORCH_CHECK(seed.is_cuda() && seed.scalar_type() == at::kLong,
+              "dropout: device int64 seed tensor required");
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(dropout_fwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)x.data_ptr(),
-                     (bf16_t*)y.data_ptr(), n, (float)p, (uint64_t)seed);
+                     (bf16_t*)y.data_ptr(), n, (float)p,
+                     (const uint64_t*)seed.data_ptr());
   HIP_CHECK_KERNEL();
 }
 
-void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p, int64_t seed) {
+void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
+                 const at::Tensor& seed) {
   const int64_t n = dy.numel();
   if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(dropout_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)dy.data_ptr(),
-                     (bf16_t*)dx.data_ptr(), n, (float)p, (uint64_t)seed);
+                     (bf16_t*)dx.data_ptr(), n, (float)p,
+                     (const uint64_t*)seed.data_ptr());
   HIP_CHECK_KERNEL();
 }
 

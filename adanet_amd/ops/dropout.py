@@ -1,9 +1,10 @@
 """HipDropout (K3): stateless counter-based-RNG dropout.
 
 The mask is a pure function of (seed, element index) so backward recomputes
-it — no mask tensor is stored or transferred. Seeds advance from a
-per-process counter seeded by torch's RNG so runs are reproducible under
-torch.manual_seed.
+it — no mask tensor is stored. The seed is a DEVICE counter snapshot: the
+module's counter advances on-device each forward, which keeps the op
+hipGraph-capturable with a fresh mask per replay (graph-eligibility no
+longer excludes dropout-bearing search spaces).
 """
 
 from __future__ import annotations
@@ -13,34 +14,26 @@ from torch import nn
 
 from adanet_amd.ops import _extension
 
-_seed_counter = [0]
-
-
-def _next_seed() -> int:
-    if _seed_counter[0] == 0:
-        _seed_counter[0] = int(torch.initial_seed()) & 0x7FFFFFFFFFFF or 1
-    _seed_counter[0] += 0x9E3779B9
-    return _seed_counter[0]
-
 
 class _DropoutFn(torch.autograd.Function):
 
     @staticmethod
-    def forward(ctx, x, p, seed):
+    def forward(ctx, x, p, seed_snapshot):
         ext = _extension.require()
         x = x.contiguous()
         y = torch.empty_like(x)
-        ext.dropout_fwd(x, y, p, seed)
+        ext.dropout_fwd(x, y, p, seed_snapshot)
         ctx.p = p
-        ctx.seed = seed
+        ctx.save_for_backward(seed_snapshot)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         ext = _extension.require()
+        (seed_snapshot,) = ctx.saved_tensors
         dy = dy.contiguous()
         dx = torch.empty_like(dy)
-        ext.dropout_bwd(dy, dx, ctx.p, ctx.seed)
+        ext.dropout_bwd(dy, dx, ctx.p, seed_snapshot)
         return dx, None, None
 
 
@@ -51,11 +44,23 @@ class HipDropout(nn.Module):
         if not 0.0 <= p < 1.0:
             raise ValueError("dropout p must be in [0, 1)")
         self.p = p
+        # Device counter (int64 buffer): advanced on-device each forward so
+        # graph replays draw fresh masks. Seeded lazily from torch's RNG.
+        self.register_buffer("_seed", torch.zeros(1, dtype=torch.int64),
+                             persistent=False)
+        self._seeded = False
 
     def forward(self, x):
         if not self.training or self.p == 0.0:
             return x
         if x.is_cuda:
-            return _DropoutFn.apply(x.to(torch.bfloat16), self.p,
-                                    _next_seed())
+            if not self._seeded or self._seed.device != x.device:
+                base = int(torch.initial_seed()) & 0x7FFFFFFFFFFF
+                self._seed = torch.tensor(
+                    [base + id(self) % 100003], dtype=torch.int64,
+                    device=x.device)
+                self._seeded = True
+            snapshot = self._seed.clone()          # device op (capturable)
+            self._seed.add_(0x9E3779B9)            # device op (capturable)
+            return _DropoutFn.apply(x.to(torch.bfloat16), self.p, snapshot)
         return torch.nn.functional.dropout(x, self.p, training=True)

@@ -294,15 +294,31 @@ def test_dropout_stats_and_bwd_mask_match():
     ext = _ext()
     x = torch.ones(1 << 20, device=DEV, dtype=torch.bfloat16)
     y = torch.empty_like(x)
-    ext.dropout_fwd(x, y, 0.25, 1234)
+    seed = torch.tensor([1234], dtype=torch.int64, device=DEV)
+    ext.dropout_fwd(x, y, 0.25, seed)
     kept = (y != 0).float().mean().item()
     assert abs(kept - 0.75) < 0.01
     assert abs(y.float().mean().item() - 1.0) < 0.02  # scaled to E[x]
     dy = torch.ones_like(x)
     dx = torch.empty_like(x)
-    ext.dropout_bwd(dy, dx, 0.25, 1234)
+    ext.dropout_bwd(dy, dx, 0.25, seed)
     # identical mask in fwd and bwd
     assert torch.equal((y != 0), (dx != 0))
+
+
+def test_dropout_module_fresh_masks_and_autograd():
+    from adanet_amd.ops.dropout import HipDropout
+    torch.manual_seed(0)
+    d = HipDropout(0.5).to(DEV)
+    d.train()
+    x = torch.ones(1 << 16, device=DEV, dtype=torch.bfloat16,
+                   requires_grad=True)
+    y1 = d(x)
+    y2 = d(x)
+    # device counter advanced -> different masks across calls
+    assert not torch.equal(y1, y2)
+    y1.sum().backward()
+    assert torch.equal((x.grad != 0), (y1 != 0))
 
 
 def test_relu_bwd():
