@@ -28,32 +28,40 @@ class AutoEnsembleSubestimator:
 
 class _CallableModule(SubnetworkModule):
     """Wraps a user nn.Module whose forward returns logits (or
-    (last_layer, logits)) into the Subnetwork contract."""
+    (last_layer, logits), or a predictions dict) into the Subnetwork
+    contract. ``logits_fn`` / ``last_layer_fn`` extract the respective
+    tensors from the raw output (reference common.py:31-40 default digs
+    predictions["logits"]; custom last_layer_fn per estimator_test.py:415).
+    """
 
-    def __init__(self, module: nn.Module, logits_fn=None):
+    def __init__(self, module: nn.Module, logits_fn=None, last_layer_fn=None):
         super().__init__()
         self.inner = module
         self._logits_fn = logits_fn
+        self._last_layer_fn = last_layer_fn
         d = getattr(module, "last_layer_dim", None)
         if d is not None:
             self.last_layer_dim = d
 
     def forward(self, features):
         out = self.inner(features)
+        raw = out
         if self._logits_fn is not None:
             out = self._logits_fn(out)
         if isinstance(out, tuple):
-            return out
-        if isinstance(out, dict):
-            # Reference default logits_fn digs predictions["logits"]
-            # (common.py:31-40).
+            last, logits = out
+        elif isinstance(out, dict):
             logits = out.get("logits", None)
             if logits is None:
                 raise ValueError(
                     "Subestimator predictions need a 'logits' key or a "
                     "custom logits_fn")
-            return out.get("last_layer", logits), logits
-        return out, out
+            last = out.get("last_layer", logits)
+        else:
+            last, logits = out, out
+        if self._last_layer_fn is not None:
+            last = self._last_layer_fn(raw)
+        return last, logits
 
 
 class _BuilderFromSubestimator(Builder):
@@ -99,8 +107,11 @@ class _BuilderFromSubestimator(Builder):
     def build_subnetwork(self, features, logits_dimension, training,
                          previous_ensemble=None) -> Subnetwork:
         module = self._build_module(features, logits_dimension, training)
-        if not isinstance(module, SubnetworkModule):
-            module = _CallableModule(module, self._logits_fn)
+        if not isinstance(module, SubnetworkModule) or (
+                self._logits_fn is not None
+                or self._last_layer_fn is not None):
+            module = _CallableModule(module, self._logits_fn,
+                                     self._last_layer_fn)
         return Subnetwork(module=module, complexity=0.0, name=self._name)
 
     def build_optimizer(self, params, iteration: int = 0):
