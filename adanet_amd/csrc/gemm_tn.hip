@@ -1,0 +1,287 @@
+// Transposed-operand GEMM staging via ds_read_b64_tr_b16 (gfx950 hardware
+// transpose read) — lets the backward GEMMs consume K-major operands
+// (dz/W/x as stored) with NO separate transpose kernels:
+//
+//   dX[B,Kin]  = dz[B,Nout] @ W[Nout,Kin]   -> B-operand transposed
+//   dW[N,Kin]  = dz^T @ x                   -> BOTH operands transposed
+//
+// Scheme per transposed operand (KSTEP=32):
+//   * LDS holds [n_colblocks][32 k][16 cols] subtiles. One global_load_lds
+//     (64 lanes x 16 B = 1 KiB) stages EXACTLY one subtile: lane l fetches
+//     G[k0 + l/2][col0 + blk*16 + (l%2)*8 .. +8] (per-lane source, linear
+//     wave-uniform LDS dest) — same op count as normal staging.
+//   * fragment read: mfma b-frag lane l needs T[k=(l>>4)*8+j][l&15],
+//     j=0..7. ds_read_b64_tr_b16 delivers lane l elems j=0..3 from
+//     lds_half[(l&15) + j*16 + (l>>4)*64 + base]; choosing per-lane
+//     base = subtile + 64*(l>>4) (and +64 for j=4..7) maps exactly onto
+//     the needed [k][col] elements: two tr-reads build one 8-halfword
+//     fragment. (Layout verified by probe_tr16_layout below.)
+//
+// Inline-asm ds_read needs an explicit lgkmcnt(0) + sched_barrier(0) fence
+// before the MFMAs (the compiler does not order register-only MFMAs
+// against asm loads — CDNA4 guide rule 18).
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+typedef s16x8 frag_ab;
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
+
+typedef __attribute__((address_space(3))) const bf16_t* lds_cptr_t;
+
+__device__ __forceinline__ u32x2 tr_b16_read(const bf16_t* lds_half_addr) {
+  // AS(3) pointers are 32-bit LDS offsets on AMDGPU; the addrspacecast
+  // recovers the DS-instruction operand from the generic pointer.
+  const unsigned off = (unsigned)(uintptr_t)(lds_cptr_t)lds_half_addr;
+  u32x2 out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1"
+               : "=v"(out)
+               : "v"(off)
+               : "memory");
+  return out;
+}
+
+// ---------------------------------------------------------------- probe
+// Writes the tr-read of a ramp-filled [32][16] subtile so python can verify
+// the lane->element mapping before trusting the GEMM.
+__global__ void probe_tr16_layout_kernel(float* __restrict__ out) {
+  __shared__ bf16_t tile[32 * 16];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 32 * 16; i += blockDim.x) {
+    tile[i] = f2bf((float)i);  // ramp: value == halfword offset
+  }
+  __syncthreads();
+  const bf16_t* base0 = &tile[64 * (lane >> 4)];
+  const bf16_t* base1 = &tile[64 * (lane >> 4) + 64];
+  u32x2 r0 = tr_b16_read(base0);
+  u32x2 r1 = tr_b16_read(base1);
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
+  const unsigned short* h0 = (const unsigned short*)&r0;
+  const unsigned short* h1 = (const unsigned short*)&r1;
+  if (blockIdx.x == 0) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const unsigned short a = h0[j];
+      const unsigned short b = h1[j];
+      out[lane * 8 + j] = bits2f(*(const short*)&a);
+      out[lane * 8 + 4 + j] = bits2f(*(const short*)&b);
+    }
+  }
+}
+
+void probe_tr16_layout(at::Tensor& out) {
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(probe_tr16_layout_kernel, dim3(1), dim3(64), 0,
+                     stream.stream(), out.data_ptr<float>());
+  HIP_CHECK_KERNEL();
+}
+
+// ------------------------------------------------------- transposed stage
+// Stage a [ROWS x 32] K-minor logical tile whose GLOBAL source is K-major
+// (G[k][col]): one 1 KiB op per [32][16] subtile, wave w takes subtiles
+// strided by NWAVES.
+template <int ROWS, int NWAVES>
+__device__ __forceinline__ void stage_tile_tr(
+    const bf16_t* __restrict__ G, int ld, int tile_col0, int max_col, int k0,
+    int max_k, bf16_t* __restrict__ lds, int wid, int lane) {
+  constexpr int SUBTILES = ROWS / 16;
+#pragma unroll
+  for (int st = wid; st < SUBTILES; st += NWAVES) {
+    const int k = k0 + (lane >> 1);
+    int col = tile_col0 + st * 16 + (lane & 1) * 8;
+    if (col + 8 > max_col) {
+      // clamp to the last 8-aligned window (masked on C-store anyway).
+      col = max(0, (max_col - 8) & ~7);
+    }
+    const bf16_t* gp = G + (int64_t)min(k, max_k - 1) * ld + col;
+    bf16_t* lp = lds + st * 32 * 16;  // wave-uniform subtile base
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+// Fragment from a transposed-staged operand: two tr-reads.
+__device__ __forceinline__ frag_ab frag_from_tr(const bf16_t* lds,
+                                                int row_block, int lane) {
+  const bf16_t* sub = lds + row_block * 32 * 16;  // 16 rows' subtile
+  const bf16_t* b0 = sub + 64 * (lane >> 4);
+  const bf16_t* b1 = b0 + 64;
+  u32x2 r0 = tr_b16_read(b0);
+  u32x2 r1 = tr_b16_read(b1);
+  union {
+    frag_ab f;
+    struct { u32x2 lo, hi; } u;
+  } pack;
+  pack.u.lo = r0;
+  pack.u.hi = r1;
+  return pack.f;
+}
+
+// TRB: B-operand K-major (B'[K][N]); TRA additionally A K-major (A'[K][M]).
+template <int BM, int BN, int FM, int FN, int MINWAVES, int WGM, int WGN,
+          bool TRA, bool TRB>
+__global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles) {
+  constexpr int KSTEP = 32;
+  __shared__ bf16_t As[2][BM * KSTEP];
+  __shared__ bf16_t Bs[2][BN * KSTEP];
+  const int nwg = mtiles * ntiles;
+  const int orig = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
+                 (orig >> 3);
+  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  static_assert(BM == WGM * FM * 16 && BN == WGN * FN * 16, "geometry");
+  constexpr int NWAVES = WGM * WGN;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int wm = wid / WGN, wn = wid % WGN;
+  const int row0 = tile_m * BM;
+  const int col0 = tile_n * BN;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int ktiles = K / KSTEP;
+
+#define STAGE_A(buf, k0)                                                      \
+  do {                                                                        \
+    if (TRA)                                                                  \
+      stage_tile_tr<BM, NWAVES>(A, lda, row0, M, k0, K, As[buf], wid, lane);  \
+    else {                                                                    \
+      constexpr int RPS = 512 / KSTEP;                                        \
+      constexpr int LPR = KSTEP / 8;                                          \
+      for (int seg = wid; seg < BM / RPS; seg += NWAVES) {                    \
+        const int rt = seg * RPS + lane / LPR;                                \
+        int grow = row0 + rt;                                                 \
+        grow = grow < M ? grow : M - 1;                                       \
+        const bf16_t* gp = A + (int64_t)grow * lda + k0 + (lane % LPR) * 8;   \
+        bf16_t* lp = As[buf] + seg * RPS * KSTEP;                             \
+        __builtin_amdgcn_global_load_lds(                                     \
+            (const __attribute__((address_space(1))) void*)gp,                \
+            (__attribute__((address_space(3))) void*)lp, 16, 0, 0);           \
+      }                                                                       \
+    }                                                                         \
+  } while (0)
+#define STAGE_B(buf, k0)                                                      \
+  do {                                                                        \
+    if (TRB)                                                                  \
+      stage_tile_tr<BN, NWAVES>(B, ldb, col0, N, k0, K, Bs[buf], wid, lane);  \
+    else {                                                                    \
+      constexpr int RPS = 512 / KSTEP;                                        \
+      constexpr int LPR = KSTEP / 8;                                          \
+      for (int seg = wid; seg < BN / RPS; seg += NWAVES) {                    \
+        const int rt = seg * RPS + lane / LPR;                                \
+        int grow = col0 + rt;                                                 \
+        grow = grow < N ? grow : N - 1;                                       \
+        const bf16_t* gp = B + (int64_t)grow * ldb + k0 + (lane % LPR) * 8;   \
+        bf16_t* lp = Bs[buf] + seg * RPS * KSTEP;                             \
+        __builtin_amdgcn_global_load_lds(                                     \
+            (const __attribute__((address_space(1))) void*)gp,                \
+            (__attribute__((address_space(3))) void*)lp, 16, 0, 0);           \
+      }                                                                       \
+    }                                                                         \
+  } while (0)
+
+  STAGE_A(0, 0);
+  STAGE_B(0, 0);
+  int buf = 0;
+  for (int kt = 0; kt < ktiles; ++kt) {
+    __syncthreads();
+    if (kt + 1 < ktiles) {
+      STAGE_A(buf ^ 1, (kt + 1) * KSTEP);
+      STAGE_B(buf ^ 1, (kt + 1) * KSTEP);
+    }
+    frag_ab a[FM], b[FN];
+    const int kofs = (lane >> 4) * 8;
+#pragma unroll
+    for (int f = 0; f < FM; ++f) {
+      if (TRA) {
+        a[f] = frag_from_tr(As[buf], wm * FM + f, lane);
+      } else {
+        const int arow = wm * (FM * 16) + (lane & 15);
+        a[f] = *(const frag_ab*)&As[buf][(arow + f * 16) * KSTEP + kofs];
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < FN; ++f) {
+      if (TRB) {
+        b[f] = frag_from_tr(Bs[buf], wn * FN + f, lane);
+      } else {
+        const int brow = wn * (FN * 16) + (lane & 15);
+        b[f] = *(const frag_ab*)&Bs[buf][(brow + f * 16) * KSTEP + kofs];
+      }
+    }
+    if (TRA || TRB) {
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+    }
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+      for (int j = 0; j < FN; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j],
+                                                            acc[i][j], 0, 0,
+                                                            0);
+    buf ^= 1;
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr] + bv;
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        C[(int64_t)row * ldc + col] = f2bf(v);
+      }
+    }
+  }
+#undef STAGE_A
+#undef STAGE_B
+}
+
+// Probe entry: trans flags select operand layout. A is [M,K] (or [K,M] when
+// trans_a), B is [N,K] (or [K,N] when trans_b); C[M,N].
+void gemm_tr_probe(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+                   int64_t trans_a, int64_t trans_b, int64_t variant) {
+  const int M = (int)C.size(0), N = (int)C.size(1);
+  const int K = (int)(trans_a ? A.size(0) : A.size(1));
+  const int lda = (int)A.stride(0), ldb = (int)B.stride(0),
+            ldc = (int)C.stride(0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+  const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+#define L(TRA, TRB)                                                           \
+  hipLaunchKernelGGL((gemm_tr_kernel<128, 128, 2, 2, 4, 4, 4, TRA, TRB>),     \
+                     dim3(mt * nt), dim3(1024), 0, stream.stream(), a, b, c, \
+                     nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt)
+  if (trans_a && trans_b) {
+    L(true, true);
+  } else if (trans_b) {
+    L(false, true);
+  } else {
+    L(false, false);
+  }
+#undef L
+  HIP_CHECK_KERNEL();
+}
