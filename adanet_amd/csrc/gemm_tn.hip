@@ -42,6 +42,25 @@ __device__ __forceinline__ u32x2 tr_b16_read(const bf16_t* lds_half_addr) {
   return out;
 }
 
+// Fragment-read addressing (MEASURED semantics, see benchmarks/tr_probe.py
+// round 1: with uniform quad addresses lane l received halfword
+// base + (l&3) replicated — i.e. the instruction is a QUAD-cooperative
+// 4x4 halfword transpose: lane 4q+j supplies row j's byte address; lane
+// 4q+c receives column c of the quad's 4x4 block as its 4 halfwords):
+//   for an mfma fragment from T[32 k][16 cols], lane l needs
+//   T[8g + j][l&15] (g = l>>4, j = 0..7). Column l&15 = 4*((l>>2)&3) +
+//   (l&3), so quad (l>>2) targets column group m4 = 4*((l>>2)&3) and lane
+//   l provides row (8g + (l&3)) [read 1] / (8g + 4 + (l&3)) [read 2].
+__device__ __forceinline__ void tr_frag_addrs(const bf16_t* sub, int lane,
+                                              const bf16_t** a1,
+                                              const bf16_t** a2) {
+  const int g = lane >> 4;
+  const int m4 = ((lane >> 2) & 3) * 4;
+  const int r = lane & 3;
+  *a1 = sub + (8 * g + r) * 16 + m4;
+  *a2 = sub + (8 * g + 4 + r) * 16 + m4;
+}
+
 // ---------------------------------------------------------------- probe
 // Writes the tr-read of a ramp-filled [32][16] subtile so python can verify
 // the lane->element mapping before trusting the GEMM.
@@ -52,10 +71,10 @@ __global__ void probe_tr16_layout_kernel(float* __restrict__ out) {
     tile[i] = f2bf((float)i);  // ramp: value == halfword offset
   }
   __syncthreads();
-  const bf16_t* base0 = &tile[64 * (lane >> 4)];
-  const bf16_t* base1 = &tile[64 * (lane >> 4) + 64];
-  u32x2 r0 = tr_b16_read(base0);
-  u32x2 r1 = tr_b16_read(base1);
+  const bf16_t *a1, *a2;
+  tr_frag_addrs(tile, lane, &a1, &a2);
+  u32x2 r0 = tr_b16_read(a1);
+  u32x2 r1 = tr_b16_read(a2);
   asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
   __builtin_amdgcn_sched_barrier(0);
   const unsigned short* h0 = (const unsigned short*)&r0;
@@ -103,14 +122,14 @@ __device__ __forceinline__ void stage_tile_tr(
   }
 }
 
-// Fragment from a transposed-staged operand: two tr-reads.
+// Fragment from a transposed-staged operand: two quad-transpose reads.
 __device__ __forceinline__ frag_ab frag_from_tr(const bf16_t* lds,
                                                 int row_block, int lane) {
   const bf16_t* sub = lds + row_block * 32 * 16;  // 16 rows' subtile
-  const bf16_t* b0 = sub + 64 * (lane >> 4);
-  const bf16_t* b1 = b0 + 64;
-  u32x2 r0 = tr_b16_read(b0);
-  u32x2 r1 = tr_b16_read(b1);
+  const bf16_t *a1, *a2;
+  tr_frag_addrs(sub, lane, &a1, &a2);
+  u32x2 r0 = tr_b16_read(a1);
+  u32x2 r1 = tr_b16_read(a2);
   union {
     frag_ab f;
     struct { u32x2 lo, hi; } u;

@@ -28,19 +28,13 @@ def main():
     print("lane31:", vals[31])
     print("lane32:", vals[32])
     print("lane48:", vals[48])
-    # expected if m162 formula includes internal lane offsets relative to a
-    # per-lane base of 64*(l>>4): value[j<4] = (l&15) + j*16 + 128*(l>>4)?
-    ok_formula = all(
-        vals[l][j] == (l & 15) + j * 16 + 64 * (l >> 4)
-        and vals[l][4 + j] == (l & 15) + j * 16 + 64 * (l >> 4) + 64
+    # Expected under the quad-transpose model with tr_frag_addrs: lane l
+    # elem j == T[8g + j][l&15] == (8*(l>>4)+j)*16 + (l&15).
+    want_ok = all(
+        vals[l][j] == (8 * (l >> 4) + j) * 16 + (l & 15)
+        and vals[l][4 + j] == (8 * (l >> 4) + 4 + j) * 16 + (l & 15)
         for l in range(64) for j in range(4))
-    print("matches 'internal lane formula on top of per-lane base':",
-          ok_formula)
-    plain = all(
-        vals[l][j] == 64 * (l >> 4) + j
-        and vals[l][4 + j] == 64 * (l >> 4) + 64 + j
-        for l in range(64) for j in range(4))
-    print("matches 'plain per-lane 4-halfword read':", plain)
+    print("matches fragment mapping T[8g+j][l&15]:", want_ok)
 
     # 2) numerics of the transposed GEMM probes
     torch.manual_seed(0)
