@@ -51,14 +51,21 @@ __device__ __forceinline__ u32x2 tr_b16_read(const bf16_t* lds_half_addr) {
 //   T[8g + j][l&15] (g = l>>4, j = 0..7). Column l&15 = 4*((l>>2)&3) +
 //   (l&3), so quad (l>>2) targets column group m4 = 4*((l>>2)&3) and lane
 //   l provides row (8g + (l&3)) [read 1] / (8g + 4 + (l&3)) [read 2].
+// MEASURED (tr_probe rounds 1-2): ds_read_b64_tr_b16 is a 16-lane-group
+// cooperative transpose. Each lane mu loads 4 contiguous halfwords at its
+// own address; the result redistributes them as
+//     result(l, j) = data(lane 16*(l>>4) + 4*j + ((l>>2)&3)) [l&3].
+// For an mfma fragment from T[32 k][16 c] (lane l needs T[8g+j][l&15],
+// j=0..7) the source-lane address must therefore be
+//     A(mu) = (8*(mu>>4) + ((mu>>2)&3) + 4*phase)*16 + 4*(mu&3)
+// with phase 0/1 for j=0..3 / 4..7 (A2 = A1 + 64 halfwords).
 __device__ __forceinline__ void tr_frag_addrs(const bf16_t* sub, int lane,
                                               const bf16_t** a1,
                                               const bf16_t** a2) {
-  const int g = lane >> 4;
-  const int m4 = ((lane >> 2) & 3) * 4;
-  const int r = lane & 3;
-  *a1 = sub + (8 * g + r) * 16 + m4;
-  *a2 = sub + (8 * g + 4 + r) * 16 + m4;
+  const int row = 8 * (lane >> 4) + ((lane >> 2) & 3);
+  const int col4 = 4 * (lane & 3);
+  *a1 = sub + row * 16 + col4;
+  *a2 = sub + (row + 4) * 16 + col4;
 }
 
 // ---------------------------------------------------------------- probe
