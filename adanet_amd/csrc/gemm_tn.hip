@@ -311,3 +311,66 @@ void gemm_tr_probe(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
 #undef L
   HIP_CHECK_KERNEL();
 }
+
+// Production entry for transposed-operand GEMMs (the backward dX/dW calls):
+// C[M,N] = op(A) @ op(B)^T with A [K,M] when trans_a (else [M,K]) and
+// B [K,N] when trans_b (else [N,K]). Geometry dispatch mirrors gemm.hip.
+void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+                  const c10::optional<at::Tensor>& bias, int64_t act,
+                  int64_t trans_a, int64_t trans_b) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda() && C.is_cuda(), "gemm_tr: GPU only");
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
+                  B.scalar_type() == at::kBFloat16,
+              "gemm_tr: bf16 required");
+  const int M = (int)C.size(0), N = (int)C.size(1);
+  const int K = (int)(trans_a ? A.size(0) : A.size(1));
+  TORCH_CHECK((int)(trans_b ? B.size(0) : B.size(1)) == K, "K mismatch");
+  TORCH_CHECK((int)(trans_a ? A.size(1) : A.size(0)) == M, "M mismatch");
+  TORCH_CHECK((int)(trans_b ? B.size(1) : B.size(0)) == N, "N mismatch");
+  const int lda = (int)A.stride(0), ldb = (int)B.stride(0),
+            ldc = (int)C.stride(0);
+  TORCH_CHECK(K % 32 == 0 && lda % 8 == 0 && ldb % 8 == 0,
+              "gemm_tr: fast-path alignment required "
+              "(K%32, strides%8) — pad like HipLinear does");
+  if (M == 0 || N == 0) return;
+  const float* bias_ptr = nullptr;
+  if (bias.has_value() && bias->defined()) bias_ptr = bias->data_ptr<float>();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+  const int64_t b128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  const int64_t t_256x128 = (int64_t)((M + 255) / 256) * ((N + 127) / 128);
+  const int64_t t_128x256 = (int64_t)((M + 127) / 128) * ((N + 255) / 256);
+#define LTR(BM, BN, FM, FN, MW, WGM, WGN)                                     \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    if (trans_a && trans_b) {                                                 \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, true, true>),         \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                \
+    } else if (trans_b) {                                                     \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, false, true>),        \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                \
+    } else {                                                                  \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, true, false>),        \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                \
+    }                                                                         \
+  } while (0)
+  if (M >= N && t_256x128 >= 512) {
+    LTR(256, 128, 4, 4, 2, 4, 2);
+  } else if (N > M && t_128x256 >= 512) {
+    LTR(128, 256, 4, 4, 2, 2, 4);
+  } else if (b128 >= 160) {
+    LTR(128, 128, 2, 2, 4, 4, 4);
+  } else {
+    LTR(64, 64, 2, 2, 6, 2, 2);
+  }
+#undef LTR
+  HIP_CHECK_KERNEL();
+}

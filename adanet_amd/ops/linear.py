@@ -118,13 +118,31 @@ class _LinearFn(torch.autograd.Function):
             else:
                 dz = dy
             dx = dw = None
+            B, N = dz.shape
+            K = weight.shape[1]
             if need_dx:
-                wt = transpose2d(weight)  # [K, N]
-                dx = gemm_nt(dz, wt)      # [B, K]
+                # dX = dz @ W: the transposed-staging GEMM consumes W
+                # K-major directly (ds_read_b64_tr_b16 fragments) — no
+                # W^T materialization. K_r = N is always 32-padded.
+                if weight.stride(0) % 8 == 0 and N % 32 == 0:
+                    dx = torch.empty((B, K), device=dz.device,
+                                     dtype=torch.bfloat16)
+                    ext.gemm_tr_bf16(dz, weight, dx, None, 0, 0, 1)
+                else:
+                    wt = transpose2d(weight)  # [K, N]
+                    dx = gemm_nt(dz, wt)      # [B, K]
             if need_dw:
-                dzt = transpose2d(dz)     # [N, B]
-                xt = transpose2d(x)       # [K, B]
-                dw = gemm_nt(dzt, xt)     # [N, K]
+                # dW = dz^T @ X: both operands K_r(=batch)-major -> both
+                # transposed-staged when the batch is 32-aligned.
+                if (B % 32 == 0 and dz.stride(0) % 8 == 0
+                        and x.stride(0) % 8 == 0):
+                    dw = torch.empty((N, K), device=dz.device,
+                                     dtype=torch.bfloat16)
+                    ext.gemm_tr_bf16(dz, x, dw, None, 0, 1, 1)
+                else:
+                    dzt = transpose2d(dz)     # [N, B]
+                    xt = transpose2d(x)       # [K, B]
+                    dw = gemm_nt(dzt, xt)     # [N, K]
             db = None
             if ctx.has_bias:
                 db = torch.empty((dz.shape[1],), device=dz.device,
