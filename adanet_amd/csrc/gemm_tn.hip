@@ -75,7 +75,7 @@ __global__ void probe_tr16_layout_kernel(float* __restrict__ out) {
   __shared__ bf16_t tile[32 * 16];
   const int lane = threadIdx.x & 63;
   for (int i = threadIdx.x; i < 32 * 16; i += blockDim.x) {
-    tile[i] = f2bf((float)i);  // ramp: value == halfword offset
+    tile[i] = f2bf((float)(i % 256));  // bf16-exact ramp
   }
   __syncthreads();
   const bf16_t *a1, *a2;

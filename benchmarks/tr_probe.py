@@ -31,8 +31,8 @@ def main():
     # Expected under the quad-transpose model with tr_frag_addrs: lane l
     # elem j == T[8g + j][l&15] == (8*(l>>4)+j)*16 + (l&15).
     want_ok = all(
-        vals[l][j] == (8 * (l >> 4) + j) * 16 + (l & 15)
-        and vals[l][4 + j] == (8 * (l >> 4) + 4 + j) * 16 + (l & 15)
+        vals[l][j] == ((8 * (l >> 4) + j) * 16 + (l & 15)) % 256
+        and vals[l][4 + j] == ((8 * (l >> 4) + 4 + j) * 16 + (l & 15)) % 256
         for l in range(64) for j in range(4))
     print("matches fragment mapping T[8g+j][l&15]:", want_ok)
 
