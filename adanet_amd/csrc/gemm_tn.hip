@@ -11,11 +11,9 @@
 //     G[k0 + l/2][col0 + blk*16 + (l%2)*8 .. +8] (per-lane source, linear
 //     wave-uniform LDS dest) — same op count as normal staging.
 //   * fragment read: mfma b-frag lane l needs T[k=(l>>4)*8+j][l&15],
-//     j=0..7. ds_read_b64_tr_b16 delivers lane l elems j=0..3 from
-//     lds_half[(l&15) + j*16 + (l>>4)*64 + base]; choosing per-lane
-//     base = subtile + 64*(l>>4) (and +64 for j=4..7) maps exactly onto
-//     the needed [k][col] elements: two tr-reads build one 8-halfword
-//     fragment. (Layout verified by probe_tr16_layout below.)
+//     j=0..7 — built by two ds_read_b64_tr_b16 ops using the MEASURED
+//     cooperative-transpose semantics (see tr_frag_addrs; verified by
+//     probe_tr16_layout + random-data refchecks).
 //
 // Inline-asm ds_read needs an explicit lgkmcnt(0) + sched_barrier(0) fence
 // before the MFMAs (the compiler does not order register-only MFMAs
@@ -42,15 +40,6 @@ __device__ __forceinline__ u32x2 tr_b16_read(const bf16_t* lds_half_addr) {
   return out;
 }
 
-// Fragment-read addressing (MEASURED semantics, see benchmarks/tr_probe.py
-// round 1: with uniform quad addresses lane l received halfword
-// base + (l&3) replicated — i.e. the instruction is a QUAD-cooperative
-// 4x4 halfword transpose: lane 4q+j supplies row j's byte address; lane
-// 4q+c receives column c of the quad's 4x4 block as its 4 halfwords):
-//   for an mfma fragment from T[32 k][16 cols], lane l needs
-//   T[8g + j][l&15] (g = l>>4, j = 0..7). Column l&15 = 4*((l>>2)&3) +
-//   (l&3), so quad (l>>2) targets column group m4 = 4*((l>>2)&3) and lane
-//   l provides row (8g + (l&3)) [read 1] / (8g + 4 + (l&3)) [read 2].
 // MEASURED (tr_probe rounds 1-2): ds_read_b64_tr_b16 is a 16-lane-group
 // cooperative transpose. Each lane mu loads 4 contiguous halfwords at its
 // own address; the result redistributes them as
