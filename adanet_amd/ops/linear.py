@@ -124,7 +124,12 @@ class _LinearFn(torch.autograd.Function):
                 # dX = dz @ W: the transposed-staging GEMM consumes W
                 # K-major directly (ds_read_b64_tr_b16 fragments) — no
                 # W^T materialization. K_r = N is always 32-padded.
-                if weight.stride(0) % 8 == 0 and N % 32 == 0:
+                # Measured (profiles/tr_vs_composite_r01.json): tr wins
+                # unless the reduction dim dominates both outputs dims
+                # (K_r=4096 > M,N=2048 → 0.87x), so fall back to the
+                # transpose+NT composite there.
+                if (weight.stride(0) % 8 == 0 and N % 32 == 0
+                        and N < max(B, K)):
                     dx = torch.empty((B, K), device=dz.device,
                                      dtype=torch.bfloat16)
                     ext.gemm_tr_bf16(dz, weight, dx, None, 0, 0, 1)
