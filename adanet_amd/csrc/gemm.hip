@@ -160,7 +160,10 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
         } else {
           float v = acc[i][j][rr] + bv;
           if (act == 1) v = v > 0.f ? v : 0.f;
-          C[(int64_t)row * ldc + col] = f2bf(v);
+          // act==2: accumulate into C (direct-to-arena gradient writes).
+          bf16_t* cp = &C[(int64_t)row * ldc + col];
+          if (act == 2) v += bf2f(*cp);
+          *cp = f2bf(v);
         }
       }
     }
@@ -198,6 +201,7 @@ __global__ void gemm_nt_generic_kernel(const bf16_t* __restrict__ A,
       acc += bf2f(A[(int64_t)m * lda + k]) * bf2f(B[(int64_t)n * ldb + k]);
     if (bias) acc += bias[n];
     if (act == 1) acc = acc > 0.f ? acc : 0.f;
+    if (act == 2) acc += bf2f(C[(int64_t)m * ldc + n]);
     C[(int64_t)m * ldc + n] = f2bf(acc);
   }
 }

@@ -259,13 +259,18 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
       const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
       if (col >= N) continue;
       const float bv = bias ? bias[col] : 0.f;
+      // act==2: accumulate into C (direct-to-arena gradient writes — the
+      // backward GEMM adds into the optimizer's grad view, replacing
+      // autograd's separate AccumulateGrad add kernel).
 #pragma unroll
       for (int rr = 0; rr < 4; ++rr) {
         const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
         if (row >= M) continue;
         float v = acc[i][j][rr] + bv;
         if (act == 1) v = v > 0.f ? v : 0.f;
-        C[(int64_t)row * ldc + col] = f2bf(v);
+        bf16_t* cp = &C[(int64_t)row * ldc + col];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
       }
     }
   }

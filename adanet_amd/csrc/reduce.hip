@@ -14,16 +14,18 @@
 // `out` MUST be zero-initialized when gridDim.y > 1.
 __global__ __launch_bounds__(256) void colsum_bf16_kernel(
     const bf16_t* __restrict__ x, float* __restrict__ out, int B, int C,
-    int ldx, int rows_per_block) {
+    int ldx, int rows_per_block, int accum) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const int r0 = blockIdx.y * rows_per_block;
   const int r1 = min(B, r0 + rows_per_block);
   float acc = 0.f;
   for (int b = r0; b < r1; ++b) acc += bf2f(x[(int64_t)b * ldx + c]);
-  if (gridDim.y == 1) {
+  if (gridDim.y == 1 && !accum) {
     out[c] = acc;
   } else {
+    // accum mode: out already holds the running gradient (direct-to-arena
+    // bias-grad write) — always add, never overwrite, never pre-zero.
     atomicAdd(&out[c], acc);
   }
 }
@@ -51,15 +53,15 @@ __global__ __launch_bounds__(256) void argmax_correct_kernel(
   }
 }
 
-void colsum_bf16(const at::Tensor& x, at::Tensor& out) {
+void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum) {
   const int B = (int)x.size(0), C = (int)x.size(1);
   if (C == 0) return;
-  if (B == 0) { out.zero_(); return; }
+  if (B == 0) { if (!accum) out.zero_(); return; }
   auto stream = at::cuda::getCurrentCUDAStream();
   const int stripes = (C + 255) / 256;
   // Fill the chip: aim for ~1024 blocks, at least 8 rows per chunk.
   int row_chunks = std::max(1, std::min(1024 / stripes, (B + 7) / 8));
-  if (row_chunks > 1) {
+  if (row_chunks > 1 && !accum) {
     out.zero_();
   }
   const int rows_per_block = (B + row_chunks - 1) / row_chunks;
@@ -67,7 +69,7 @@ void colsum_bf16(const at::Tensor& x, at::Tensor& out) {
                      dim3((unsigned)stripes, (unsigned)row_chunks), dim3(256),
                      0, stream.stream(), (const bf16_t*)x.data_ptr(),
                      out.data_ptr<float>(), B, C, (int)x.stride(0),
-                     rows_per_block);
+                     rows_per_block, (int)accum);
   HIP_CHECK_KERNEL();
 }
 

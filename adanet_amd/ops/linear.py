@@ -32,6 +32,37 @@ from torch import nn
 from adanet_amd.ops import _extension
 
 
+_DIRECT_GRAD = False
+
+
+class direct_grad_writes:
+    """Enable direct-to-arena gradient writes inside this context.
+
+    With this on, _LinearFn.backward writes dW/db straight into the
+    parameter's existing `.grad` tensor using the GEMM accumulate epilogue
+    (act=2) / colsum accum mode, and returns None to autograd — eliminating
+    AccumulateGrad's separate add kernel and the colsum pre-zero fill
+    (measured 5-6% of step GPU time, profiles/bench_kernel_stats_r01d.txt).
+
+    Semantics match autograd exactly (add into existing grad), so it is only
+    valid when gradients are consumed via `.grad` (optimizer steps), not via
+    `torch.autograd.grad`. The training engine wraps its backward pass in
+    this context; arena-pinned grad views (ops/optim.py) make the target
+    addresses stable across hipGraph replays.
+    """
+
+    def __enter__(self):
+        global _DIRECT_GRAD
+        self._prev = _DIRECT_GRAD
+        _DIRECT_GRAD = True
+        return self
+
+    def __exit__(self, *exc):
+        global _DIRECT_GRAD
+        _DIRECT_GRAD = self._prev
+        return False
+
+
 def _pad8(n: int) -> int:
     # Pad out-features to a multiple of 32: keeps the row stride 16B-aligned
     # for global_load_lds AND keeps N usable as the reduction dim of the
@@ -101,6 +132,11 @@ class _LinearFn(torch.autograd.Function):
         y = gemm_nt(x, weight, bias, activation)
         ctx.activation = activation
         ctx.has_bias = bias is not None
+        # Keep the live Parameter objects so backward can reach their .grad
+        # views for direct-to-arena writes (saved_tensors unwraps to plain
+        # tensors that drop the identity we need).
+        ctx.param_refs = (weight if isinstance(weight, nn.Parameter) else None,
+                          bias if isinstance(bias, nn.Parameter) else None)
         ctx.save_for_backward(x, weight, y if activation == "relu" else None)
         return y
 
@@ -136,23 +172,41 @@ class _LinearFn(torch.autograd.Function):
                 else:
                     wt = transpose2d(weight)  # [K, N]
                     dx = gemm_nt(dz, wt)      # [B, K]
+            wparam, bparam = ctx.param_refs
             if need_dw:
                 # dW = dz^T @ X: both operands K_r(=batch)-major -> both
                 # transposed-staged when the batch is 32-aligned.
+                # Direct mode: accumulate (act=2) into the arena grad view
+                # and return None — no AccumulateGrad add kernel.
+                wg = wparam.grad if (_DIRECT_GRAD and wparam is not None) \
+                    else None
+                direct_w = (wg is not None and wg.is_cuda
+                            and wg.dtype == torch.bfloat16
+                            and wg.shape == (N, K) and wg.is_contiguous())
+                dw_out = wg if direct_w else torch.empty(
+                    (N, K), device=dz.device, dtype=torch.bfloat16)
+                epi = 2 if direct_w else 0
                 if (B % 32 == 0 and dz.stride(0) % 8 == 0
                         and x.stride(0) % 8 == 0):
-                    dw = torch.empty((N, K), device=dz.device,
-                                     dtype=torch.bfloat16)
-                    ext.gemm_tr_bf16(dz, x, dw, None, 0, 1, 1)
+                    ext.gemm_tr_bf16(dz, x, dw_out, None, epi, 1, 1)
                 else:
                     dzt = transpose2d(dz)     # [N, B]
                     xt = transpose2d(x)       # [K, B]
-                    dw = gemm_nt(dzt, xt)     # [N, K]
+                    ext.gemm_nt_bf16(dzt, xt, dw_out, None, epi)
+                dw = None if direct_w else dw_out
             db = None
             if ctx.has_bias:
-                db = torch.empty((dz.shape[1],), device=dz.device,
-                                 dtype=torch.float32)
-                ext.colsum_bf16(dz, db)
+                bg = bparam.grad if (_DIRECT_GRAD and bparam is not None) \
+                    else None
+                direct_b = (bg is not None and bg.is_cuda
+                            and bg.dtype == torch.float32
+                            and bg.shape == (N,) and bg.is_contiguous())
+                if direct_b:
+                    ext.colsum_bf16(dz, bg, 1)
+                else:
+                    db = torch.empty((dz.shape[1],), device=dz.device,
+                                     dtype=torch.float32)
+                    ext.colsum_bf16(dz, db)
         else:
             dzf = dy.float()
             if ctx.activation == "relu":

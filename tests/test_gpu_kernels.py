@@ -414,3 +414,58 @@ def test_arena_flattened_sgd_matches_unflattened_training():
     bufs = opt.flat_grad_buffers()
     assert sum(b.numel() for b in bufs) == sum(
         p.numel() for p in model.parameters())
+
+
+def test_direct_grad_writes_match_autograd():
+    """direct_grad_writes (accumulate epilogue into .grad views) must
+    produce the same dW/db as the standard autograd accumulate path,
+    including double-use accumulation within one backward."""
+    from adanet_amd.ops.linear import HipLinear, direct_grad_writes
+    torch.manual_seed(1)
+    B, K, N = 128, 256, 96
+    lin = HipLinear(K, N, activation="relu").to(DEV)
+    x = torch.randn(B, K, device=DEV).to(torch.bfloat16)
+
+    # reference: plain autograd
+    lin.zero_grad(set_to_none=True)
+    (lin(x).float().mean() + lin(x * 0.5).float().mean()).backward()
+    ref_w = lin.weight.grad.detach().clone()
+    ref_b = lin.bias.grad.detach().clone()
+
+    # direct: pre-pinned zeroed grads + context
+    lin.weight.grad = torch.zeros_like(lin.weight)
+    lin.bias.grad = torch.zeros_like(lin.bias)
+    with direct_grad_writes():
+        (lin(x).float().mean() + lin(x * 0.5).float().mean()).backward()
+    for got, want in [(lin.weight.grad.float(), ref_w.float()),
+                      (lin.bias.grad, ref_b)]:
+        rel = (got - want).abs().mean() / (want.abs().mean() + 1e-6)
+        assert rel < 0.02, rel
+
+
+def test_gemm_accumulate_epilogue():
+    """act=2 adds into C for both the NT and transposed-staging GEMMs."""
+    ext = _ext()
+    torch.manual_seed(2)
+    M, N, K = 256, 128, 64
+    A = torch.randn(M, K, device=DEV).to(torch.bfloat16)
+    Bm = torch.randn(N, K, device=DEV).to(torch.bfloat16)
+    C0 = torch.randn(M, N, device=DEV).to(torch.bfloat16)
+    C = C0.clone()
+    ext.gemm_nt_bf16(A, Bm, C, None, 2)
+    ref = (C0.float() + A.float() @ Bm.float().t())
+    rel = (C.float() - ref).abs().mean() / (ref.abs().mean() + 1e-3)
+    assert rel < 0.02, rel
+    # tr (both transposed), accumulate
+    At = A.t().contiguous()
+    Bt = Bm.t().contiguous()
+    C = C0.clone()
+    ext.gemm_tr_bf16(At, Bt, C, None, 2, 1, 1)
+    rel = (C.float() - ref).abs().mean() / (ref.abs().mean() + 1e-3)
+    assert rel < 0.02, rel
+    # colsum accumulate mode
+    out0 = torch.randn(K, device=DEV, dtype=torch.float32)
+    out = out0.clone()
+    ext.colsum_bf16(A, out, 1)
+    refc = out0 + A.float().sum(dim=0)
+    assert (out - refc).abs().max() < 0.5
