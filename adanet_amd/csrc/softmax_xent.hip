@@ -18,7 +18,8 @@
 __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
     const bf16_t* __restrict__ logits, const int64_t* __restrict__ labels,
     float* __restrict__ loss, bf16_t* __restrict__ probs, int B, int C,
-    int ldl, int ldp, float eps, float* __restrict__ mean_out) {
+    int ldl, int ldp, float eps, float* __restrict__ mean_out,
+    float* __restrict__ partials) {
   const int wave_in_block = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int waves = (gridDim.x * blockDim.x) >> 6;
@@ -52,9 +53,11 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
       block_sum += l;
     }
   }
-  // Fused mean: per-block LDS reduction -> ONE global atomic per block
-  // (a per-row atomic to a single address serialized ~6x the kernel).
-  if (mean_out) {
+  // Fused mean: per-block LDS reduction, then either a plain per-block
+  // partial store (preferred: no pre-zero fill needed, a tiny reducer
+  // kernel finishes) or ONE global atomic per block (legacy path, needs a
+  // zeroed mean_out; a per-row atomic to one address serialized ~6x).
+  if (mean_out || partials) {
     __shared__ float partial[4];
     if (lane == 0) partial[wave_in_block] = block_sum;
     __syncthreads();
@@ -62,9 +65,25 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
       float s = 0.f;
       const int nwaves = blockDim.x >> 6;
       for (int w = 0; w < nwaves; ++w) s += partial[w];
-      atomicAdd(mean_out, s / B);
+      if (partials) {
+        partials[blockIdx.x] = s;  // overwrite: scratch never zeroed
+      } else {
+        atomicAdd(mean_out, s / B);
+      }
     }
   }
+}
+
+// Finishes the fused mean: mean = sum(partials) / B. One wave, overwrite
+// store — the output scalar needs no pre-zeroing (removes one fill kernel
+// per loss call from every captured train step).
+__global__ __launch_bounds__(64) void xent_mean_reduce_kernel(
+    const float* __restrict__ partials, int n, float inv_b,
+    float* __restrict__ out) {
+  float s = 0.f;
+  for (int i = threadIdx.x; i < n; i += 64) s += partials[i];
+  s = wave_reduce_sum(s);
+  if (threadIdx.x == 0) out[0] = s * inv_b;
 }
 
 __global__ __launch_bounds__(256) void softmax_xent_bwd_kernel(
@@ -110,12 +129,27 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   float* mean_ptr = (mean_out.has_value() && mean_out->defined())
                         ? mean_out->data_ptr<float>()
                         : nullptr;
+  // Fused mean via per-block partials + one-wave reducer: mean_out is
+  // overwritten, never pre-zeroed (no fill kernel per loss call).
+  at::Tensor scratch;
+  float* partials_ptr = nullptr;
+  if (fused_mean) {
+    scratch = at::empty({blocks}, logits.options().dtype(at::kFloat));
+    partials_ptr = scratch.data_ptr<float>();
+  }
   hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks), dim3(256), 0,
                      stream.stream(), (const bf16_t*)logits.data_ptr(),
                      labels.data_ptr<int64_t>(), loss_ptr,
                      (bf16_t*)probs.data_ptr(), B, C, (int)logits.stride(0),
-                     (int)probs.stride(0), (float)eps, mean_ptr);
+                     (int)probs.stride(0), (float)eps,
+                     partials_ptr ? nullptr : mean_ptr, partials_ptr);
   HIP_CHECK_KERNEL();
+  if (fused_mean) {
+    hipLaunchKernelGGL(xent_mean_reduce_kernel, dim3(1), dim3(64), 0,
+                       stream.stream(), partials_ptr, blocks, 1.f / B,
+                       mean_ptr);
+    HIP_CHECK_KERNEL();
+  }
 }
 
 void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,

@@ -52,7 +52,9 @@ class _SoftmaxXentMeanFn(torch.autograd.Function):
     def forward(ctx, logits, labels, label_smoothing):
         ext = _extension.require()
         B, C = logits.shape
-        mean = torch.zeros((), device=logits.device, dtype=torch.float32)
+        # empty, not zeros: the partials+reducer path overwrites the scalar
+        # (no per-step fill kernel in the captured graph).
+        mean = torch.empty((), device=logits.device, dtype=torch.float32)
         probs = torch.empty((B, C), device=logits.device,
                             dtype=torch.bfloat16)
         ext.softmax_xent_fwd(logits, labels, None, probs,
