@@ -76,12 +76,14 @@ class Evaluator(object):
             except StopIteration:
                 break
             for i, fn in enumerate(metric_fns):
-                sums[i] += float(fn(features, labels))
+                # Metric fns may return device tensors; keep accumulation
+                # on-device and sync once per candidate at the end.
+                sums[i] = sums[i] + fn(features, labels)
             count += 1
             step += 1
         if count == 0:
             return [float("nan")] * len(metric_fns)
-        return [s / count for s in sums]
+        return [float(s) / count for s in sums]
 
     def best_index(self, values: Sequence[float]) -> int:
         arr = np.asarray(values, dtype=np.float64)

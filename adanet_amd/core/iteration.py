@@ -754,6 +754,9 @@ class _Iteration(object):
         (reference evaluator.py:97-140: same batches for all candidates).
         Round-robin: each rank evaluates the candidates it owns; results
         are merged by adanet_losses()-style all-gather in the caller."""
+        # Accumulate per-candidate losses as DEVICE tensors: one host sync
+        # per candidate at the end instead of candidates x eval-batches
+        # blocking .item() round-trips mid-phase.
         sums = [0.0] * len(self.ensemble_specs)
         count = 0
         step = 0
@@ -777,7 +780,8 @@ class _Iteration(object):
                     loss = self._ensemble_adanet_loss(spec, frozen_out,
                                                       labels, train=False)
                     if loss is not None:
-                        sums[i] += float(loss)
+                        sums[i] = sums[i] + (loss.detach() if isinstance(
+                            loss, torch.Tensor) else loss)
             count += 1
             step += 1
         if count == 0:
@@ -787,7 +791,7 @@ class _Iteration(object):
             if spec.ensemble is None:
                 out.append(float("nan"))
             else:
-                spec.eval_loss = sums[i] / count
+                spec.eval_loss = float(sums[i]) / count
                 out.append(spec.eval_loss)
         return out
 
