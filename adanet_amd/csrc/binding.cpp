@@ -59,6 +59,8 @@ void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p,
 void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
                  const at::Tensor& seed);
 void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
+void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
+                     at::Tensor& dz, at::Tensor& db);
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred, at::Tensor& correct);
@@ -85,6 +87,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("relu_bwd", &relu_bwd);
+  m.def("relu_bwd_colsum", &relu_bwd_colsum);
   m.def("colsum_bf16", &colsum_bf16, py::arg("x"), py::arg("out"),
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);

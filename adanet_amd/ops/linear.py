@@ -148,9 +148,23 @@ class _LinearFn(torch.autograd.Function):
         need_dw = ctx.needs_input_grad[1]
         if dy.is_cuda:
             ext = _extension.require()
+            wparam, bparam = ctx.param_refs
+            bg = bparam.grad if (_DIRECT_GRAD and bparam is not None) \
+                else None
+            direct_b = (ctx.has_bias and bg is not None and bg.is_cuda
+                        and bg.dtype == torch.float32
+                        and bg.shape == (dy.shape[1],)
+                        and bg.is_contiguous())
+            db_done = False
             if ctx.activation == "relu":
                 dz = torch.empty_like(dy)
-                ext.relu_bwd(dy, y, dz)
+                if direct_b and dy.is_contiguous() and y.is_contiguous():
+                    # Fused mask + bias-grad colsum: one read of dY/Y
+                    # produces dz AND accumulates db into the arena view.
+                    ext.relu_bwd_colsum(dy, y, dz, bg)
+                    db_done = True
+                else:
+                    ext.relu_bwd(dy, y, dz)
             else:
                 dz = dy
             dx = dw = None
@@ -172,7 +186,6 @@ class _LinearFn(torch.autograd.Function):
                 else:
                     wt = transpose2d(weight)  # [K, N]
                     dx = gemm_nt(dz, wt)      # [B, K]
-            wparam, bparam = ctx.param_refs
             if need_dw:
                 # dW = dz^T @ X: both operands K_r(=batch)-major -> both
                 # transposed-staged when the batch is 32-aligned.
@@ -195,12 +208,7 @@ class _LinearFn(torch.autograd.Function):
                     ext.gemm_nt_bf16(dzt, xt, dw_out, None, epi)
                 dw = None if direct_w else dw_out
             db = None
-            if ctx.has_bias:
-                bg = bparam.grad if (_DIRECT_GRAD and bparam is not None) \
-                    else None
-                direct_b = (bg is not None and bg.is_cuda
-                            and bg.dtype == torch.float32
-                            and bg.shape == (N,) and bg.is_contiguous())
+            if ctx.has_bias and not db_done:
                 if direct_b:
                     ext.colsum_bf16(dz, bg, 1)
                 else:
