@@ -59,7 +59,7 @@ __device__ __forceinline__ void stage_tile_nt(
 // separate epilogue kernel finishes bias+ReLU+bf16 — trades a little
 // output traffic for filling all 256 CUs with the efficient big tile.
 template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32,
-          bool SPLITK = false, int WGM = 2, int WGN = 2>
+          bool SPLITK = false, int WGM = 2, int WGN = 2, int GROUPM = 0>
 __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -77,7 +77,23 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
   const int tiles = mtiles * ntiles;
   const int split = SPLITK ? (wg / tiles) : 0;
   const int tile_lin = SPLITK ? (wg % tiles) : wg;
-  const int tile_m = tile_lin / ntiles, tile_n = tile_lin % ntiles;
+  // GROUPM > 0: L2 supertile grouping — consecutive tiles walk M within a
+  // GROUPM-tall column slab, so each XCD's contiguous chunk touches a
+  // square-ish C region and re-reads A/B slabs from its own L2 instead of
+  // streaming all of B per M-band (row-major order). Bijective incl. the
+  // tail band (last band's height gm < GROUPM).
+  int tile_m, tile_n;
+  if (GROUPM > 0) {
+    const int per_band = GROUPM * ntiles;
+    const int band = tile_lin / per_band;
+    const int in_band = tile_lin % per_band;
+    const int gm = min(GROUPM, mtiles - band * GROUPM);
+    tile_m = band * GROUPM + in_band % gm;
+    tile_n = in_band / gm;
+  } else {
+    tile_m = tile_lin / ntiles;
+    tile_n = tile_lin % ntiles;
+  }
 
   static_assert(BM == WGM * FM * 16 && BN == WGN * FN * 16,
                 "tile must equal wave grid x fragments");
@@ -381,6 +397,26 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
           N, K, lda, ldb, ldc, 0, mt, nt);
       break;
     }
+#define LAUNCH_VG(BM, BN, FM, FN, MW, WGM, WGN, GM)                           \
+  do {                                                                        \
+    TORCH_CHECK(K % 32 == 0, "probe: K %% 32");                               \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL(                                                       \
+        (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN, GM>),   \
+        dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
+        nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);                          \
+  } while (0)
+    // L2 supertile grouping sweep (GROUPM column-slab order)
+    case 31: LAUNCH_VG(128, 128, 2, 2, 4, 4, 4, 2); break;
+    case 32: LAUNCH_VG(128, 128, 2, 2, 4, 4, 4, 4); break;
+    case 33: LAUNCH_VG(128, 128, 2, 2, 4, 4, 4, 8); break;
+    case 34: LAUNCH_VG(128, 128, 2, 2, 4, 4, 4, 16); break;
+    case 35: LAUNCH_VG(256, 128, 4, 4, 2, 4, 2, 2); break;
+    case 36: LAUNCH_VG(256, 128, 4, 4, 2, 4, 2, 4); break;
+    case 37: LAUNCH_VG(64, 64, 2, 2, 6, 2, 2, 4); break;
+    case 38: LAUNCH_VG(64, 64, 2, 2, 6, 2, 2, 8); break;
+    case 39: LAUNCH_VG(128, 256, 4, 4, 2, 2, 4, 2); break;
+    case 40: LAUNCH_VG(128, 256, 4, 4, 2, 2, 4, 4); break;
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V
