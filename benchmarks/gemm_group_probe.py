@@ -23,7 +23,7 @@ VARIANTS = {
     24: "128x256/8w G0",
     39: "128x256 G2",
     40: "128x256 G4",
-]
+}
 
 SHAPES = [(2048, 2048, 3072), (2048, 3072, 2048), (4096, 4096, 4096)]
 
