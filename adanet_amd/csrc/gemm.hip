@@ -264,11 +264,11 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                        b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, \
                        nt);                                                   \
   } while (0)
-#define LAUNCH_CFG_W(BM, BN, FM, FN, MW, WGM, WGN)                            \
+#define LAUNCH_CFG_W(BM, BN, FM, FN, MW, WGM, WGN, GM)                        \
   do {                                                                        \
     const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
     hipLaunchKernelGGL(                                                       \
-        (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN>),       \
+        (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN, GM>),   \
         dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
         bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                  \
   } while (0)
@@ -279,13 +279,20 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     if (M >= N && t_256x128 >= 512) {
       // 256x128, 8 waves (4x2), 64x64/wave: big-tile L2/LLC traffic with
       // 2 waves/SIMD (980 TF @4096^3, profiles/gemm_variants_r01c).
-      LAUNCH_CFG_W(256, 128, 4, 4, 2, 4, 2);
+      LAUNCH_CFG_W(256, 128, 4, 4, 2, 4, 2, 0);
     } else if (N > M && t_128x256 >= 512) {
-      LAUNCH_CFG_W(128, 256, 4, 4, 2, 2, 4);
+      LAUNCH_CFG_W(128, 256, 4, 4, 2, 2, 4, 0);
+    } else if (K > M && K > N && b128 >= 160) {
+      // K-dominant mid shapes: many small blocks beat the 128^2 tile
+      // (long K-loop at thin occupancy) — measured +5% at
+      // 2048x2048x3072 (profiles/gemm_group_probe_r01.json).
+      LAUNCH_CFG(64, 64, 2, 2, 6);
     } else if (b128 >= 160) {
-      // 128^2, 16 waves (4x4), 32x32/wave: half the 64^2 tile's LLC
-      // re-reads at full occupancy even at 1 block/CU.
-      LAUNCH_CFG_W(128, 128, 2, 2, 4, 4, 4);
+      // 128^2, 16 waves (4x4), 32x32/wave, G8 L2 supertile grouping:
+      // half the 64^2 tile's LLC re-reads at full occupancy; grouping
+      // +21% at 4096-class shapes that miss the 256-tile threshold,
+      // neutral at 2048-class (gemm_group_probe_r01.json).
+      LAUNCH_CFG_W(128, 128, 2, 2, 4, 4, 4, 8);
     } else {
       LAUNCH_CFG(64, 64, 2, 2, 6);
     }

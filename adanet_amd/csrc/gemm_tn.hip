@@ -137,7 +137,7 @@ __device__ __forceinline__ frag_ab frag_from_tr(const bf16_t* lds,
 
 // TRB: B-operand K-major (B'[K][N]); TRA additionally A K-major (A'[K][M]).
 template <int BM, int BN, int FM, int FN, int MINWAVES, int WGM, int WGN,
-          bool TRA, bool TRB>
+          bool TRA, bool TRB, int GROUPM = 0>
 __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -150,7 +150,19 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
   const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
   const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
                  (orig >> 3);
-  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  // GROUPM: L2 supertile grouping (see gemm.hip) — column-slab tile walk.
+  int tile_m, tile_n;
+  if (GROUPM > 0) {
+    const int per_band = GROUPM * ntiles;
+    const int band = wg / per_band;
+    const int in_band = wg % per_band;
+    const int gm = min(GROUPM, mtiles - band * GROUPM);
+    tile_m = band * GROUPM + in_band % gm;
+    tile_n = in_band / gm;
+  } else {
+    tile_m = wg / ntiles;
+    tile_n = wg % ntiles;
+  }
   static_assert(BM == WGM * FM * 16 && BN == WGN * FN * 16, "geometry");
   constexpr int NWAVES = WGM * WGN;
   const int tid = threadIdx.x;
@@ -290,19 +302,37 @@ void gemm_tr_probe(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   const bf16_t* a = (const bf16_t*)A.data_ptr();
   const bf16_t* b = (const bf16_t*)B.data_ptr();
   bf16_t* c = (bf16_t*)C.data_ptr();
-  const int mt = (M + 127) / 128, nt = (N + 127) / 128;
-#define L(TRA, TRB)                                                           \
-  hipLaunchKernelGGL((gemm_tr_kernel<128, 128, 2, 2, 4, 4, 4, TRA, TRB>),     \
-                     dim3(mt * nt), dim3(1024), 0, stream.stream(), a, b, c, \
-                     nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt)
-  if (trans_a && trans_b) {
-    L(true, true);
-  } else if (trans_b) {
-    L(false, true);
-  } else {
-    L(false, false);
+#define LV(BM, BN, FM, FN, MW, WGM, WGN, GM)                                  \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    if (trans_a && trans_b) {                                                 \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, true, true, GM>),     \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);                        \
+    } else if (trans_b) {                                                     \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, false, true, GM>),    \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);                        \
+    } else {                                                                  \
+      hipLaunchKernelGGL(                                                     \
+          (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, false, false, GM>),   \
+          dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,  \
+          nullptr, M, N, K, lda, ldb, ldc, 0, mt, nt);                        \
+    }                                                                         \
+  } while (0)
+  switch (variant) {
+    case 0: LV(128, 128, 2, 2, 4, 4, 4, 0); break;
+    case 1: LV(128, 128, 2, 2, 4, 4, 4, 8); break;
+    case 2: LV(64, 64, 2, 2, 6, 2, 2, 0); break;
+    case 3: LV(64, 64, 2, 2, 6, 2, 2, 8); break;
+    case 4: LV(256, 128, 4, 4, 2, 4, 2, 0); break;
+    case 5: LV(128, 256, 4, 4, 2, 2, 4, 0); break;
+    case 6: LV(128, 128, 2, 2, 4, 4, 4, 4); break;
+    default: TORCH_CHECK(false, "unknown tr variant");
   }
-#undef L
+#undef LV
   HIP_CHECK_KERNEL();
 }
 
@@ -323,6 +353,8 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   TORCH_CHECK((int)(trans_b ? B.size(1) : B.size(0)) == N, "N mismatch");
   const int lda = (int)A.stride(0), ldb = (int)B.stride(0),
             ldc = (int)C.stride(0);
+  TORCH_CHECK(trans_a || trans_b,
+              "gemm_tr: use gemm_nt_bf16 for the non-transposed case");
   TORCH_CHECK(K % 32 == 0 && lda % 8 == 0 && ldb % 8 == 0,
               "gemm_tr: fast-path alignment required "
               "(K%32, strides%8) — pad like HipLinear does");
