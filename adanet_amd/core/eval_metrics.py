@@ -30,6 +30,74 @@ class _MeanAccumulator(object):
         return self.total / self.count if self.count else float("nan")
 
 
+class AUCAccumulator(object):
+    """Streaming thresholded AUC / precision / recall for binary heads.
+
+    Mirrors tf.metrics.auc's bucketed accumulation (the reference wires
+    tf.estimator binary heads whose eval dict includes auc/auc_pr,
+    reference adanet/core/eval_metrics.py metric-op passthrough): scores
+    in [0,1] are bucketed into `num_thresholds` bins per class; suffix
+    sums give TP/FP at every threshold and the ROC is integrated by
+    trapezoid. On GPU the histogram is the native binary_histogram kernel
+    (csrc/reduce.hip); on CPU a torch bincount.
+    """
+
+    def __init__(self, num_thresholds: int = 200):
+        self.T = int(num_thresholds)
+        self._hist = None  # lazily placed [2, T] int on first update
+
+    def update(self, scores, labels):
+        import torch
+        from adanet_amd.ops import _extension
+        scores = scores.detach().reshape(-1).float()
+        labels = labels.detach().reshape(-1).long()
+        if self._hist is None:
+            self._hist = torch.zeros((2, self.T), dtype=torch.int32,
+                                     device=scores.device)
+        if scores.is_cuda:
+            ext = _extension.require()
+            ext.binary_histogram(scores.contiguous(), labels.contiguous(),
+                                 self._hist)
+        else:
+            b = (scores.clamp(0, 1) * self.T).long().clamp(max=self.T - 1)
+            pos = torch.bincount(b[labels != 0], minlength=self.T)
+            neg = torch.bincount(b[labels == 0], minlength=self.T)
+            self._hist[0] += pos.to(torch.int32)
+            self._hist[1] += neg.to(torch.int32)
+
+    def _counts(self):
+        import torch
+        h = self._hist.cpu().long()
+        pos, neg = h[0], h[1]
+        # TP/FP predicting positive at threshold = bucket lower edge
+        # (suffix-inclusive), thresholds 0..T-1 plus the all-negative end.
+        tp = torch.flip(torch.cumsum(torch.flip(pos, [0]), 0), [0])
+        fp = torch.flip(torch.cumsum(torch.flip(neg, [0]), 0), [0])
+        return tp, fp, int(pos.sum()), int(neg.sum())
+
+    def value(self) -> Dict[str, float]:
+        if self._hist is None:
+            return {"auc": float("nan"), "precision": float("nan"),
+                    "recall": float("nan")}
+        tp, fp, p, n = self._counts()
+        if p == 0 or n == 0:
+            auc = float("nan")
+        else:
+            tpr = tp.double() / p
+            fpr = fp.double() / n
+            # thresholds descend left->right along the ROC; append (0,0)
+            import torch
+            tpr = torch.cat([tpr, torch.zeros(1, dtype=torch.float64)])
+            fpr = torch.cat([fpr, torch.zeros(1, dtype=torch.float64)])
+            auc = float(torch.trapz(tpr.flip(0), fpr.flip(0)))
+        mid = self.T // 2
+        tp5, fp5 = int(tp[mid]), int(fp[mid])
+        fn5 = p - tp5
+        precision = tp5 / (tp5 + fp5) if (tp5 + fp5) else float("nan")
+        recall = tp5 / (tp5 + fn5) if (tp5 + fn5) else float("nan")
+        return {"auc": auc, "precision": precision, "recall": recall}
+
+
 class _EvalMetricsStore(object):
     """Accumulates streaming means of named metrics
     (reference eval_metrics.py:41-68)."""

@@ -62,6 +62,8 @@ void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
 void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
                      at::Tensor& dz, at::Tensor& db);
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum);
+void binary_histogram(const at::Tensor& scores, const at::Tensor& labels,
+                      at::Tensor& hist);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred, at::Tensor& correct);
 
@@ -91,4 +93,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("colsum_bf16", &colsum_bf16, py::arg("x"), py::arg("out"),
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);
+  m.def("binary_histogram", &binary_histogram);
 }

@@ -53,6 +53,52 @@ __global__ __launch_bounds__(256) void argmax_correct_kernel(
   }
 }
 
+// Score histograms for streaming binary-classification metrics (K10 tail:
+// AUC / precision / recall, reference tf.metrics.auc's thresholded
+// TP/FP/TN/FN accumulation). Buckets scores in [0,1] into T bins, one
+// histogram per class, LDS-accumulated then merged with one atomic per
+// bin per block. Suffix sums over the histograms give TP/FP at every
+// threshold (core/eval_metrics.py _AUCAccumulator).
+__global__ __launch_bounds__(256) void binary_histogram_kernel(
+    const float* __restrict__ scores, const int64_t* __restrict__ labels,
+    int* __restrict__ hist, int B, int T) {
+  extern __shared__ int lh[];  // [2][T]
+  for (int i = threadIdx.x; i < 2 * T; i += blockDim.x) lh[i] = 0;
+  __syncthreads();
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < B;
+       i += stride) {
+    float s = scores[i];
+    s = s < 0.f ? 0.f : (s > 1.f ? 1.f : s);
+    int b = (int)(s * T);
+    b = b >= T ? T - 1 : b;
+    atomicAdd(&lh[(labels[i] != 0 ? 0 : 1) * T + b], 1);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 2 * T; i += blockDim.x) {
+    if (lh[i]) atomicAdd(&hist[i], lh[i]);
+  }
+}
+
+void binary_histogram(const at::Tensor& scores, const at::Tensor& labels,
+                      at::Tensor& hist) {
+  TORCH_CHECK(scores.scalar_type() == at::kFloat && scores.is_contiguous(),
+              "binary_histogram: contiguous fp32 scores");
+  TORCH_CHECK(labels.scalar_type() == at::kLong, "binary_histogram: int64 labels");
+  TORCH_CHECK(hist.scalar_type() == at::kInt && hist.dim() == 2 &&
+                  hist.size(0) == 2,
+              "binary_histogram: int32 [2,T] hist");
+  const int B = (int)scores.numel(), T = (int)hist.size(1);
+  if (B == 0) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const int blocks = std::min((B + 255) / 256, 512);
+  hipLaunchKernelGGL(binary_histogram_kernel, dim3(blocks), dim3(256),
+                     2 * T * sizeof(int), stream.stream(),
+                     scores.data_ptr<float>(), labels.data_ptr<int64_t>(),
+                     hist.data_ptr<int>(), B, T);
+  HIP_CHECK_KERNEL();
+}
+
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum) {
   const int B = (int)x.size(0), C = (int)x.size(1);
   if (C == 0) return;

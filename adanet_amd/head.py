@@ -110,11 +110,17 @@ class BinaryClassHead(Head):
                     "class_ids": (p > 0.5).long()}
 
     def metrics(self, logits, labels):
+        from adanet_amd.core.eval_metrics import AUCAccumulator
         with torch.no_grad():
             p = torch.sigmoid(logits.float()).reshape(-1)
-            acc = float(((p > 0.5).long() == labels.long().reshape(-1)).float().mean())
-            return {"accuracy": acc,
-                    "average_loss": float(self.loss(logits, labels).cpu())}
+            y = labels.long().reshape(-1)
+            acc = float(((p > 0.5).long() == y).float().mean())
+            auc = AUCAccumulator()
+            auc.update(p, y)
+            out = {"accuracy": acc,
+                   "average_loss": float(self.loss(logits, labels).cpu())}
+            out.update(auc.value())
+            return out
 
 
 class MultiHead(Head):

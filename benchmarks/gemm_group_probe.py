@@ -25,7 +25,8 @@ VARIANTS = {
     40: "128x256 G4",
 }
 
-SHAPES = [(2048, 2048, 3072), (2048, 3072, 2048), (4096, 4096, 4096)]
+SHAPES = [(2048, 2048, 3072), (2048, 3072, 2048), (2048, 2048, 2048),
+          (4096, 4096, 4096)]
 
 
 def main():

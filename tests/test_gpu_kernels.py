@@ -469,3 +469,17 @@ def test_gemm_accumulate_epilogue():
     ext.colsum_bf16(A, out, 1)
     refc = out0 + A.float().sum(dim=0)
     assert (out - refc).abs().max() < 0.5
+
+
+def test_binary_histogram_kernel_matches_cpu():
+    from adanet_amd.core.eval_metrics import AUCAccumulator
+    torch.manual_seed(3)
+    s = torch.rand(4096)
+    y = (torch.rand(4096) > 0.5).long()
+    cpu = AUCAccumulator(200)
+    cpu.update(s, y)
+    gpu = AUCAccumulator(200)
+    gpu.update(s.to(DEV), y.to(DEV))
+    assert torch.equal(cpu._hist.cpu(), gpu._hist.cpu())
+    g, c = gpu.value(), cpu.value()
+    assert abs(g["auc"] - c["auc"]) < 1e-9

@@ -82,3 +82,35 @@ def test_torchscript_export_standalone(tmp_path, synthetic_classification):
     # live path rounds through bf16; the portable artifact is full fp32.
     assert torch.allclose(out, ref.float(), atol=1e-2), (
         (out - ref.float()).abs().max())
+
+
+def test_auc_accumulator_vs_sklearn():
+    import torch
+    from sklearn.metrics import roc_auc_score
+    from adanet_amd.core.eval_metrics import AUCAccumulator
+    torch.manual_seed(0)
+    acc = AUCAccumulator(num_thresholds=400)
+    all_s, all_y = [], []
+    for _ in range(5):  # streaming over batches
+        y = (torch.rand(512) > 0.6).long()
+        s = (0.3 * torch.randn(512) + 0.3 + 0.35 * y.float()).clamp(0, 1)
+        acc.update(s, y)
+        all_s.append(s)
+        all_y.append(y)
+    got = acc.value()
+    want = roc_auc_score(torch.cat(all_y).numpy(), torch.cat(all_s).numpy())
+    assert abs(got["auc"] - want) < 0.01, (got["auc"], want)
+    assert 0.0 <= got["precision"] <= 1.0
+    assert 0.0 <= got["recall"] <= 1.0
+
+
+def test_binary_head_metrics_include_auc():
+    import torch
+    from adanet_amd.head import BinaryClassHead
+    torch.manual_seed(1)
+    h = BinaryClassHead()
+    logits = torch.randn(256, 1)
+    labels = (torch.rand(256) > 0.5).long()
+    m = h.metrics(logits, labels)
+    for k in ("accuracy", "average_loss", "auc", "precision", "recall"):
+        assert k in m
