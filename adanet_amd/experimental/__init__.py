@@ -12,6 +12,7 @@ from adanet_amd.experimental.phases import (AllStrategy, AutoEnsemblePhase,
                                             RandomKStrategy, RepeatPhase,
                                             TrainerPhase, TunerPhase)
 from adanet_amd.experimental.schedulers import (InProcessScheduler,
+                                                MultiGpuScheduler,
                                                 Scheduler,
                                                 ThreadedScheduler)
 from adanet_amd.experimental.storages import (InMemoryStorage,
@@ -23,6 +24,6 @@ __all__ = [
     "WeightedEnsemble", "CompiledModel", "ModelSearch", "AllStrategy",
     "AutoEnsemblePhase", "GrowStrategy", "InputPhase", "Phase",
     "RandomKStrategy", "RepeatPhase", "TrainerPhase", "TunerPhase",
-    "InProcessScheduler", "Scheduler", "ThreadedScheduler", "InMemoryStorage", "ModelContainer",
+    "InProcessScheduler", "MultiGpuScheduler", "Scheduler", "ThreadedScheduler", "InMemoryStorage", "ModelContainer",
     "Storage", "TrainerWorkUnit", "WorkUnit",
 ]

@@ -31,6 +31,8 @@ class TrainerWorkUnit(WorkUnit):
         self._epochs = epochs
         self._steps_per_epoch = steps_per_epoch
         self._eval_steps = eval_steps
+        #: routing hint for MultiGpuScheduler (CompiledModel.device)
+        self.device = getattr(model, "device", None)
 
     def execute(self):
         self._model.fit(self._train, epochs=self._epochs,

@@ -156,3 +156,25 @@ def test_threaded_scheduler_repeat_phase_serial():
     ModelSearch(SequentialController(phases),
                 scheduler=ThreadedScheduler()).run()
     assert len(storage.get_models()) == 3
+
+
+def test_multigpu_scheduler_cpu_fallback_and_routing():
+    from adanet_amd.experimental.schedulers import MultiGpuScheduler
+
+    class _WU:
+        def __init__(self, device=None):
+            self.device = device
+            self.ran = 0
+
+        def execute(self):
+            self.ran += 1
+
+    # CPU: n_gpus resolves to 0 -> serial fallback
+    sched = MultiGpuScheduler()
+    units = [_WU(), _WU("cuda:1"), _WU()]
+    sched.schedule(iter(units))
+    assert all(u.ran == 1 for u in units)
+    # phased with parallel_ok falls back serially too
+    units2 = [_WU(), _WU()]
+    sched.schedule_phased([(iter(units2), True)])
+    assert all(u.ran == 1 for u in units2)
