@@ -113,6 +113,35 @@ def export_torchscript(estimator, example_features, path: str):
     return path
 
 
+def export_program(estimator, example_features, path: str,
+                   dynamic_batch: bool = True):
+    """torch.export artifact of the frozen best ensemble (.pt2).
+
+    Unlike the TorchScript trace, the ExportedProgram records a
+    functionalized ATen graph with explicit input shapes (batch
+    symbolically dynamic by default) — loadable with `torch.export.load`
+    in any adanet_amd-free environment, and the graph is the stable
+    serving/compile interface (the reference's SavedModel analog;
+    estimator.py export_saved_model_for_serving).
+    """
+    t = estimator._iteration_number
+    built, _ = estimator._rebuild_previous_ensemble(t, example_features)
+    if built is None:
+        raise ValueError("No trained ensemble to export — train() first.")
+    portable = to_portable_module(built)
+    wrapper = _TracedEnsembleWrapper(portable)
+    ex = example_features
+    if torch.is_tensor(ex):
+        ex = ex.detach().cpu().float()
+    dynamic_shapes = None
+    if dynamic_batch and torch.is_tensor(ex):
+        batch = torch.export.Dim("batch", min=1)
+        dynamic_shapes = ((batch,) + (None,) * (ex.dim() - 1),)
+    ep = torch.export.export(wrapper, (ex,), dynamic_shapes=dynamic_shapes)
+    torch.export.save(ep, path)
+    return path
+
+
 class ServableEnsemble(torch.nn.Module):
     """Frozen best ensemble + prediction helpers."""
 

@@ -114,3 +114,26 @@ def test_binary_head_metrics_include_auc():
     m = h.metrics(logits, labels)
     for k in ("accuracy", "average_loss", "auc", "precision", "recall"):
         assert k in m
+
+
+def test_export_program_roundtrip(tmp_path, synthetic_classification):
+    """torch.export artifact: loadable with torch.export.load, dynamic
+    batch honored, outputs match the live ensemble."""
+    X, Y, input_fn = synthetic_classification
+    gen = simple_dnn.Generator(layer_size=8)
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(4), subnetwork_generator=gen,
+        max_iteration_steps=10, model_dir=str(tmp_path / "m"),
+        config=adanet_amd.RunConfig(tf_random_seed=42))
+    est.train(input_fn, max_steps=20)
+    path = str(tmp_path / "ensemble.pt2")
+    serving.export_program(est, X[:8], path)
+    ep = torch.export.load(path)
+    with torch.no_grad():
+        out = ep.module()(X[:8].float())
+        out16 = ep.module()(X[:16].float())  # dynamic batch
+    live, _ = est._load_frozen_best()
+    with torch.no_grad():
+        ref = live(X[:8])
+    assert out.shape == ref.shape and out16.shape[0] == 16
+    assert torch.allclose(out, ref.float(), atol=1e-2)
