@@ -30,6 +30,13 @@ void mixer_bwd_dw(const at::Tensor& stack, const at::Tensor& dY,
                   at::Tensor& dw, int64_t vector_mode);
 void mixer_bwd_dlogits(const at::Tensor& dY, const at::Tensor& weights,
                        at::Tensor& dL, int64_t j, int64_t vector_mode);
+void mixer_fwd_direct(const std::vector<at::Tensor>& members,
+                      const at::Tensor& weights,
+                      const c10::optional<at::Tensor>& bias, at::Tensor& out,
+                      int64_t vector_mode);
+void mixer_bwd_dw_direct(const std::vector<at::Tensor>& members,
+                         const at::Tensor& dY, at::Tensor& dw,
+                         int64_t vector_mode);
 void fused_sgd(at::Tensor& master, at::Tensor& param, const at::Tensor& grad,
                const c10::optional<at::Tensor>& momentum_buf, double lr,
                double momentum, double dampening, double weight_decay,
@@ -80,6 +87,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mixer_fwd", &mixer_fwd);
   m.def("mixer_bwd_dw", &mixer_bwd_dw);
   m.def("mixer_bwd_dlogits", &mixer_bwd_dlogits);
+  m.def("mixer_fwd_direct", &mixer_fwd_direct);
+  m.def("mixer_bwd_dw_direct", &mixer_bwd_dw_direct);
   m.def("fused_sgd", &fused_sgd);
   m.def("fused_adam", &fused_adam);
   m.def("fused_sgd_fp32", &fused_sgd_fp32);

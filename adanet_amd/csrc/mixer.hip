@@ -82,8 +82,146 @@ __global__ __launch_bounds__(256) void mixer_bwd_dlogits_kernel(
   }
 }
 
+// Direct-pointer variants: members are read in place (frozen static
+// buffers + the live candidate's logits) — no per-step [J,B,C] stack copy.
+// Pointers ride in the kernel-arg struct, so hipGraph capture bakes the
+// STATIC buffer addresses (stable across replays), unlike the host-built
+// pointer-table-in-memory approach that captured stale H2D copies.
+// J > 8 chunks through repeated launches (accum=1 adds into out).
+struct MixerPtrs {
+  const bf16_t* p[8];
+  int ld[8];
+};
+
+__global__ __launch_bounds__(256) void mixer_fwd_ptrs_kernel(
+    MixerPtrs Ls, const float* __restrict__ w,
+    const float* __restrict__ bias, bf16_t* __restrict__ out, int J, int B,
+    int C, int ldo, int vector_mode, int accum) {
+  const int64_t total = (int64_t)B * C;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += (int64_t)gridDim.x * blockDim.x) {
+    const int b = (int)(p / C), c = (int)(p % C);
+    bf16_t* op = out + (int64_t)b * ldo + c;
+    float acc = accum ? bf2f(*op) : (bias ? bias[c] : 0.f);
+    for (int j = 0; j < J; ++j) {
+      const float wj = vector_mode ? w[j * C + c] : w[j];
+      acc += wj * bf2f(Ls.p[j][(int64_t)b * Ls.ld[j] + c]);
+    }
+    *op = f2bf(acc);
+  }
+}
+
+__global__ __launch_bounds__(256) void mixer_bwd_dw_scalar_ptrs_kernel(
+    MixerPtrs Ls, const bf16_t* __restrict__ dY, float* __restrict__ dw,
+    int B, int C, int ldy) {
+  const int j = blockIdx.y;
+  const bf16_t* L = Ls.p[j];
+  const int ldl = Ls.ld[j];
+  const int64_t total = (int64_t)B * C;
+  float acc = 0.f;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += (int64_t)gridDim.x * blockDim.x) {
+    const int b = (int)(p / C), c = (int)(p % C);
+    acc += bf2f(dY[(int64_t)b * ldy + c]) * bf2f(L[(int64_t)b * ldl + c]);
+  }
+  acc = wave_reduce_sum(acc);
+  __shared__ float partial[4];
+  if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    atomicAdd(&dw[j], partial[0] + partial[1] + partial[2] + partial[3]);
+  }
+}
+
+__global__ __launch_bounds__(256) void mixer_bwd_dw_vector_ptrs_kernel(
+    MixerPtrs Ls, const bf16_t* __restrict__ dY, float* __restrict__ dw,
+    int B, int C, int ldy) {
+  const int j = blockIdx.y;
+  const bf16_t* L = Ls.p[j];
+  const int ldl = Ls.ld[j];
+  for (int c = blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (int b = 0; b < B; ++b)
+      acc += bf2f(dY[(int64_t)b * ldy + c]) * bf2f(L[(int64_t)b * ldl + c]);
+    dw[j * C + c] = acc;
+  }
+}
+
 static int grid_for(int64_t total) {
   return (int)std::min<int64_t>((total + 255) / 256, 2048);
+}
+
+static MixerPtrs pack_ptrs(const std::vector<at::Tensor>& members, size_t j0,
+                           int n, int B, int C) {
+  MixerPtrs P;
+  for (int k = 0; k < n; ++k) {
+    const at::Tensor& m = members[j0 + k];
+    TORCH_CHECK(m.is_cuda() && m.scalar_type() == at::kBFloat16 &&
+                    m.dim() == 2 && (int)m.size(0) == B &&
+                    (int)m.size(1) == C && m.stride(1) == 1,
+                "mixer: member logits must be bf16 [B,C] row-strided");
+    P.p[k] = (const bf16_t*)m.data_ptr();
+    P.ld[k] = (int)m.stride(0);
+  }
+  for (int k = n; k < 8; ++k) {
+    P.p[k] = nullptr;
+    P.ld[k] = 0;
+  }
+  return P;
+}
+
+void mixer_fwd_direct(const std::vector<at::Tensor>& members,
+                      const at::Tensor& weights,
+                      const c10::optional<at::Tensor>& bias, at::Tensor& out,
+                      int64_t vector_mode) {
+  const int J = (int)members.size();
+  TORCH_CHECK(J > 0, "mixer: need members");
+  const int B = (int)members[0].size(0), C = (int)members[0].size(1);
+  if ((int64_t)B * C == 0) return;
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const float* bias_ptr =
+      (bias.has_value() && bias->defined()) ? bias->data_ptr<float>() : nullptr;
+  for (int j0 = 0; j0 < J; j0 += 8) {
+    const int n = std::min(8, J - j0);
+    MixerPtrs P = pack_ptrs(members, j0, n, B, C);
+    const float* wp = weights.data_ptr<float>() +
+                      (vector_mode ? (int64_t)j0 * C : (int64_t)j0);
+    hipLaunchKernelGGL(mixer_fwd_ptrs_kernel,
+                       dim3(grid_for((int64_t)B * C)), dim3(256), 0,
+                       stream.stream(), P, wp, bias_ptr,
+                       (bf16_t*)out.data_ptr(), n, B, C, (int)out.stride(0),
+                       (int)vector_mode, j0 > 0 ? 1 : 0);
+    HIP_CHECK_KERNEL();
+  }
+}
+
+void mixer_bwd_dw_direct(const std::vector<at::Tensor>& members,
+                         const at::Tensor& dY, at::Tensor& dw,
+                         int64_t vector_mode) {
+  const int J = (int)members.size();
+  const int B = (int)members[0].size(0), C = (int)members[0].size(1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  for (int j0 = 0; j0 < J; j0 += 8) {
+    const int n = std::min(8, J - j0);
+    MixerPtrs P = pack_ptrs(members, j0, n, B, C);
+    float* dwp = dw.data_ptr<float>() +
+                 (vector_mode ? (int64_t)j0 * C : (int64_t)j0);
+    if (vector_mode) {
+      dim3 grid((unsigned)((C + 255) / 256), (unsigned)n);
+      hipLaunchKernelGGL(mixer_bwd_dw_vector_ptrs_kernel, grid, dim3(256), 0,
+                         stream.stream(), P, (const bf16_t*)dY.data_ptr(),
+                         dwp, B, C, (int)dY.stride(0));
+    } else {
+      dim3 grid(
+          (unsigned)std::min<int64_t>(((int64_t)B * C + 2047) / 2048, 256),
+          (unsigned)n);
+      hipLaunchKernelGGL(mixer_bwd_dw_scalar_ptrs_kernel, grid, dim3(256), 0,
+                         stream.stream(), P, (const bf16_t*)dY.data_ptr(),
+                         dwp, B, C, (int)dY.stride(0));
+    }
+    HIP_CHECK_KERNEL();
+  }
 }
 
 void mixer_fwd(const at::Tensor& stack, const at::Tensor& weights,

@@ -24,27 +24,32 @@ class _MixerFn(torch.autograd.Function):
     def forward(ctx, stacked_w, bias, vector_mode, *logits):
         ext = _extension.require()
         B, C = logits[0].shape
-        # Device-side stack ([J,B,C] bf16): hipGraph-capturable, unlike a
-        # host-built pointer table whose H2D copy would bake capture-time
-        # addresses into the graph.
-        stack = torch.stack([l.detach() for l in logits]).contiguous()
+        # Direct-pointer path: the kernel reads the J member buffers in
+        # place (frozen members' static HBM-cache buffers + the live
+        # candidate's logits) — no per-step [J,B,C] stack copy. Pointers
+        # live in the kernel args, so hipGraph capture bakes the stable
+        # static-buffer addresses (unlike the old host-built pointer
+        # table whose H2D copy captured stale addresses).
+        members = [l.detach() for l in logits]
         out = torch.empty((B, C), device=logits[0].device,
                           dtype=torch.bfloat16)
-        ext.mixer_fwd(stack, stacked_w, bias, out, 1 if vector_mode else 0)
+        ext.mixer_fwd_direct(members, stacked_w, bias, out,
+                             1 if vector_mode else 0)
         ctx.vector_mode = vector_mode
         ctx.has_bias = bias is not None
         ctx.logit_requires = [t.requires_grad for t in logits]
-        ctx.save_for_backward(stacked_w, stack)
+        ctx.save_for_backward(stacked_w, *members)
         return out
 
     @staticmethod
     def backward(ctx, dy):
         ext = _extension.require()
-        stacked_w, stack = ctx.saved_tensors
-        J, B, C = stack.shape
+        stacked_w = ctx.saved_tensors[0]
+        members = list(ctx.saved_tensors[1:])
+        B, C = members[0].shape
         dy = dy.contiguous()
         dw = torch.zeros_like(stacked_w)
-        ext.mixer_bwd_dw(stack, dy, dw, 1 if ctx.vector_mode else 0)
+        ext.mixer_bwd_dw_direct(members, dy, dw, 1 if ctx.vector_mode else 0)
         dbias = None
         if ctx.has_bias:
             dbias = torch.empty((C,), device=dy.device, dtype=torch.float32)
