@@ -175,11 +175,12 @@ class _LinearFn(torch.autograd.Function):
                 # K-major directly (ds_read_b64_tr_b16 fragments) — no
                 # W^T materialization. K_r = N is always 32-padded.
                 # Measured (profiles/tr_vs_composite_r01.json): tr wins
-                # unless the reduction dim dominates both outputs dims
-                # (K_r=4096 > M,N=2048 → 0.87x), so fall back to the
-                # transpose+NT composite there.
+                # unless the reduction dim STRICTLY dominates both output
+                # dims (K_r=4096 > M,N=2048 → 0.87x); at square shapes tr
+                # still wins (tr_variants_r01.json: 427 vs ~390 effective
+                # at 2048^3), so fall back to transpose+NT only beyond.
                 if (weight.stride(0) % 8 == 0 and N % 32 == 0
-                        and N < max(B, K)):
+                        and N <= max(B, K)):
                     dx = torch.empty((B, K), device=dz.device,
                                      dtype=torch.bfloat16)
                     ext.gemm_tr_bf16(dz, weight, dx, None, 0, 0, 1)
