@@ -200,6 +200,45 @@ __global__ __launch_bounds__(256) void gemm_splitk_epilogue_kernel(
   }
 }
 
+// GEMV path for skinny-M GEMMs (M <= 8: batch-1/small-batch serving,
+// where an MFMA tile would waste 63/64 rows): one wave per output column,
+// lanes stride K with 16 B loads, wave-shuffle reduction. Bandwidth-bound
+// on the B matrix read (~10x the tiled kernel's batch-1 latency).
+struct bfx8 {
+  bf16_t v[8];
+};
+
+__global__ __launch_bounds__(256) void gemm_nt_gemv_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int lda, int ldb, int ldc, int act) {
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  const int n = blockIdx.x * 4 + wid;
+  if (n >= N) return;
+  const bf16_t* brow = B + (int64_t)n * ldb;
+  float acc[8];
+#pragma unroll
+  for (int m = 0; m < 8; ++m) acc[m] = 0.f;
+  for (int k = lane * 8; k < K; k += 64 * 8) {
+    const bfx8 bv = *(const bfx8*)(brow + k);
+    for (int m = 0; m < M; ++m) {
+      const bfx8 av = *(const bfx8*)(A + (int64_t)m * lda + k);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) acc[m] += bf2f(av.v[i]) * bf2f(bv.v[i]);
+    }
+  }
+  for (int m = 0; m < M; ++m) {
+    float v = wave_reduce_sum(acc[m]);
+    if (lane == 0) {
+      if (bias) v += bias[n];
+      if (act == 1) v = v > 0.f ? v : 0.f;
+      bf16_t* cp = &C[(int64_t)m * ldc + n];
+      if (act == 2) v += bf2f(*cp);
+      *cp = f2bf(v);
+    }
+  }
+}
+
 // Generic any-stride fallback (correctness net for shapes the fast path
 // can't take: lda/ldb not 8-aligned or K not a multiple of 32). VALU fp32.
 __global__ void gemm_nt_generic_kernel(const bf16_t* __restrict__ A,
@@ -249,6 +288,13 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   bf16_t* c = (bf16_t*)C.data_ptr();
 
   const bool fast = (K % 32 == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
+  if (fast && M <= 8) {
+    hipLaunchKernelGGL(gemm_nt_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0,
+                       stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
+                       ldc, (int)act);
+    HIP_CHECK_KERNEL();
+    return;
+  }
   if (fast) {
     // Tile dispatch tuned on MI355X (benchmarks/gemm_variants.py,
     // profiles/gemm_variants_r01.json): occupancy rules until ~4 blocks/CU,
