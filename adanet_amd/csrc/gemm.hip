@@ -470,6 +470,15 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     case 38: LAUNCH_VG(64, 64, 2, 2, 6, 2, 2, 8); break;
     case 39: LAUNCH_VG(128, 256, 4, 4, 2, 2, 4, 2); break;
     case 40: LAUNCH_VG(128, 256, 4, 4, 2, 2, 4, 4); break;
+    // mid-tile alternatives for the 2048-class bench shapes
+    case 41: LAUNCH_VG(96, 96, 3, 3, 4, 2, 2, 0); break;
+    case 42: LAUNCH_VG(96, 96, 3, 3, 4, 2, 2, 8); break;
+    case 43: LAUNCH_VG(64, 128, 2, 4, 4, 2, 2, 0); break;
+    case 44: LAUNCH_VG(128, 64, 4, 2, 4, 2, 2, 0); break;
+    case 45: LAUNCH_VG(64, 128, 2, 2, 4, 2, 4, 0); break;   // 8w 32x32/wave
+    case 46: LAUNCH_VG(96, 96, 3, 3, 2, 2, 2, 0); break;    // minwaves 2
+    case 47: LAUNCH_VG(64, 64, 2, 2, 4, 2, 2, 0); break;    // 64^2 mw4
+    case 48: LAUNCH_VG(64, 64, 2, 2, 8, 2, 2, 0); break;    // 64^2 mw8
     default: TORCH_CHECK(false, "unknown variant");
   }
 #undef LAUNCH_V
