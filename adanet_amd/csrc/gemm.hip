@@ -328,16 +328,13 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
       LAUNCH_CFG_W(256, 128, 4, 4, 2, 4, 2, 0);
     } else if (N > M && t_128x256 >= 512) {
       LAUNCH_CFG_W(128, 256, 4, 4, 2, 2, 4, 0);
-    } else if (K > M && K > N && b128 >= 160) {
-      // K-dominant mid shapes: many small blocks beat the 128^2 tile
-      // (long K-loop at thin occupancy) — measured +5% at
-      // 2048x2048x3072 (profiles/gemm_group_probe_r01.json).
-      LAUNCH_CFG(64, 64, 2, 2, 6);
     } else if (b128 >= 160) {
       // 128^2, 16 waves (4x4), 32x32/wave, G8 L2 supertile grouping:
       // half the 64^2 tile's LLC re-reads at full occupancy; grouping
       // +21% at 4096-class shapes that miss the 256-tile threshold,
-      // neutral at 2048-class (gemm_group_probe_r01.json).
+      // neutral at 2048-class (gemm_group_probe_r01.json). A 64^2
+      // K-dominant carve-out was measured BOTH ways across boxes
+      // (±5% box noise, gemm_mid_probe) — dropped for stability.
       LAUNCH_CFG_W(128, 128, 2, 2, 4, 4, 4, 8);
     } else {
       LAUNCH_CFG(64, 64, 2, 2, 6);
