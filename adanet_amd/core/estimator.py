@@ -644,6 +644,11 @@ class Estimator(object):
                         builder_hooks.extend(
                             getattr(opt, "chief_hooks", ()) or ())
                     opt = opt.optimizer
+                if opt is not None:
+                    # Candidate train steps run under direct_grad_writes
+                    # (iteration.py): single-write arenas may skip the
+                    # per-step grad memset (ops/optim.py zero_grad).
+                    opt._overwrite_grads = True
                 sub_specs.append(
                     _SubnetworkSpec(
                         name=name, builder=b, subnetwork=sub, optimizer=opt,

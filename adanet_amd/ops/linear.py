@@ -199,7 +199,11 @@ class _LinearFn(torch.autograd.Function):
                             and wg.shape == (N, K) and wg.is_contiguous())
                 dw_out = wg if direct_w else torch.empty(
                     (N, K), device=dz.device, dtype=torch.bfloat16)
-                epi = 2 if direct_w else 0
+                # overwrite epilogue when the optimizer skipped the arena
+                # zero (single-write params, ops/optim.py zero_grad).
+                epi = 0 if not direct_w else (
+                    0 if getattr(wparam, "_adanet_grad_overwrite", False)
+                    else 2)
                 if (B % 32 == 0 and dz.stride(0) % 8 == 0
                         and x.stride(0) % 8 == 0):
                     ext.gemm_tr_bf16(dz, x, dw_out, None, epi, 1, 1)
@@ -242,6 +246,10 @@ class HipLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.empty((self.padded_out, in_features), device=device,
                         dtype=dtype))
+        # exactly one dW write per backward (module used once per step in
+        # every shipped builder) — lets the optimizer skip the arena grad
+        # memset and the backward GEMM use the overwrite epilogue.
+        self.weight._adanet_single_write = True
         if bias:
             self.bias = nn.Parameter(
                 torch.zeros((self.padded_out,), device=device,

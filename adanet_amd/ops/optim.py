@@ -78,6 +78,12 @@ class _ArenaMixin(object):
                     "group": gi, "dtype": dtype, "params": params,
                     "flat_p": flat_p, "flat_g": flat_g, "views": views,
                     "state": {},
+                    # every param single-write (HipLinear weights): with
+                    # overwrite-mode direct grad writes the per-step arena
+                    # zero fill can be skipped entirely.
+                    "all_single_write": all(
+                        getattr(p, "_adanet_single_write", False)
+                        for p in params),
                 })
         self._arenas = arenas
 
@@ -105,8 +111,16 @@ class _ArenaMixin(object):
     def zero_grad(self, set_to_none: bool = True):
         if self._arenas:
             for a in self._arenas:
-                a["flat_g"].zero_()
+                # Overwrite-mode (engine opt-in via _overwrite_grads, only
+                # when every param in the arena is written exactly once
+                # per backward by a direct-grad producer): skip the memset
+                # and tell producers to use the overwrite epilogue.
+                skip = (getattr(self, "_overwrite_grads", False)
+                        and a["all_single_write"])
+                if not skip:
+                    a["flat_g"].zero_()
                 for p, off, k in a["views"]:
+                    p._adanet_grad_overwrite = skip
                     if p.grad is None or p.grad.data_ptr() != a[
                             "flat_g"][off:off + k].data_ptr():
                         p.grad = a["flat_g"][off:off + k].view_as(p.data)

@@ -499,3 +499,40 @@ def test_gemm_gemv_path_small_m(M):
     ref = torch.relu(A.float() @ Bm.float().t() + bias)
     rel = (C.float() - ref).abs().mean() / (ref.abs().mean() + 1e-3)
     assert rel < 0.02, rel
+
+
+def test_overwrite_grad_mode_matches_standard():
+    """_overwrite_grads (skip arena memset + overwrite epilogue) must
+    train identically to the standard accumulate path."""
+    from adanet_amd.ops.linear import HipLinear, direct_grad_writes
+    from adanet_amd.ops.optim import FusedSGD
+    torch.manual_seed(5)
+
+    def build():
+        torch.manual_seed(5)
+        return torch.nn.Sequential(
+            HipLinear(64, 96, activation="relu"),
+            HipLinear(96, 32),
+        ).to(DEV)
+
+    outs = []
+    for overwrite in (False, True):
+        model = build()
+        opt = FusedSGD(model.parameters(), lr=0.05, momentum=0.9)
+        if overwrite:
+            opt._overwrite_grads = True
+        torch.manual_seed(7)
+        for _ in range(4):
+            x = torch.randn(128, 64, device=DEV).to(torch.bfloat16)
+            opt.zero_grad(set_to_none=True)
+            with direct_grad_writes():
+                model(x).float().pow(2).mean().backward()
+            opt.step()
+        outs.append([p.detach().float().clone() for p in model.parameters()])
+        if overwrite:
+            assert any(a["all_single_write"] is False or True
+                       for a in opt._arenas)  # arenas engaged
+            assert any(a["all_single_write"] for a in opt._arenas), \
+                "weight arena should be single-write"
+    for a, b in zip(*outs):
+        assert torch.equal(a, b), (a - b).abs().max()
