@@ -14,6 +14,9 @@ void gemm_tr_probe(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
 void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                   const c10::optional<at::Tensor>& bias, int64_t act,
                   int64_t trans_a, int64_t trans_b);
+void gemm_tr_batched(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+                     const c10::optional<at::Tensor>& bias, int64_t act,
+                     int64_t trans_a, int64_t trans_b);
 void transpose_bf16(const at::Tensor& in, at::Tensor& out);
 void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
                       const c10::optional<at::Tensor>& loss,
@@ -81,6 +84,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("probe_tr16_layout", &probe_tr16_layout);
   m.def("gemm_tr_probe", &gemm_tr_probe);
   m.def("gemm_tr_bf16", &gemm_tr_bf16);
+  m.def("gemm_tr_batched", &gemm_tr_batched);
   m.def("transpose_bf16", &transpose_bf16, "bf16 2-D transpose");
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);

@@ -141,10 +141,16 @@ template <int BM, int BN, int FM, int FN, int MINWAVES, int WGM, int WGN,
 __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
-    int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles) {
+    int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles,
+    int64_t stride_a = 0, int64_t stride_b = 0, int64_t stride_c = 0) {
   constexpr int KSTEP = 32;
   __shared__ bf16_t As[2][BM * KSTEP];
   __shared__ bf16_t Bs[2][BN * KSTEP];
+  // Batched mode (pointwise conv: one GEMM per image): grid.z selects the
+  // batch element; zero strides share the operand (the weight matrix).
+  A += (int64_t)blockIdx.z * stride_a;
+  B += (int64_t)blockIdx.z * stride_b;
+  C += (int64_t)blockIdx.z * stride_c;
   const int nwg = mtiles * ntiles;
   const int orig = blockIdx.x;
   const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
@@ -405,5 +411,72 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     LTR(64, 64, 2, 2, 6, 2, 2);
   }
 #undef LTR
+  HIP_CHECK_KERNEL();
+}
+
+// Batched transposed-operand GEMM: one GEMM per grid.z element (the
+// pointwise-conv workhorse — K9: y_b[Co,HW] = W[Co,Ci] @ x_b[Ci,HW] is
+// trans_b staging per image, dX_b = W^T @ dz_b is tt). 2-D operands are
+// shared across the batch (stride 0); 3-D operands advance by stride(0).
+void gemm_tr_batched(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+                     const c10::optional<at::Tensor>& bias, int64_t act,
+                     int64_t trans_a, int64_t trans_b) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda() && C.is_cuda(), "gemm_trb: GPU only");
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
+                  B.scalar_type() == at::kBFloat16,
+              "gemm_trb: bf16 required");
+  TORCH_CHECK(C.dim() == 3, "gemm_trb: C must be [Z,M,N]");
+  const int Z = (int)C.size(0);
+  const int M = (int)C.size(1), N = (int)C.size(2);
+  auto dims2 = [](const at::Tensor& t) {
+    return std::make_pair((int)t.size(t.dim() - 2), (int)t.size(t.dim() - 1));
+  };
+  auto a2 = dims2(A), b2 = dims2(B);
+  const int K = trans_a ? a2.first : a2.second;
+  TORCH_CHECK((trans_b ? b2.second : b2.first) == N, "gemm_trb: N mismatch");
+  TORCH_CHECK((trans_b ? b2.first : b2.second) == K, "gemm_trb: K mismatch");
+  TORCH_CHECK((trans_a ? a2.second : a2.first) == M, "gemm_trb: M mismatch");
+  const int lda = (int)A.stride(A.dim() - 2), ldb = (int)B.stride(B.dim() - 2),
+            ldc = (int)C.stride(1);
+  const int64_t sa = A.dim() == 3 ? A.stride(0) : 0;
+  const int64_t sb = B.dim() == 3 ? B.stride(0) : 0;
+  const int64_t sc = C.stride(0);
+  TORCH_CHECK(K % 32 == 0 && lda % 8 == 0 && ldb % 8 == 0,
+              "gemm_trb: fast-path alignment required (K%32, strides%8)");
+  if (Z == 0 || M == 0 || N == 0) return;
+  const float* bias_ptr = nullptr;
+  if (bias.has_value() && bias->defined()) bias_ptr = bias->data_ptr<float>();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+#define LTRB(BM, BN, FM, FN, MW, WGM, WGN, TRA, TRB)                          \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL(                                                       \
+        (gemm_tr_kernel<BM, BN, FM, FN, MW, WGM, WGN, TRA, TRB>),             \
+        dim3(mt * nt, 1, Z), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, \
+        c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt, sa, sb, sc);   \
+  } while (0)
+#define LTRB_DISPATCH(TRA, TRB)                                               \
+  do {                                                                        \
+    const int64_t b128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);      \
+    if (b128 >= 64) {                                                         \
+      LTRB(128, 128, 2, 2, 4, 4, 4, TRA, TRB);                                \
+    } else {                                                                  \
+      LTRB(64, 64, 2, 2, 6, 2, 2, TRA, TRB);                                  \
+    }                                                                         \
+  } while (0)
+  if (trans_a && trans_b) {
+    LTRB_DISPATCH(true, true);
+  } else if (trans_b) {
+    LTRB_DISPATCH(false, true);
+  } else if (trans_a) {
+    LTRB_DISPATCH(true, false);
+  } else {
+    LTRB_DISPATCH(false, false);
+  }
+#undef LTRB
+#undef LTRB_DISPATCH
   HIP_CHECK_KERNEL();
 }

@@ -47,17 +47,20 @@ class SepConv(nn.Module):
 
     def __init__(self, c_in, c_out, kernel, stride):
         super().__init__()
+        from adanet_amd.ops.conv import HipConv1x1
         pad = kernel // 2
+        # pointwise convs run on the batched MFMA GEMM when channels align
+        # (ops/conv.py); depthwise stays on MIOpen (grouped conv).
         self.op = nn.Sequential(
             nn.ReLU(),
             nn.Conv2d(c_in, c_in, kernel, stride=stride, padding=pad,
                       groups=c_in, bias=False),
-            nn.Conv2d(c_in, c_out, 1, bias=False),
+            HipConv1x1(c_in, c_out, bias=False),
             BNfp32(c_out),
             nn.ReLU(),
             nn.Conv2d(c_out, c_out, kernel, stride=1, padding=pad,
                       groups=c_out, bias=False),
-            nn.Conv2d(c_out, c_out, 1, bias=False),
+            HipConv1x1(c_out, c_out, bias=False),
             BNfp32(c_out),
         )
 
@@ -69,10 +72,14 @@ class ReluConvBN(nn.Module):
 
     def __init__(self, c_in, c_out, kernel=1, stride=1):
         super().__init__()
+        from adanet_amd.ops.conv import HipConv1x1
+        conv = (HipConv1x1(c_in, c_out, bias=False)
+                if kernel == 1 and stride == 1 else
+                nn.Conv2d(c_in, c_out, kernel, stride=stride,
+                          padding=kernel // 2, bias=False))
         self.op = nn.Sequential(
             nn.ReLU(),
-            nn.Conv2d(c_in, c_out, kernel, stride=stride,
-                      padding=kernel // 2, bias=False),
+            conv,
             BNfp32(c_out),
         )
 

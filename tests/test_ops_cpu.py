@@ -147,3 +147,14 @@ def test_dropout_validation_and_eval_mode():
     d.eval()
     x = torch.randn(4, 4)
     assert torch.equal(d(x), x)
+
+
+def test_hip_conv1x1_cpu_fallback():
+    import torch
+    from adanet_amd.ops.conv import HipConv1x1
+    m = HipConv1x1(8, 12)
+    x = torch.randn(2, 8, 5, 5)
+    y = m(x)
+    ref = torch.nn.functional.conv2d(
+        x, m.weight, m.bias.to(x.dtype))
+    assert torch.allclose(y, ref, atol=1e-5)
