@@ -535,7 +535,11 @@ def test_overwrite_grad_mode_matches_standard():
             assert any(a["all_single_write"] for a in opt._arenas), \
                 "weight arena should be single-write"
     for a, b in zip(*outs):
-        assert torch.equal(a, b), (a - b).abs().max()
+        # not bit-exact: the bias-grad colsum accumulates via fp32 atomics
+        # whose order varies run-to-run (within EITHER mode); compare with
+        # a tight tolerance instead.
+        rel = (a - b).abs().max() / (b.abs().max() + 1e-6)
+        assert rel < 5e-3, rel
 
 
 def test_conv1x1_native_fwd_bwd_vs_fp32():
