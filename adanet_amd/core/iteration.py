@@ -258,11 +258,23 @@ class _Iteration(object):
         keep_last = self._need_frozen_last_layer()
         if cacheable and key in self._frozen_cache:
             cached = self._frozen_cache[key]
+            srcs, dsts = [], []
             for name, (last_c, logits_c) in cached.items():
                 last_s, logits_s = self._frozen_static[name]
-                logits_s.copy_(logits_c)
+                srcs.append(logits_c)
+                dsts.append(logits_s)
                 if keep_last and last_c is not None:
-                    last_s.copy_(last_c)
+                    srcs.append(last_c)
+                    dsts.append(last_s)
+            if srcs and srcs[0].is_cuda and all(
+                    s.dtype == torch.bfloat16 and s.is_contiguous()
+                    and d.is_contiguous() for s, d in zip(srcs, dsts)):
+                # One batched-copy launch instead of J-1 copyBuffer calls.
+                from adanet_amd.ops import _extension
+                _extension.require().multi_copy_bf16(srcs, dsts)
+            else:
+                for s, d in zip(srcs, dsts):
+                    d.copy_(s)
             return self._frozen_static
         fresh = self.compute_frozen_outputs(features, training=training)
         if self._frozen_static is None or any(

@@ -71,6 +71,8 @@ void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
 void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
 void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
                      at::Tensor& dz, at::Tensor& db);
+void multi_copy_bf16(const std::vector<at::Tensor>& srcs,
+                     const std::vector<at::Tensor>& dsts);
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum);
 void binary_histogram(const at::Tensor& scores, const at::Tensor& labels,
                       at::Tensor& hist);
@@ -103,6 +105,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &dropout_bwd);
   m.def("relu_bwd", &relu_bwd);
   m.def("relu_bwd_colsum", &relu_bwd_colsum);
+  m.def("multi_copy_bf16", &multi_copy_bf16);
   m.def("colsum_bf16", &colsum_bf16, py::arg("x"), py::arg("out"),
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);

@@ -582,3 +582,15 @@ def test_hip_conv1x1_module_native_and_fallback():
     y2 = m2(x2)
     ref2 = torch.nn.functional.conv2d(x2.float(), m2.weight.float())
     assert (y2.float() - ref2).abs().mean() / (ref2.abs().mean() + 1e-3) < 0.03
+
+
+def test_multi_copy_bf16():
+    ext = _ext()
+    torch.manual_seed(8)
+    srcs = [torch.randn(37 + 100 * i, device=DEV).to(torch.bfloat16)
+            for i in range(10)]
+    dsts = [torch.zeros_like(s) for s in srcs]
+    ext.multi_copy_bf16(srcs, dsts)
+    torch.cuda.synchronize()
+    for s, d in zip(srcs, dsts):
+        assert torch.equal(s, d)
