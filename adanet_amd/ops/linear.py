@@ -159,7 +159,8 @@ class _LinearFn(torch.autograd.Function):
             db_done = False
             if ctx.activation == "relu":
                 dz = torch.empty_like(dy)
-                if direct_b and dy.is_contiguous() and y.is_contiguous():
+                if (direct_b and dy.is_contiguous() and y.is_contiguous()
+                        and dy.shape[1] % 8 == 0):
                     # Fused mask + bias-grad colsum: one read of dY/Y
                     # produces dz AND accumulates db into the arena view.
                     ext.relu_bwd_colsum(dy, y, dz, bg)
