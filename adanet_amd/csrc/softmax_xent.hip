@@ -201,7 +201,11 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   // overwritten, never pre-zeroed (no fill kernel per loss call).
   at::Tensor scratch;
   float* partials_ptr = nullptr;
-  const bool small_c = C <= 32;
+  static const bool small_enabled = [] {
+    const char* e = getenv("ADANET_XENT_SMALL");
+    return !(e && e[0] == '0');
+  }();
+  const bool small_c = C <= 32 && small_enabled;
   const int sblocks = std::min((B + 255) / 256, small_c ? 256 : blocks);
   if (fused_mean) {
     scratch = at::empty({small_c ? sblocks : blocks},
@@ -248,7 +252,11 @@ void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
                         ? grad_scalar->data_ptr<float>()
                         : nullptr;
   TORCH_CHECK(gr || gs, "xent bwd: need grad_rows or grad_scalar");
-  if (C <= 32) {
+  static const bool small_enabled = [] {
+    const char* e = getenv("ADANET_XENT_SMALL");
+    return !(e && e[0] == '0');
+  }();
+  if (C <= 32 && small_enabled) {
     const int sblocks = std::min((B + 255) / 256, 2048);
     hipLaunchKernelGGL(softmax_xent_bwd_small_kernel, dim3(sblocks),
                        dim3(256), 0, stream.stream(),
