@@ -59,50 +59,22 @@ __global__ __launch_bounds__(256) void relu_bwd_kernel(
 // Layout mirrors colsum: thread-per-column (coalesced), row-chunk grid.y
 // fills the chip; db is accumulated via one atomicAdd per column per chunk
 // (direct-to-arena accum semantics — db never pre-zeroed here).
-// Each thread owns 8 consecutive columns (16 B vector loads/stores per
-// row — the scalar 2 B/thread version measured only 1.45 TB/s) and keeps 8
-// fp32 column partials in registers; one atomicAdd per column per block.
-typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_e;
-
 __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
     bf16_t* __restrict__ dz, float* __restrict__ db, int B, int C,
     int rows_per_block) {
-  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  if (c0 >= C) return;
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
   const int r0 = blockIdx.y * rows_per_block;
   const int r1 = min(B, r0 + rows_per_block);
-  float acc[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
-  if (c0 + 8 <= C) {
-    for (int b = r0; b < r1; ++b) {
-      const int64_t i = (int64_t)b * C + c0;
-      union { u32x4_e v; bf16_t h[8]; } dv, yv, zv;
-      dv.v = *(const u32x4_e*)(dy + i);
-      yv.v = *(const u32x4_e*)(y + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float g = bf2f(yv.h[j]) > 0.f ? bf2f(dv.h[j]) : 0.f;
-        zv.h[j] = f2bf(g);
-        acc[j] += g;
-      }
-      *(u32x4_e*)(dz + i) = zv.v;
-    }
-  } else {
-    for (int b = r0; b < r1; ++b) {
-      for (int j = 0; c0 + j < C; ++j) {
-        const int64_t i = (int64_t)b * C + c0 + j;
-        const float g = bf2f(y[i]) > 0.f ? bf2f(dy[i]) : 0.f;
-        dz[i] = f2bf(g);
-        acc[j] += g;
-      }
-    }
+  float acc = 0.f;
+  for (int b = r0; b < r1; ++b) {
+    const int64_t i = (int64_t)b * C + c;
+    const float g = bf2f(y[i]) > 0.f ? bf2f(dy[i]) : 0.f;
+    dz[i] = f2bf(g);
+    acc += g;
   }
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    if (c0 + j < C && acc[j] != 0.f) atomicAdd(&db[c0 + j], acc[j]);
-  }
+  atomicAdd(&db[c], acc);
 }
 
 // Batched device-to-device copy: up to 8 (src,dst) pairs per launch
@@ -212,10 +184,9 @@ void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
               "relu_bwd_colsum: contiguous tensors required");
   TORCH_CHECK(db.scalar_type() == at::kFloat, "relu_bwd_colsum: fp32 db");
   const int B = (int)dy.size(0), C = (int)dy.size(1);
-  TORCH_CHECK(C % 8 == 0, "relu_bwd_colsum: C % 8 (padded heads)");
   if (B == 0 || C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
-  const int stripes = (C + 2047) / 2048;  // 8 cols per thread
+  const int stripes = (C + 255) / 256;
   const int row_chunks =
       std::max(1, std::min(1024 / stripes, (B + 7) / 8));
   const int rows_per_block = (B + row_chunks - 1) / row_chunks;

@@ -74,74 +74,6 @@ __global__ __launch_bounds__(256) void softmax_xent_fwd_kernel(
   }
 }
 
-// Small-C specialization (C <= 32, the CIFAR-class-count regime): one
-// THREAD per row with a serial register loop over classes — the wave-per-
-// row kernel leaves 64-C lanes idle and serializes 3 wave reductions per
-// row. Here all 64 lanes process 64 rows concurrently, no shuffles.
-__global__ __launch_bounds__(256) void softmax_xent_fwd_small_kernel(
-    const bf16_t* __restrict__ logits, const int64_t* __restrict__ labels,
-    float* __restrict__ loss, bf16_t* __restrict__ probs, int B, int C,
-    int ldl, int ldp, float eps, float* __restrict__ partials) {
-  float acc = 0.f;
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < B;
-       row += stride) {
-    const bf16_t* lrow = logits + row * ldl;
-    const int y = (int)labels[row];
-    // three register-only passes over the (L2-hot, <=64 B) row — a private
-    // v[C] array would spill to scratch under the dynamic C loop.
-    float mx = -INFINITY;
-    for (int c = 0; c < C; ++c) mx = fmaxf(mx, bf2f(lrow[c]));
-    float se = 0.f, sum_logits = 0.f, ly = 0.f;
-    for (int c = 0; c < C; ++c) {
-      const float v = bf2f(lrow[c]);
-      se += __expf(v - mx);
-      sum_logits += v;
-      if (c == y) ly = v;
-    }
-    const float lse = __logf(se) + mx;
-    const float inv_se = 1.f / se;
-    if (probs) {
-      bf16_t* prow = probs + row * ldp;
-      for (int c = 0; c < C; ++c)
-        prow[c] = f2bf(__expf(bf2f(lrow[c]) - mx) * inv_se);
-    }
-    const float l = lse - (1.f - eps) * ly - (eps / C) * sum_logits;
-    if (loss) loss[row] = l;
-    acc += l;
-  }
-  if (partials) {
-    // block-sum -> one partial per block (reduced by the mean kernel).
-    __shared__ float sm[256];
-    sm[threadIdx.x] = acc;
-    __syncthreads();
-    for (int s = 128; s > 0; s >>= 1) {
-      if (threadIdx.x < s) sm[threadIdx.x] += sm[threadIdx.x + s];
-      __syncthreads();
-    }
-    if (threadIdx.x == 0) partials[blockIdx.x] = sm[0];
-  }
-}
-
-__global__ __launch_bounds__(256) void softmax_xent_bwd_small_kernel(
-    const bf16_t* __restrict__ probs, const int64_t* __restrict__ labels,
-    const float* __restrict__ grad_rows, bf16_t* __restrict__ dlogits, int B,
-    int C, int ldp, int ldd, float eps,
-    const float* __restrict__ grad_scalar) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < B;
-       row += stride) {
-    const int64_t y = labels[row];
-    const float g = grad_scalar ? (*grad_scalar) / B : grad_rows[row];
-    const bf16_t* prow = probs + row * ldp;
-    bf16_t* drow = dlogits + row * ldd;
-    for (int c = 0; c < C; ++c) {
-      const float t = (c == (int)y ? 1.f - eps : 0.f) + eps / C;
-      drow[c] = f2bf((bf2f(prow[c]) - t) * g);
-    }
-  }
-}
-
 // Finishes the fused mean: mean = sum(partials) / B. One wave, overwrite
 // store — the output scalar needs no pre-zeroing (removes one fill kernel
 // per loss call from every captured train step).
@@ -201,38 +133,21 @@ void softmax_xent_fwd(const at::Tensor& logits, const at::Tensor& labels,
   // overwritten, never pre-zeroed (no fill kernel per loss call).
   at::Tensor scratch;
   float* partials_ptr = nullptr;
-  static const bool small_enabled = [] {
-    const char* e = getenv("ADANET_XENT_SMALL");
-    return !(e && e[0] == '0');
-  }();
-  const bool small_c = C <= 32 && small_enabled;
-  const int sblocks = std::min((B + 255) / 256, small_c ? 256 : blocks);
   if (fused_mean) {
-    scratch = at::empty({small_c ? sblocks : blocks},
-                        logits.options().dtype(at::kFloat));
+    scratch = at::empty({blocks}, logits.options().dtype(at::kFloat));
     partials_ptr = scratch.data_ptr<float>();
   }
-  if (small_c) {
-    hipLaunchKernelGGL(softmax_xent_fwd_small_kernel, dim3(sblocks),
-                       dim3(256), 0, stream.stream(),
-                       (const bf16_t*)logits.data_ptr(),
-                       labels.data_ptr<int64_t>(), loss_ptr,
-                       (bf16_t*)probs.data_ptr(), B, C,
-                       (int)logits.stride(0), (int)probs.stride(0),
-                       (float)eps, partials_ptr);
-  } else {
-    hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks), dim3(256), 0,
-                       stream.stream(), (const bf16_t*)logits.data_ptr(),
-                       labels.data_ptr<int64_t>(), loss_ptr,
-                       (bf16_t*)probs.data_ptr(), B, C, (int)logits.stride(0),
-                       (int)probs.stride(0), (float)eps,
-                       partials_ptr ? nullptr : mean_ptr, partials_ptr);
-  }
+  hipLaunchKernelGGL(softmax_xent_fwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)logits.data_ptr(),
+                     labels.data_ptr<int64_t>(), loss_ptr,
+                     (bf16_t*)probs.data_ptr(), B, C, (int)logits.stride(0),
+                     (int)probs.stride(0), (float)eps,
+                     partials_ptr ? nullptr : mean_ptr, partials_ptr);
   HIP_CHECK_KERNEL();
   if (fused_mean) {
     hipLaunchKernelGGL(xent_mean_reduce_kernel, dim3(1), dim3(64), 0,
-                       stream.stream(), partials_ptr,
-                       small_c ? sblocks : blocks, 1.f / B, mean_ptr);
+                       stream.stream(), partials_ptr, blocks, 1.f / B,
+                       mean_ptr);
     HIP_CHECK_KERNEL();
   }
 }
@@ -252,25 +167,10 @@ void softmax_xent_bwd(const at::Tensor& probs, const at::Tensor& labels,
                         ? grad_scalar->data_ptr<float>()
                         : nullptr;
   TORCH_CHECK(gr || gs, "xent bwd: need grad_rows or grad_scalar");
-  static const bool small_enabled = [] {
-    const char* e = getenv("ADANET_XENT_SMALL");
-    return !(e && e[0] == '0');
-  }();
-  if (C <= 32 && small_enabled) {
-    const int sblocks = std::min((B + 255) / 256, 2048);
-    hipLaunchKernelGGL(softmax_xent_bwd_small_kernel, dim3(sblocks),
-                       dim3(256), 0, stream.stream(),
-                       (const bf16_t*)probs.data_ptr(),
-                       labels.data_ptr<int64_t>(), gr,
-                       (bf16_t*)dlogits.data_ptr(), B, C,
-                       (int)probs.stride(0), (int)dlogits.stride(0),
-                       (float)eps, gs);
-  } else {
-    hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
-                       stream.stream(), (const bf16_t*)probs.data_ptr(),
-                       labels.data_ptr<int64_t>(), gr,
-                       (bf16_t*)dlogits.data_ptr(), B, C, (int)probs.stride(0),
-                       (int)dlogits.stride(0), (float)eps, gs);
-  }
+  hipLaunchKernelGGL(softmax_xent_bwd_kernel, dim3(blocks), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)probs.data_ptr(),
+                     labels.data_ptr<int64_t>(), gr,
+                     (bf16_t*)dlogits.data_ptr(), B, C, (int)probs.stride(0),
+                     (int)dlogits.stride(0), (float)eps, gs);
   HIP_CHECK_KERNEL();
 }

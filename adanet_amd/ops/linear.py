@@ -24,7 +24,6 @@ Reference dependency being replaced: tf.layers.dense in subnetworks
 from __future__ import annotations
 
 import math
-import os
 from typing import Optional
 
 import torch
@@ -160,10 +159,7 @@ class _LinearFn(torch.autograd.Function):
             db_done = False
             if ctx.activation == "relu":
                 dz = torch.empty_like(dy)
-                if (direct_b and dy.is_contiguous() and y.is_contiguous()
-                        and dy.shape[1] % 8 == 0
-                        and os.environ.get("ADANET_FUSED_RELU_COLSUM",
-                                           "1") != "0"):
+                if direct_b and dy.is_contiguous() and y.is_contiguous():
                     # Fused mask + bias-grad colsum: one read of dY/Y
                     # produces dz AND accumulates db into the arena view.
                     ext.relu_bwd_colsum(dy, y, dz, bg)
