@@ -76,6 +76,12 @@ void multi_copy_bf16(const std::vector<at::Tensor>& srcs,
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum);
 void binary_histogram(const at::Tensor& scores, const at::Tensor& labels,
                       at::Tensor& hist);
+void depthwise_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
+                   int64_t stride, int64_t pad);
+void depthwise_bwd_dx(const at::Tensor& dy, const at::Tensor& w,
+                      at::Tensor& dx, int64_t stride, int64_t pad);
+void depthwise_bwd_dw(const at::Tensor& x, const at::Tensor& dy,
+                      at::Tensor& dw, int64_t stride, int64_t pad);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred, at::Tensor& correct);
 
@@ -110,4 +116,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);
   m.def("binary_histogram", &binary_histogram);
+  m.def("depthwise_fwd", &depthwise_fwd);
+  m.def("depthwise_bwd_dx", &depthwise_bwd_dx);
+  m.def("depthwise_bwd_dw", &depthwise_bwd_dw);
 }

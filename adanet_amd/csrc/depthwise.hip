@@ -1,0 +1,154 @@
+// Depthwise 2-D convolution (K9): the other half of the NASNet separable
+// conv (reference nasnet_utils.py:182 _stacked_separable_conv = depthwise
+// KxK then pointwise 1x1; the pointwise half runs on gemm_tr_batched).
+// Depthwise conv has no reuse across channels, so it is bandwidth-bound
+// elementwise work: one thread per output pixel, serial tap loop (9/25
+// taps), coalesced along the contiguous W dimension. Works for any channel
+// count (no GEMM alignment constraints) at stride 1 or 2.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+__global__ __launch_bounds__(256) void depthwise_fwd_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ y, int B, int C, int H, int W, int OH, int OW,
+    int KS, int stride, int pad) {
+  const int64_t total = (int64_t)B * C * OH * OW;
+  const int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += gstride) {
+    const int ow = (int)(p % OW);
+    const int oh = (int)((p / OW) % OH);
+    const int c = (int)((p / ((int64_t)OW * OH)) % C);
+    const int b = (int)(p / ((int64_t)OW * OH * C));
+    const bf16_t* xp = x + ((int64_t)b * C + c) * H * W;
+    const bf16_t* wp = w + c * KS * KS;
+    float acc = 0.f;
+    for (int kh = 0; kh < KS; ++kh) {
+      const int ih = oh * stride + kh - pad;
+      if (ih < 0 || ih >= H) continue;
+      for (int kw = 0; kw < KS; ++kw) {
+        const int iw = ow * stride + kw - pad;
+        if (iw < 0 || iw >= W) continue;
+        acc += bf2f(wp[kh * KS + kw]) * bf2f(xp[ih * W + iw]);
+      }
+    }
+    y[p] = f2bf(acc);
+  }
+}
+
+// dX: full correlation with the flipped kernel, honoring stride divisibility.
+__global__ __launch_bounds__(256) void depthwise_bwd_dx_kernel(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ dx, int B, int C, int H, int W, int OH, int OW,
+    int KS, int stride, int pad) {
+  const int64_t total = (int64_t)B * C * H * W;
+  const int64_t gstride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += gstride) {
+    const int iw = (int)(p % W);
+    const int ih = (int)((p / W) % H);
+    const int c = (int)((p / ((int64_t)W * H)) % C);
+    const int b = (int)(p / ((int64_t)W * H * C));
+    const bf16_t* dyp = dy + ((int64_t)b * C + c) * OH * OW;
+    const bf16_t* wp = w + c * KS * KS;
+    float acc = 0.f;
+    for (int kh = 0; kh < KS; ++kh) {
+      const int num_h = ih + pad - kh;
+      if (num_h < 0 || num_h % stride) continue;
+      const int oh = num_h / stride;
+      if (oh >= OH) continue;
+      for (int kw = 0; kw < KS; ++kw) {
+        const int num_w = iw + pad - kw;
+        if (num_w < 0 || num_w % stride) continue;
+        const int ow = num_w / stride;
+        if (ow >= OW) continue;
+        acc += bf2f(wp[kh * KS + kw]) * bf2f(dyp[oh * OW + ow]);
+      }
+    }
+    dx[p] = f2bf(acc);
+  }
+}
+
+// dW: one block per (channel, tap); block-parallel reduction over b,oh,ow.
+__global__ __launch_bounds__(256) void depthwise_bwd_dw_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
+    float* __restrict__ dw, int B, int C, int H, int W, int OH, int OW,
+    int KS, int stride, int pad) {
+  const int c = blockIdx.x;
+  const int kh = blockIdx.y / KS, kw = blockIdx.y % KS;
+  const int64_t total = (int64_t)B * OH * OW;
+  float acc = 0.f;
+  for (int64_t p = threadIdx.x; p < total; p += blockDim.x) {
+    const int ow = (int)(p % OW);
+    const int oh = (int)((p / OW) % OH);
+    const int b = (int)(p / ((int64_t)OW * OH));
+    const int ih = oh * stride + kh - pad;
+    const int iw = ow * stride + kw - pad;
+    if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+    acc += bf2f(dy[((int64_t)b * C + c) * OH * OW + oh * OW + ow]) *
+           bf2f(x[((int64_t)b * C + c) * H * W + ih * W + iw]);
+  }
+  acc = wave_reduce_sum(acc);
+  __shared__ float partial[4];
+  if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    dw[c * KS * KS + kh * KS + kw] =
+        partial[0] + partial[1] + partial[2] + partial[3];
+  }
+}
+
+static int dw_grid(int64_t n) {
+  return (int)std::min<int64_t>((n + 255) / 256, 4096);
+}
+
+void depthwise_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
+                   int64_t stride, int64_t pad) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 &&
+                  x.is_contiguous() && w.is_contiguous(),
+              "depthwise: contiguous bf16");
+  const int B = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3);
+  const int OH = (int)y.size(2), OW = (int)y.size(3);
+  const int KS = (int)w.size(w.dim() - 1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(depthwise_fwd_kernel,
+                     dim3(dw_grid((int64_t)B * C * OH * OW)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)x.data_ptr(),
+                     (const bf16_t*)w.data_ptr(), (bf16_t*)y.data_ptr(), B, C,
+                     H, W, OH, OW, KS, (int)stride, (int)pad);
+  HIP_CHECK_KERNEL();
+}
+
+void depthwise_bwd_dx(const at::Tensor& dy, const at::Tensor& w,
+                      at::Tensor& dx, int64_t stride, int64_t pad) {
+  const int B = (int)dx.size(0), C = (int)dx.size(1), H = (int)dx.size(2),
+            W = (int)dx.size(3);
+  const int OH = (int)dy.size(2), OW = (int)dy.size(3);
+  const int KS = (int)w.size(w.dim() - 1);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(depthwise_bwd_dx_kernel,
+                     dim3(dw_grid((int64_t)B * C * H * W)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)dy.data_ptr(),
+                     (const bf16_t*)w.data_ptr(), (bf16_t*)dx.data_ptr(), B,
+                     C, H, W, OH, OW, KS, (int)stride, (int)pad);
+  HIP_CHECK_KERNEL();
+}
+
+void depthwise_bwd_dw(const at::Tensor& x, const at::Tensor& dy,
+                      at::Tensor& dw, int64_t stride, int64_t pad) {
+  const int B = (int)x.size(0), C = (int)x.size(1), H = (int)x.size(2),
+            W = (int)x.size(3);
+  const int OH = (int)dy.size(2), OW = (int)dy.size(3);
+  const int KS = (int)(std::lround(std::sqrt((double)(dw.numel() / C))));
+  TORCH_CHECK(dw.scalar_type() == at::kFloat, "depthwise dW: fp32 out");
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(depthwise_bwd_dw_kernel,
+                     dim3((unsigned)C, (unsigned)(KS * KS)), dim3(256), 0,
+                     stream.stream(), (const bf16_t*)x.data_ptr(),
+                     (const bf16_t*)dy.data_ptr(), dw.data_ptr<float>(), B, C,
+                     H, W, OH, OW, KS, (int)stride, (int)pad);
+  HIP_CHECK_KERNEL();
+}

@@ -47,19 +47,18 @@ class SepConv(nn.Module):
 
     def __init__(self, c_in, c_out, kernel, stride):
         super().__init__()
-        from adanet_amd.ops.conv import HipConv1x1
+        from adanet_amd.ops.conv import HipConv1x1, HipDepthwiseConv2d
         pad = kernel // 2
-        # pointwise convs run on the batched MFMA GEMM when channels align
-        # (ops/conv.py); depthwise stays on MIOpen (grouped conv).
+        # Fully native separable conv: depthwise on the direct kernel
+        # (csrc/depthwise.hip, any channel count) + pointwise on the
+        # batched MFMA GEMM when channels align (ops/conv.py).
         self.op = nn.Sequential(
             nn.ReLU(),
-            nn.Conv2d(c_in, c_in, kernel, stride=stride, padding=pad,
-                      groups=c_in, bias=False),
+            HipDepthwiseConv2d(c_in, kernel, stride=stride, padding=pad),
             HipConv1x1(c_in, c_out, bias=False),
             BNfp32(c_out),
             nn.ReLU(),
-            nn.Conv2d(c_out, c_out, kernel, stride=1, padding=pad,
-                      groups=c_out, bias=False),
+            HipDepthwiseConv2d(c_out, kernel, stride=1, padding=pad),
             HipConv1x1(c_out, c_out, bias=False),
             BNfp32(c_out),
         )

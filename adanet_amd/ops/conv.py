@@ -126,3 +126,66 @@ def conv1x1(x: torch.Tensor, weight: torch.Tensor,
     return torch.nn.functional.conv2d(
         x, weight.reshape(co, ci, 1, 1).to(x.dtype),
         bias.to(x.dtype) if bias is not None else None)
+
+
+class _DepthwiseFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, pad):
+        ext = _extension.require()
+        B, C, H, W = x.shape
+        KS = weight.shape[-1]
+        OH = (H + 2 * pad - KS) // stride + 1
+        OW = (W + 2 * pad - KS) // stride + 1
+        y = torch.empty((B, C, OH, OW), device=x.device, dtype=torch.bfloat16)
+        ext.depthwise_fwd(x, weight, y, stride, pad)
+        ctx.save_for_backward(x, weight)
+        ctx.stride, ctx.pad = stride, pad
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _extension.require()
+        x, weight = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = torch.empty_like(x)
+            ext.depthwise_bwd_dx(dy, weight, dx, ctx.stride, ctx.pad)
+        if ctx.needs_input_grad[1]:
+            dwf = torch.empty(weight.numel(), device=x.device,
+                              dtype=torch.float32)
+            ext.depthwise_bwd_dw(x, dy, dwf, ctx.stride, ctx.pad)
+            dw = dwf.reshape(weight.shape).to(weight.dtype)
+        return dx, dw, None, None
+
+
+class HipDepthwiseConv2d(nn.Module):
+    """Depthwise KxK conv on the native kernel (any channel count),
+    drop-in for ``nn.Conv2d(C, C, K, stride, padding, groups=C, bias=False)``.
+    Falls back to grouped conv2d (MIOpen) off-GPU or for non-bf16 input.
+    """
+
+    def __init__(self, channels: int, kernel_size: int, stride: int = 1,
+                 padding: int = 0):
+        super().__init__()
+        self.channels = channels
+        self.kernel_size = kernel_size
+        self.stride = stride
+        self.padding = padding
+        ref = nn.Conv2d(channels, channels, kernel_size, stride=stride,
+                        padding=padding, groups=channels, bias=False)
+        self.weight = nn.Parameter(ref.weight.detach())  # [C,1,K,K]
+
+    def forward(self, x):
+        if (x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous()
+                and self.weight.dtype == torch.bfloat16):
+            return _DepthwiseFn.apply(x, self.weight.contiguous(),
+                                      self.stride, self.padding)
+        return torch.nn.functional.conv2d(
+            x, self.weight.to(x.dtype), None, self.stride, self.padding,
+            groups=self.channels)
+
+    def extra_repr(self):
+        return "C=%d, k=%d, s=%d, p=%d (native bf16 path)" % (
+            self.channels, self.kernel_size, self.stride, self.padding)

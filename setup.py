@@ -24,6 +24,7 @@ SRC = [
     "adanet_amd/csrc/layernorm.hip",
     "adanet_amd/csrc/elementwise.hip",
     "adanet_amd/csrc/reduce.hip",
+    "adanet_amd/csrc/depthwise.hip",
 ]
 
 setup(

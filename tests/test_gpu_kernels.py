@@ -594,3 +594,27 @@ def test_multi_copy_bf16():
     torch.cuda.synchronize()
     for s, d in zip(srcs, dsts):
         assert torch.equal(s, d)
+
+
+@pytest.mark.parametrize("C,KS,stride", [(24, 3, 1), (32, 5, 2), (7, 3, 2)])
+def test_depthwise_conv_fwd_bwd_vs_fp32(C, KS, stride):
+    from adanet_amd.ops.conv import _DepthwiseFn
+    torch.manual_seed(9)
+    B, H, W = 3, 17, 17
+    pad = KS // 2
+    x = torch.randn(B, C, H, W, device=DEV).to(
+        torch.bfloat16).requires_grad_(True)
+    w = torch.randn(C, 1, KS, KS, device=DEV).mul(0.2).to(
+        torch.bfloat16).requires_grad_(True)
+    y = _DepthwiseFn.apply(x, w, stride, pad)
+    xf = x.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(xf, wf, None, stride, pad, groups=C)
+    rel = (y.float() - ref).abs().mean() / (ref.abs().mean() + 1e-3)
+    assert rel < 0.02, rel
+    up = torch.randn_like(ref)
+    y.backward(up.to(torch.bfloat16))
+    ref.backward(up)
+    for got, want in [(x.grad.float(), xf.grad), (w.grad.float(), wf.grad)]:
+        rel = (got - want).abs().mean() / (want.abs().mean() + 1e-6)
+        assert rel < 0.03, rel
