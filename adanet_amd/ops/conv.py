@@ -25,8 +25,13 @@ from torch import nn
 from adanet_amd.ops import _extension
 
 
+def _native_enabled() -> bool:
+    import os
+    return os.environ.get("ADANET_NATIVE_CONV", "1") != "0"
+
+
 def _aligned(x: torch.Tensor, co: int, ci: int) -> bool:
-    if not x.is_cuda or x.dtype != torch.bfloat16:
+    if not x.is_cuda or x.dtype != torch.bfloat16 or not _native_enabled():
         return False
     hw = x.shape[2] * x.shape[3]
     return ci % 32 == 0 and co % 32 == 0 and hw % 8 == 0 and x.is_contiguous()
@@ -179,7 +184,8 @@ class HipDepthwiseConv2d(nn.Module):
 
     def forward(self, x):
         if (x.is_cuda and x.dtype == torch.bfloat16 and x.is_contiguous()
-                and self.weight.dtype == torch.bfloat16):
+                and self.weight.dtype == torch.bfloat16
+                and _native_enabled()):
             return _DepthwiseFn.apply(x, self.weight.contiguous(),
                                       self.stride, self.padding)
         return torch.nn.functional.conv2d(
