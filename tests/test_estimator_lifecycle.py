@@ -492,3 +492,20 @@ def test_regression_head_end_to_end(model_dir):
     est.train(input_fn, max_steps=20)
     res = est.evaluate(input_fn, steps=4)
     assert res["average_loss"] == res["average_loss"]
+
+
+def test_keep_checkpoint_max_gc(model_dir, synthetic_classification):
+    """Old increment checkpoints are garbage-collected down to
+    keep_checkpoint_max (reference keep_checkpoint_max semantics)."""
+    X, Y, input_fn = synthetic_classification
+    gen = simple_dnn.Generator(layer_size=8)
+    est = adanet_amd.Estimator(
+        head=MultiClassHead(4), subnetwork_generator=gen,
+        max_iteration_steps=5, model_dir=model_dir,
+        config=adanet_amd.RunConfig(tf_random_seed=1, keep_checkpoint_max=2))
+    est.train(input_fn, max_steps=25)  # 5 iterations
+    ckpts = glob.glob(os.path.join(model_dir, "increment.ckpt-*.pt"))
+    assert len(ckpts) == 2, sorted(ckpts)
+    # latest checkpoint pointer still resolves
+    with open(os.path.join(model_dir, "checkpoint")) as f:
+        assert os.path.exists(f.read().strip())
