@@ -399,11 +399,12 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     LTR(256, 128, 4, 4, 2, 4, 2);
   } else if (N > M && t_128x256 >= 512) {
     LTR(128, 256, 4, 4, 2, 2, 4);
-  } else if (K >= M && K >= N && b128 >= 160 && b128 < 512) {
-    // K-dominant MID shapes only (grid under 512 tiles): 64^2 beats
-    // 128^2 by 14-22% at 2048^3, both trans_b and tt
-    // (profiles/tr_variants_r01.json) — but at 4096-class K-dominant
-    // shapes 64^2 collapses (393-434 vs 620+ TF), hence the b128 cap.
+  } else if (K >= M && K >= N && K <= 2048 && b128 >= 160 && b128 < 512) {
+    // K-dominant MID shapes only, capped at the MEASURED territory
+    // (K<=2048, grid under 512 tiles): 64^2 beats 128^2 by 14-22% at
+    // 2048^3, both trans_b and tt (profiles/tr_variants_r01.json) — but
+    // with K=4096 64^2 collapses (393-434 vs 620+ TF; the wide-hidden
+    // bench regressed 11k->6.6k via dX 2048x3072x4096 before the K cap).
     LTR(64, 64, 2, 2, 6, 2, 2);
   } else if (b128 >= 160) {
     LTR(128, 128, 2, 2, 4, 4, 4);
