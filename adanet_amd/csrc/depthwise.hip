@@ -38,6 +38,41 @@ __global__ __launch_bounds__(256) void depthwise_fwd_kernel(
   }
 }
 
+// KS-templated fast path: one (batch, channel) image per blockIdx.y, so
+// the KSxKS tap weights load once into registers (same c for the whole
+// block -> broadcast from L1) and the tap loops fully unroll.
+template <int KS>
+__global__ __launch_bounds__(256) void depthwise_fwd_tmpl_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ y, int C, int H, int W, int OH, int OW, int stride,
+    int pad) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const bf16_t* xp = x + (int64_t)bc * H * W;
+  bf16_t* yp = y + (int64_t)bc * OH * OW;
+  float wr[KS * KS];
+#pragma unroll
+  for (int i = 0; i < KS * KS; ++i) wr[i] = bf2f(w[c * KS * KS + i]);
+  const int total = OH * OW;
+  for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += gridDim.x * blockDim.x) {
+    const int ow = p % OW, oh = p / OW;
+    float acc = 0.f;
+#pragma unroll
+    for (int kh = 0; kh < KS; ++kh) {
+      const int ih = oh * stride + kh - pad;
+      if (ih < 0 || ih >= H) continue;
+#pragma unroll
+      for (int kw = 0; kw < KS; ++kw) {
+        const int iw = ow * stride + kw - pad;
+        if (iw < 0 || iw >= W) continue;
+        acc += wr[kh * KS + kw] * bf2f(xp[ih * W + iw]);
+      }
+    }
+    yp[p] = f2bf(acc);
+  }
+}
+
 // dX: full correlation with the flipped kernel, honoring stride divisibility.
 __global__ __launch_bounds__(256) void depthwise_bwd_dx_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ w,
@@ -114,11 +149,24 @@ void depthwise_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
   const int OH = (int)y.size(2), OW = (int)y.size(3);
   const int KS = (int)w.size(w.dim() - 1);
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(depthwise_fwd_kernel,
-                     dim3(dw_grid((int64_t)B * C * OH * OW)), dim3(256), 0,
-                     stream.stream(), (const bf16_t*)x.data_ptr(),
-                     (const bf16_t*)w.data_ptr(), (bf16_t*)y.data_ptr(), B, C,
-                     H, W, OH, OW, KS, (int)stride, (int)pad);
+  const int64_t bc = (int64_t)B * C;
+  if ((KS == 3 || KS == 5) && bc <= 65535) {
+    const int px_blocks = std::max(1, std::min((OH * OW + 255) / 256,
+                                               (int)(2048 / bc) + 1));
+    auto kern = KS == 3 ? depthwise_fwd_tmpl_kernel<3>
+                        : depthwise_fwd_tmpl_kernel<5>;
+    hipLaunchKernelGGL(kern, dim3((unsigned)px_blocks, (unsigned)bc),
+                       dim3(256), 0, stream.stream(),
+                       (const bf16_t*)x.data_ptr(),
+                       (const bf16_t*)w.data_ptr(), (bf16_t*)y.data_ptr(), C,
+                       H, W, OH, OW, (int)stride, (int)pad);
+  } else {
+    hipLaunchKernelGGL(depthwise_fwd_kernel,
+                       dim3(dw_grid((int64_t)B * C * OH * OW)), dim3(256), 0,
+                       stream.stream(), (const bf16_t*)x.data_ptr(),
+                       (const bf16_t*)w.data_ptr(), (bf16_t*)y.data_ptr(), B,
+                       C, H, W, OH, OW, KS, (int)stride, (int)pad);
+  }
   HIP_CHECK_KERNEL();
 }
 
