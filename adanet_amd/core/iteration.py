@@ -635,7 +635,11 @@ class _Iteration(object):
             creg = spec.ensemble.complexity_regularization()
             adanet_loss = loss + creg
             spec.optimizer.zero_grad(set_to_none=True)
-            adanet_loss.backward()
+            # mixer dW/db write direct-to-arena (ops/mixer.py backward);
+            # the L1-penalty grad still flows through AccumulateGrad and
+            # ADDS into the same views.
+            with direct_grad_writes():
+                adanet_loss.backward()
             spec.optimizer.step()
             return adanet_loss
         with torch.no_grad():

@@ -144,7 +144,10 @@ __global__ __launch_bounds__(256) void mixer_bwd_dw_vector_ptrs_kernel(
     float acc = 0.f;
     for (int b = 0; b < B; ++b)
       acc += bf2f(dY[(int64_t)b * ldy + c]) * bf2f(L[(int64_t)b * ldl + c]);
-    dw[j * C + c] = acc;
+    // ACCUMULATE (dw pre-zeroed): direct-to-arena mode shares the grad
+    // view with the L1-penalty term's AccumulateGrad add — overwrite
+    // would clobber whichever landed first.
+    dw[j * C + c] += acc;
   }
 }
 

@@ -38,22 +38,49 @@ class _MixerFn(torch.autograd.Function):
         ctx.vector_mode = vector_mode
         ctx.has_bias = bias is not None
         ctx.logit_requires = [t.requires_grad for t in logits]
+        # live Parameter identities for direct-to-arena grad writes
+        # (mixture phase runs under ops.linear.direct_grad_writes).
+        ctx.param_refs = (
+            stacked_w if isinstance(stacked_w, torch.nn.Parameter) else None,
+            bias if isinstance(bias, torch.nn.Parameter) else None)
         ctx.save_for_backward(stacked_w, *members)
         return out
 
     @staticmethod
     def backward(ctx, dy):
+        from adanet_amd.ops import linear as _lin
         ext = _extension.require()
         stacked_w = ctx.saved_tensors[0]
         members = list(ctx.saved_tensors[1:])
         B, C = members[0].shape
         dy = dy.contiguous()
-        dw = torch.zeros_like(stacked_w)
-        ext.mixer_bwd_dw_direct(members, dy, dw, 1 if ctx.vector_mode else 0)
+        wparam, bparam = ctx.param_refs
+        # Direct-to-arena: the mixture weights are used once per step, so
+        # the scalar kernel's atomicAdds (into the zeroed arena view) /
+        # the vector kernel's overwrites land the final gradient and we
+        # return None — no AccumulateGrad add, no zeros_like fill.
+        wg = wparam.grad if (_lin._DIRECT_GRAD and wparam is not None) \
+            else None
+        direct_w = (wg is not None and wg.is_cuda
+                    and wg.dtype == torch.float32
+                    and wg.shape == stacked_w.shape and wg.is_contiguous())
+        dw_out = wg if direct_w else torch.zeros_like(stacked_w)
+        ext.mixer_bwd_dw_direct(members, dy, dw_out,
+                                1 if ctx.vector_mode else 0)
+        dw = None if direct_w else dw_out
         dbias = None
         if ctx.has_bias:
-            dbias = torch.empty((C,), device=dy.device, dtype=torch.float32)
-            ext.colsum_bf16(dy, dbias)
+            bg = bparam.grad if (_lin._DIRECT_GRAD and bparam is not None) \
+                else None
+            direct_b = (bg is not None and bg.is_cuda
+                        and bg.dtype == torch.float32
+                        and bg.shape == (C,) and bg.is_contiguous())
+            if direct_b:
+                ext.colsum_bf16(dy, bg, 1)
+            else:
+                dbias = torch.empty((C,), device=dy.device,
+                                    dtype=torch.float32)
+                ext.colsum_bf16(dy, dbias)
         dlogits = []
         for j, req in enumerate(ctx.logit_requires):
             if req:
