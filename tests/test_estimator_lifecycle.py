@@ -509,3 +509,48 @@ def test_keep_checkpoint_max_gc(model_dir, synthetic_classification):
     # latest checkpoint pointer still resolves
     with open(os.path.join(model_dir, "checkpoint")) as f:
         assert os.path.exists(f.read().strip())
+
+
+def test_resume_across_processes(model_dir, tmp_path):
+    """Kill-and-restart robustness: a FRESH PROCESS resumes mid-iteration
+    from disk alone (no in-memory state survives)."""
+    import subprocess
+    import sys
+    script = tmp_path / "resume_driver.py"
+    script.write_text("""
+import sys
+import torch
+import adanet_amd
+from adanet_amd.head import MultiClassHead
+from adanet_amd.models import simple_dnn
+
+model_dir, steps = sys.argv[1], int(sys.argv[2])
+torch.manual_seed(0)
+X = torch.randn(64, 12)
+Y = (X[:, :4].sum(1) > 0).long() + 2 * (X[:, 4:8].sum(1) > 0).long()
+
+def input_fn():
+    while True:
+        yield X, Y
+
+est = adanet_amd.Estimator(
+    head=MultiClassHead(4),
+    subnetwork_generator=simple_dnn.Generator(layer_size=8),
+    max_iteration_steps=10, model_dir=model_dir,
+    config=adanet_amd.RunConfig(tf_random_seed=7))
+est.train(input_fn, steps=steps)
+print("ITER=%d STEP=%d" % (est.iteration_number, est.global_step))
+""")
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+    def run(steps):
+        env = dict(os.environ)
+        env["PYTHONPATH"] = repo + os.pathsep + env.get("PYTHONPATH", "")
+        out = subprocess.run(
+            [sys.executable, str(script), model_dir, str(steps)],
+            capture_output=True, text=True, timeout=600, env=env)
+        assert out.returncode == 0, out.stderr[-2000:]
+        return out.stdout.strip().splitlines()[-1]
+
+    assert run(12) == "ITER=1 STEP=12"   # process A dies mid-iteration 1
+    assert run(18) == "ITER=3 STEP=30"   # process B resumes + continues
