@@ -72,10 +72,11 @@ class ReluConvBN(nn.Module):
     def __init__(self, c_in, c_out, kernel=1, stride=1):
         super().__init__()
         from adanet_amd.ops.conv import HipConv1x1
+        from adanet_amd.ops.conv import HipConvNxN
         conv = (HipConv1x1(c_in, c_out, bias=False)
                 if kernel == 1 and stride == 1 else
-                nn.Conv2d(c_in, c_out, kernel, stride=stride,
-                          padding=kernel // 2, bias=False))
+                HipConvNxN(c_in, c_out, kernel, stride=stride,
+                           padding=kernel // 2, bias=False))
         self.op = nn.Sequential(
             nn.ReLU(),
             conv,
@@ -93,10 +94,13 @@ class FactorizedReduction(nn.Module):
     def __init__(self, c_in, c_out):
         super().__init__()
         assert c_out % 2 == 0
+        from adanet_amd.ops.conv import HipConvNxN
         self.relu = nn.ReLU()
-        self.p1 = nn.Conv2d(c_in, c_out // 2, 1, stride=2, bias=False)
-        self.p2 = nn.Conv2d(c_in, c_out - c_out // 2, 1, stride=2,
-                            bias=False)
+        # stride-2 1x1 paths via unfold+MFMA GEMM (MIOpen's naive bf16
+        # wrw kernel was 6.9 ms/call on these; ops/conv.py HipConvNxN).
+        self.p1 = HipConvNxN(c_in, c_out // 2, 1, stride=2, bias=False)
+        self.p2 = HipConvNxN(c_in, c_out - c_out // 2, 1, stride=2,
+                             bias=False)
         self.bn = BNfp32(c_out)
 
     def forward(self, x):
@@ -217,7 +221,9 @@ class NasNetCIFAR(nn.Module):
         super().__init__()
         c_stem = stem_multiplier * num_conv_filters
         self.stem = nn.Sequential(
-            nn.Conv2d(in_channels, c_stem, 3, padding=1, bias=False),
+            __import__("adanet_amd.ops.conv", fromlist=["HipConvNxN"]
+                       ).HipConvNxN(in_channels, c_stem, 3, padding=1,
+                                    bias=False),
             BNfp32(c_stem))
         self.cells = nn.ModuleList()
         reduction_points = set()

@@ -195,3 +195,64 @@ class HipDepthwiseConv2d(nn.Module):
     def extra_repr(self):
         return "C=%d, k=%d, s=%d, p=%d (native bf16 path)" % (
             self.channels, self.kernel_size, self.stride, self.padding)
+
+
+class HipConvNxN(nn.Module):
+    """Dense KxK conv as unfold + the batched MFMA GEMM (K9 tail).
+
+    MIOpen falls back to a naive weight-gradient kernel for these NCHW
+    bf16 shapes (6.9 ms/call — 35% of the NASNet step,
+    profiles/conv_bench_r01.txt); expressing the conv as im2col (one
+    torch.unfold copy, a few MB) + gemm_tr_batched keeps fwd/bwd on the
+    MFMA path. The im2col K dim (Ci*K*K) is zero-padded to 32 inside the
+    padded weight, so ANY in-channel count works; requires Co % 32 == 0
+    (else torch fallback). Drop-in for
+    ``nn.Conv2d(ci, co, K, stride, padding, bias=...)``.
+    """
+
+    def __init__(self, in_channels: int, out_channels: int, kernel_size: int,
+                 stride: int = 1, padding: int = 0, bias: bool = True):
+        super().__init__()
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.kernel_size = kernel_size
+        self.stride = stride
+        self.padding = padding
+        ref = nn.Conv2d(in_channels, out_channels, kernel_size,
+                        stride=stride, padding=padding, bias=bias)
+        self.weight = nn.Parameter(ref.weight.detach())  # [Co, Ci, K, K]
+        if bias:
+            self.bias = nn.Parameter(ref.bias.detach().float())
+        else:
+            self.register_parameter("bias", None)
+
+    def forward(self, x):
+        K = self.kernel_size
+        ckk = self.in_channels * K * K
+        ckk_pad = (ckk + 31) // 32 * 32
+        if (x.is_cuda and x.dtype == torch.bfloat16 and _native_enabled()
+                and self.weight.dtype == torch.bfloat16
+                and self.out_channels % 32 == 0):
+            B = x.shape[0]
+            oh = (x.shape[2] + 2 * self.padding - K) // self.stride + 1
+            ow = (x.shape[3] + 2 * self.padding - K) // self.stride + 1
+            u = torch.nn.functional.unfold(
+                x, K, padding=self.padding, stride=self.stride)  # [B,ckk,L]
+            if ckk_pad != ckk:
+                u = torch.nn.functional.pad(u, (0, 0, 0, ckk_pad - ckk))
+            w2 = self.weight.reshape(self.out_channels, ckk)
+            if ckk_pad != ckk:
+                w2 = torch.nn.functional.pad(w2, (0, ckk_pad - ckk))
+            y = _Conv1x1Fn.apply(
+                u.reshape(B, ckk_pad, oh, ow).contiguous(), w2.contiguous(),
+                self.bias)
+            return y
+        return torch.nn.functional.conv2d(
+            x, self.weight.to(x.dtype),
+            self.bias.to(x.dtype) if self.bias is not None else None,
+            self.stride, self.padding)
+
+    def extra_repr(self):
+        return "in=%d, out=%d, k=%d, s=%d (im2col+MFMA when Co%%32==0)" % (
+            self.in_channels, self.out_channels, self.kernel_size,
+            self.stride)

@@ -78,10 +78,11 @@ def restore_fp32_params(module) -> None:
     fp32 epilogues/statistics); `module.to(torch.bfloat16)` downcasts them,
     so call this afterwards.
     """
-    from adanet_amd.ops.conv import HipConv1x1
+    from adanet_amd.ops.conv import HipConv1x1, HipConvNxN
     from adanet_amd.ops.layernorm import HipLayerNorm
     for m in module.modules():
-        if isinstance(m, (HipLinear, HipConv1x1)) and m.bias is not None:
+        if isinstance(m, (HipLinear, HipConv1x1, HipConvNxN)) \
+                and m.bias is not None:
             m.bias.data = m.bias.data.float()
         if isinstance(m, HipLayerNorm) and m.weight is not None:
             m.weight.data = m.weight.data.float()

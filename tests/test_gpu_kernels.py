@@ -618,3 +618,38 @@ def test_depthwise_conv_fwd_bwd_vs_fp32(C, KS, stride):
     for got, want in [(x.grad.float(), xf.grad), (w.grad.float(), wf.grad)]:
         rel = (got - want).abs().mean() / (want.abs().mean() + 1e-6)
         assert rel < 0.03, rel
+
+
+def test_conv_nxn_unfold_gemm_vs_fp32():
+    from adanet_amd.ops.conv import HipConvNxN
+    torch.manual_seed(11)
+    m = HipConvNxN(3, 32, 3, padding=1).to(DEV).to(torch.bfloat16)
+    from adanet_amd.ops.linear import restore_fp32_params
+    restore_fp32_params(m)
+    x = torch.randn(4, 3, 16, 16, device=DEV).to(
+        torch.bfloat16).requires_grad_(True)
+    y = m(x)
+    xf = x.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(xf, m.weight.float(), m.bias, 1, 1)
+    rel = (y.float() - ref).abs().mean() / (ref.abs().mean() + 1e-3)
+    assert rel < 0.02, rel
+    up = torch.randn_like(ref)
+    y.backward(up.to(torch.bfloat16))
+    ref.backward(up)
+    rel = (m.weight.grad.float().flatten()
+           - torch.autograd.grad(
+               torch.nn.functional.conv2d(
+                   x.detach().float().requires_grad_(False),
+                   (wf := m.weight.detach().float().requires_grad_(True)),
+                   m.bias.detach(), 1, 1), wf,
+               grad_outputs=up)[0].flatten()).abs().mean()
+    assert rel < 0.05, rel
+    # stride-2 1x1 variant (FactorizedReduction path)
+    m2 = HipConvNxN(32, 32, 1, stride=2, bias=False).to(DEV).to(
+        torch.bfloat16)
+    x2 = torch.randn(2, 32, 16, 16, device=DEV).to(torch.bfloat16)
+    y2 = m2(x2)
+    ref2 = torch.nn.functional.conv2d(x2.float(), m2.weight.float(),
+                                      None, 2, 0)
+    assert (y2.float() - ref2).abs().mean() / (
+        ref2.abs().mean() + 1e-3) < 0.02
