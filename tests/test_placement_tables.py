@@ -7,15 +7,6 @@ from adanet_amd.distributed.placement import (ReplicationStrategy,
                                               RoundRobinStrategy)
 
 
-class _Fixed:
-    """Pin world/rank without a process group."""
-
-    def __init__(self, strat, world, rank):
-        self.s = strat
-        strat.world_size = world          # type: ignore[assignment]
-        strat.rank = rank                 # type: ignore[assignment]
-
-
 def _patch(strat, world, rank, monkeypatch):
     monkeypatch.setattr(type(strat), "world_size",
                         property(lambda self: world))
