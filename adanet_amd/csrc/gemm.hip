@@ -16,9 +16,12 @@
 //     publishes the staged tile (the m97 structure from the CDNA4 guide).
 //   * XCD-aware bijective blockIdx swizzle so neighboring output tiles
 //     share a chiplet-local L2 (8 XCDs).
-// All A/B/C layouts are row-major with the reduction dim (K) minor; the
-// python wrappers materialize transposes for the backward GEMMs with the
-// LDS-tiled transpose kernel (transpose.hip).
+// All A/B/C layouts are row-major with the reduction dim (K) minor.
+// Backward GEMMs consume K-major operands WITHOUT transposition via the
+// ds_read_b64_tr_b16 transposed-staging kernels in gemm_tn.hip; skinny-M
+// (M<=8, batch-1 serving) dispatches to a wave-per-column GEMV here.
+// Epilogue act codes: 0 = none, 1 = ReLU, 2 = accumulate into C
+// (direct-to-arena gradient writes).
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>

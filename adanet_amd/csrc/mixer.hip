@@ -4,9 +4,10 @@
 // adanet/ensemble/weighted.py:427-454,545-561) computed as ONE fused kernel
 // over the J member logit buffers: out[b,c] = bias[c] + sum_j w_j(*)L[j,b,c],
 // where w_j is a scalar (MixtureWeightType.SCALAR) or per-class vector
-// (VECTOR). Members arrive as a stacked [J, B, C] bf16 tensor (the stack is
-// a device copy, so the whole op is hipGraph-capturable — a host-built
-// pointer table would bake stale addresses into the graph).
+// (VECTOR). The production path reads the J member buffers IN PLACE via
+// kernel-arg pointer structs (mixer_*_direct below — no per-step stack
+// copy; hipGraph capture bakes the stable static-buffer addresses). The
+// older stacked-[J,B,C] entry points are kept for the probe/tests.
 
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>

@@ -1,13 +1,17 @@
 """HipLinear: bf16 dense layer on the fused MFMA GEMM (K1).
 
 Forward is one gemm_nt_bf16 call with the bias+ReLU epilogue fused into the
-GEMM (adanet_amd/csrc/gemm.hip). Backward runs two more MFMA GEMMs with
-K-minor operands materialized by the LDS-tiled transpose kernel, plus a
-column-sum for the bias gradient:
+GEMM (adanet_amd/csrc/gemm.hip). Backward is TRANSPOSE-FREE: the
+transposed-staging GEMMs (csrc/gemm_tn.hip, ds_read_b64_tr_b16) consume
+dz/W/x in their stored K-major layouts, and under `direct_grad_writes`
+their epilogues land dW/db straight in the optimizer's arena grad views
+(accumulate act=2, or plain overwrite when the optimizer skipped the
+arena memset for single-write params):
 
-    dX = dReLU(dY) @ W          (gemm_nt with W^T)
-    dW = dReLU(dY)^T @ X        (gemm_nt with dY^T, X^T)
-    db = colsum(dReLU(dY))
+    dz = dReLU(dY)              (fused with the db colsum when direct)
+    dX = dz @ W                 (trans_b staging; per-shape vs composite)
+    dW = dz^T @ X               (tt staging -> arena view)
+    db = colsum(dz)             (-> arena view)
 
 The ReLU mask is recovered from the saved forward OUTPUT (y > 0), so the
 fused epilogue never materializes a mask tensor.
