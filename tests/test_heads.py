@@ -63,3 +63,25 @@ def test_binary_head_contract():
     assert abs(float(h.loss(logits, labels)) - float(want)) < 1e-5
     p = h.predictions(logits)
     assert set(p) == {"logits", "probabilities", "class_ids"}
+
+
+def test_train_hooks_fire_and_stop():
+    from adanet_amd.hooks import EveryNSteps, StopAfterSteps
+    seen = []
+    h1 = EveryNSteps(2, lambda step: seen.append(step))
+    h2 = StopAfterSteps(5)
+    for s in range(1, 8):
+        h1.after_step(s)
+        h2.after_step(s)
+    assert seen == [2, 4, 6]
+    assert h2.should_stop
+
+
+def test_comm_singleprocess_fallbacks():
+    """comm helpers degrade to no-ops without a process group."""
+    from adanet_amd.distributed import comm
+    assert not comm.is_initialized()
+    assert comm.world_size() == 1 and comm.rank() == 0 and comm.is_chief()
+    assert comm.broadcast_object({"x": 1}) == {"x": 1}
+    assert comm.all_gather_objects("v") == ["v"]
+    comm.barrier()  # no-op
