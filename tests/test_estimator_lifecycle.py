@@ -554,3 +554,26 @@ print("ITER=%d STEP=%d" % (est.iteration_number, est.global_step))
 
     assert run(12) == "ITER=1 STEP=12"   # process A dies mid-iteration 1
     assert run(18) == "ITER=3 STEP=30"   # process B resumes + continues
+
+
+def test_adanet_lambda_steers_selection_toward_low_complexity(
+        model_dir, synthetic_classification):
+    """The complexity penalty changes WHICH architectures win: with a
+    strong lambda, zero-complexity (linear) candidates beat deeper DNNs
+    (the reference adanet_objective tutorial's headline behavior)."""
+    X, Y, input_fn = synthetic_classification
+
+    def total_complexity(lam, sub_dir):
+        est = adanet_amd.Estimator(
+            head=MultiClassHead(4),
+            subnetwork_generator=simple_dnn.Generator(
+                layer_size=8, learn_mixture_weights=True),
+            max_iteration_steps=15, adanet_lambda=lam,
+            model_dir=os.path.join(model_dir, sub_dir),
+            config=adanet_amd.RunConfig(tf_random_seed=11))
+        est.train(input_fn, max_steps=45)
+        ens, _ = est._rebuild_previous_ensemble(est.iteration_number, X[:8])
+        return sum(float(ws.subnetwork.complexity)
+                   for ws in ens.weighted_subnetworks)
+
+    assert total_complexity(5.0, "strong") <= total_complexity(0.0, "none")
