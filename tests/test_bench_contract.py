@@ -57,3 +57,22 @@ def test_bench_torchrun_two_ranks_cpu(tmp_path):
     assert d["config"]["candidates_per_iter"] == 4  # 2 per rank (weak)
     assert "round_robin2" in d["config"]["parallelism"]
     assert d["value"] > 0
+
+
+def test_bench_torchrun_replication_cpu(tmp_path):
+    """DP placement at N=2 over gloo: flat-bucket all-reduce path."""
+    port = _free_port()
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--cpu", "--steps", "1", "--warmup", "0",
+         "--batch", "64", "--hidden", "32", "--train-steps-per-iter", "3",
+         "--eval-batches", "2", "--placement", "replication"],
+        capture_output=True, text=True, timeout=900, cwd=str(tmp_path))
+    assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-2000:])
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1
+    d = json.loads(lines[0])
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["value"] > 0
