@@ -31,7 +31,9 @@ import torch
 
 def parse_args():
     p = argparse.ArgumentParser()
-    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--gpus", type=int, default=1,
+                   help="accepted for the driver contract; the actual "
+                        "world size comes from the torchrun env")
     p.add_argument("--steps", type=int, default=3,
                    help="AdaNet iterations to time")
     p.add_argument("--warmup", type=int, default=1,
