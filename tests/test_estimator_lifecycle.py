@@ -577,3 +577,32 @@ def test_adanet_lambda_steers_selection_toward_low_complexity(
                    for ws in ens.weighted_subnetworks)
 
     assert total_complexity(5.0, "strong") <= total_complexity(0.0, "none")
+
+
+def test_replicate_ensemble_in_training(model_dir, synthetic_classification):
+    """replicate_ensemble_in_training=True recomputes frozen members in
+    TRAIN mode each step (reference iteration.py:569-572 semantics: no
+    eval-mode freeze); training still completes and grows."""
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(model_dir, input_fn,
+                          replicate_ensemble_in_training=True)
+    est.train(input_fn, max_steps=20)
+    assert est.iteration_number == 2
+    res = est.evaluate(input_fn, steps=3)
+    assert math.isfinite(res["loss"])
+
+
+def test_save_checkpoints_steps_mid_iteration(model_dir,
+                                              synthetic_classification):
+    """save_checkpoints_steps writes MID-iteration checkpoints (resume
+    granularity below one boosting round)."""
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        config=adanet_amd.RunConfig(tf_random_seed=42,
+                                    save_checkpoints_steps=4))
+    est.train(input_fn, steps=6)  # inside iteration 0
+    est._join_ckpt_writer()
+    assert glob.glob(os.path.join(model_dir, "increment.ckpt-0.pt"))
+    est2 = _make_estimator(model_dir, input_fn)
+    assert est2.global_step in (4, 6)  # resumed from a mid-iteration save
