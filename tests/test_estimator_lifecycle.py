@@ -606,3 +606,21 @@ def test_save_checkpoints_steps_mid_iteration(model_dir,
     assert glob.glob(os.path.join(model_dir, "increment.ckpt-0.pt"))
     est2 = _make_estimator(model_dir, input_fn)
     assert est2.global_step in (4, 6)  # resumed from a mid-iteration save
+
+
+def test_seeded_runs_reproduce_architectures(model_dir,
+                                             synthetic_classification):
+    """Same tf_random_seed -> identical selected architectures across two
+    independent runs (CPU is deterministic; the reference's determinism
+    contract for Generators, generator.py:288)."""
+    X, Y, input_fn = synthetic_classification
+
+    def run(sub):
+        d = os.path.join(model_dir, sub)
+        est = _make_estimator(d, input_fn)
+        est.train(input_fn, max_steps=30)
+        return [json.loads(open(os.path.join(
+            d, "architecture-%d.json" % t)).read())["subnetworks"]
+            for t in range(3)]
+
+    assert run("a") == run("b")
