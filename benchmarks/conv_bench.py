@@ -37,8 +37,12 @@ def main():
             step()
         torch.cuda.synchronize()
         ms = (time.perf_counter() - t0) / 10 * 1e3
-        print("filters=%d (native=%s): %.1f ms/step" % (
-            filters, filters % 32 == 0, ms))
+        mode = ("fully-native" if filters % 32 == 0 else
+                "native-depthwise (GEMM paths unaligned -> library)")
+        import os
+        if os.environ.get("ADANET_NATIVE_CONV") == "0":
+            mode = "all-library (MIOpen)"
+        print("filters=%d [%s]: %.1f ms/step" % (filters, mode, ms))
 
 
 if __name__ == "__main__":
