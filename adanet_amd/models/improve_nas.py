@@ -57,6 +57,11 @@ class Hparams:
     learn_mixture_weights: bool = False
     drop_path_keep: float = 0.9
     train_steps: int = 10000  # cosine horizon per iteration
+    #: round filter counts up to a multiple of this so the native conv
+    #: stack (batched-MFMA pointwise / unfold-GEMM adaptors, ops/conv.py)
+    #: engages; 1 = exact reference filter counts (library conv fallback
+    #: for unaligned channels). 32 is the GEMM alignment contract.
+    channel_multiple: int = 1
 
 
 class NasNetBuilder(Builder):
@@ -67,7 +72,9 @@ class NasNetBuilder(Builder):
                  name_suffix: str = "", seed: Optional[int] = None):
         self._hp = hparams
         self._num_cells = num_cells or hparams.num_cells
-        self._filters = num_conv_filters or hparams.num_conv_filters
+        f = num_conv_filters or hparams.num_conv_filters
+        m = max(1, int(getattr(hparams, "channel_multiple", 1)))
+        self._filters = (f + m - 1) // m * m
         self._suffix = name_suffix
         self._seed = seed
 

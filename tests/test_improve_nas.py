@@ -82,3 +82,16 @@ def test_improve_nas_estimator_lifecycle(tmp_path):
     assert est.iteration_number == 2
     res = est.evaluate(input_fn, steps=2)
     assert "accuracy" in res
+
+
+def test_channel_multiple_rounds_filters():
+    from adanet_amd.models.improve_nas import (DynamicGenerator, Hparams,
+                                               NasNetBuilder)
+    hp = Hparams(num_conv_filters=10, channel_multiple=32)
+    assert NasNetBuilder(hp)._filters == 32
+    # widened candidates stay aligned too
+    gen = DynamicGenerator(hp)
+    cands = gen.generate_candidates(None, 0, [], [])
+    assert all(c._filters % 32 == 0 for c in cands)
+    # default keeps the reference's exact counts
+    assert NasNetBuilder(Hparams(num_conv_filters=10))._filters == 10
