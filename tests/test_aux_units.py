@@ -68,3 +68,25 @@ def test_countdown_timer():
     assert t.secs_remaining() > 0
     time.sleep(0.25)
     assert t.secs_remaining() == 0
+
+
+def test_public_api_surface():
+    """Every symbol the docs/README reference resolves (reference
+    adanet_test.py:23-25 pattern: assert public attrs exist)."""
+    import adanet_amd
+    from adanet_amd import distributed, ensemble, replay, serving, subnetwork
+    for sym in ("Estimator", "AutoEnsembleEstimator",
+                "AutoEnsembleSubestimator", "Evaluator",
+                "ReportMaterializer", "RunConfig", "Summary"):
+        assert hasattr(adanet_amd, sym), sym
+    assert distributed.RoundRobinStrategy and distributed.ReplicationStrategy
+    assert serving.export_torchscript and serving.export_program
+    assert replay.Config and subnetwork.Builder and subnetwork.Generator
+    assert ensemble.ComplexityRegularizedEnsembler and ensemble.MeanEnsembler
+    from adanet_amd.experimental import (ModelSearch,  # noqa: F401
+                                         MultiGpuScheduler,
+                                         ThreadedScheduler)
+    from adanet_amd.head import (BinaryClassHead,  # noqa: F401
+                                 MultiClassHead, MultiHead, RegressionHead)
+    from adanet_amd.ops.conv import (HipConv1x1,  # noqa: F401
+                                     HipConvNxN, HipDepthwiseConv2d)
