@@ -112,3 +112,56 @@ class DNNEstimator(object):
 
     def make_optimizer(self, params):
         return self._optimizer(params)
+
+
+class _DNNLinearCombinedModule(SubnetworkModule):
+    """Wide & deep: logits = linear(x) + dnn_head(dnn(x)); last layer is
+    the deep tower's top (the part downstream ensembles can extend)."""
+
+    def __init__(self, in_dim, hidden_units, logits_dim, dropout):
+        super().__init__()
+        self.flatten = _FlattenInput()
+        self.wide = HipLinear(in_dim, logits_dim)
+        layers = []
+        d = in_dim
+        for h in hidden_units:
+            layers.append(HipLinear(d, h, activation="relu"))
+            if dropout:
+                layers.append(HipDropout(dropout))
+            d = h
+        self.deep = nn.Sequential(*layers)
+        self.deep_logits = HipLinear(d, logits_dim, bias=False)
+        self.last_layer_dim = d
+
+    def forward(self, features):
+        x = self.flatten(features)
+        last = self.deep(x)
+        return last, self.wide(x) + self.deep_logits(last)
+
+
+class DNNLinearCombinedEstimator(object):
+    """Wide & deep (analog of tf.estimator.DNNLinearCombinedEstimator —
+    the canned family the reference's AutoEnsemble examples pool)."""
+
+    def __init__(self, head=None, hidden_units: Sequence[int] = (300,),
+                 optimizer: Optional[Callable] = None,
+                 linear_optimizer: Optional[Callable] = None,
+                 dropout: float = 0.0, seed: Optional[int] = None):
+        self.head = head
+        self.hidden_units = list(hidden_units)
+        # One fused optimizer over both towers (torch joint training; the
+        # reference's separate FTRL/Adagrad split is an async-PS artifact).
+        self._optimizer = (optimizer or linear_optimizer
+                           or functools.partial(FusedSGD, lr=0.01))
+        self.dropout = dropout
+        self._seed = seed
+
+    def build_model(self, features, logits_dimension, training):
+        if self._seed is not None:
+            torch.manual_seed(self._seed)
+        return _DNNLinearCombinedModule(
+            _feature_dim(features), self.hidden_units, logits_dimension,
+            self.dropout if training else 0.0)
+
+    def make_optimizer(self, params):
+        return self._optimizer(params)

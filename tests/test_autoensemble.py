@@ -173,3 +173,37 @@ def test_custom_logits_and_last_layer_fn(tmp_path):
     # last layer flowed through: spec outputs have hidden width 8
     spec = est._current_iteration
     assert est.iteration_number >= 0  # trained without error
+
+
+def test_dnn_linear_combined_in_pool(tmp_path):
+    """Wide & deep canned estimator trains inside an AutoEnsemble pool."""
+    import functools
+    from adanet_amd.models.canned import (DNNEstimator,
+                                          DNNLinearCombinedEstimator)
+    from adanet_amd.ops.optim import FusedSGD
+    torch.manual_seed(0)
+    X = torch.randn(256, 16)
+    Y = (X[:, :4].sum(dim=1) > 0).long()
+
+    def input_fn():
+        def gen():
+            while True:
+                yield X[:128], Y[:128]
+        return gen()
+
+    head = MultiClassHead(2)
+    est = AutoEnsembleEstimator(
+        head=head,
+        candidate_pool={
+            "wide_deep": DNNLinearCombinedEstimator(
+                head=head, hidden_units=[16],
+                optimizer=functools.partial(FusedSGD, lr=0.05)),
+            "dnn": DNNEstimator(head=head, hidden_units=[16]),
+        },
+        max_iteration_steps=10,
+        model_dir=str(tmp_path / "wd"),
+        config=adanet_amd.RunConfig(tf_random_seed=3))
+    est.train(input_fn, max_steps=20)
+    assert est.iteration_number == 2
+    res = est.evaluate(input_fn, steps=2)
+    assert res["accuracy"] >= 0.0
