@@ -624,3 +624,20 @@ def test_seeded_runs_reproduce_architectures(model_dir,
             for t in range(3)]
 
     assert run("a") == run("b")
+
+
+def test_custom_metric_fn(model_dir, synthetic_classification):
+    """metric_fn(predictions, features, labels) merges into evaluate()
+    results (reference estimator metric_fn contract)."""
+    X, Y, input_fn = synthetic_classification
+
+    def metric_fn(predictions, features, labels):
+        top1 = predictions["class_ids"]
+        return {"custom_error": float((top1 != labels).float().mean())}
+
+    est = _make_estimator(model_dir, input_fn, metric_fn=metric_fn)
+    est.train(input_fn, max_steps=20)
+    res = est.evaluate(input_fn, steps=3)
+    assert "custom_error" in res
+    assert res["custom_error"] == pytest.approx(1.0 - res["accuracy"],
+                                                abs=1e-6)
