@@ -526,11 +526,25 @@ class Estimator(object):
                         raise RuntimeError(
                             "Builder %r for iteration %d not regenerated — "
                             "Generators must be deterministic" % (bname, i))
+                    extra = {}
+                    try:
+                        import inspect
+                        params = inspect.signature(
+                            by_name[bname].build_subnetwork).parameters
+                        if "summary" in params:
+                            # rebuild is eval-mode: disabled (skip) summary
+                            extra["summary"] = self._make_summary(
+                                "subnetwork", bname, it_num, False)
+                        if "iteration_step" in params:
+                            extra["iteration_step"] = 0
+                    except (TypeError, ValueError):
+                        pass
                     sub = by_name[bname].build_subnetwork(
                         features,
                         logits_dimension=self._head.logits_dimension,
                         training=False,
-                        previous_ensemble=prev_ensemble)
+                        previous_ensemble=prev_ensemble,
+                        **extra)
                     # Frozen members are named t<i>_<builder> so the same
                     # builder chosen at two iterations stays distinct.
                     sub.name = "t{}_{}".format(it_num, bname)
@@ -623,6 +637,21 @@ class Estimator(object):
             summary = self._make_summary("subnetwork", b.name, t,
                                          self._enable_subnetwork_summaries)
             if build_here:
+                # Reference parity (generator.py:162-270): builders that
+                # declare `summary`/`labels`/`iteration_step` params get
+                # them — detected by introspection like the reference's
+                # optional `config` arg (estimator.py:1994-2006).
+                extra = {}
+                try:
+                    import inspect
+                    params = inspect.signature(
+                        b.build_subnetwork).parameters
+                    if "summary" in params:
+                        extra["summary"] = summary
+                    if "iteration_step" in params:
+                        extra["iteration_step"] = 0
+                except (TypeError, ValueError):  # builtins/partials
+                    pass
                 with torch.device(self._device):
                     # Factory calls inside builders allocate directly on the
                     # target device (no CPU init + transfer per iteration).
@@ -630,7 +659,8 @@ class Estimator(object):
                         features,
                         logits_dimension=self._head.logits_dimension,
                         training=True,
-                        previous_ensemble=prev_ensemble)
+                        previous_ensemble=prev_ensemble,
+                        **extra)
                 sub.name = b.name
                 sub.module.to(self._device)
                 if self._device.type == "cuda":
