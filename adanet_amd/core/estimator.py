@@ -485,10 +485,13 @@ class Estimator(object):
         return builders
 
     def _example_features(self, input_fn):
+        features, _ = self._example_batch(input_fn)
+        return features
+
+    def _example_batch(self, input_fn):
         it = iter(input_fn())
         features, labels = next(it)
-        features, labels = _to_device(features, labels, self._device)
-        return features
+        return _to_device(features, labels, self._device)
 
     def _frozen_key(self, iteration_number: int, builder_name: str) -> str:
         return "t{}|{}".format(iteration_number, builder_name)
@@ -620,7 +623,7 @@ class Estimator(object):
         t = self._iteration_number
         if self._config.random_seed is not None:
             torch.manual_seed(self._config.random_seed + t)
-        features = self._example_features(input_fn)
+        features, probe_labels = self._example_batch(input_fn)
         prev_ensemble, frozen = self._rebuild_previous_ensemble(t, features)
         builders = self._generate_builders(prev_ensemble, t)
 
@@ -650,6 +653,8 @@ class Estimator(object):
                         extra["summary"] = summary
                     if "iteration_step" in params:
                         extra["iteration_step"] = 0
+                    if "labels" in params:
+                        extra["labels"] = probe_labels
                 except (TypeError, ValueError):  # builtins/partials
                     pass
                 with torch.device(self._device):
