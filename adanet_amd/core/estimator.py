@@ -862,7 +862,8 @@ class Estimator(object):
         elif self._evaluator is not None:
             local = iteration.evaluate_candidates(
                 iter(self._evaluator.input_fn()), self._evaluator.steps,
-                lambda f, l: _to_device(f, l, self._device))
+                lambda f, l: _to_device(f, l, self._device),
+                metric_name=self._evaluator.metric_name)
             merged = comm.all_gather_objects({
                 i: v for i, v in enumerate(local)
                 if iteration.ensemble_specs[i].ensemble is not None

@@ -648,6 +648,23 @@ class _Iteration(object):
             creg = spec.ensemble.complexity_regularization()
             return loss + creg
 
+    def _ensemble_eval_metric(self, spec: _EnsembleSpec, frozen_out, labels,
+                              metric_name: str):
+        """A head metric of the candidate ensemble on one batch (custom
+        Evaluator metric_name, reference evaluator.py:31-60: any metric in
+        the candidate's eval dict may drive selection)."""
+        sub_logits, sub_last = self._gather_member_outputs(spec, frozen_out)
+        if sub_logits is None:
+            return None
+        with torch.no_grad():
+            logits = spec.ensemble.logits_from(sub_logits, sub_last)
+            metrics = self.head.metrics(logits, labels)
+        if metric_name not in metrics:
+            raise ValueError(
+                "Evaluator metric_name %r not produced by the head "
+                "(available: %s)" % (metric_name, sorted(metrics)))
+        return float(metrics[metric_name])
+
     # ------------------------------------------------------------------
     # loss ring buffer -> host EMAs
     # ------------------------------------------------------------------
@@ -765,9 +782,12 @@ class _Iteration(object):
     # ------------------------------------------------------------------
 
     def evaluate_candidates(self, input_iter, steps: Optional[int],
-                            to_device: Callable) -> List[float]:
-        """Mean adanet_loss per candidate over shared eval batches
+                            to_device: Callable,
+                            metric_name: str = "adanet_loss") -> List[float]:
+        """Mean eval metric per candidate over shared eval batches
         (reference evaluator.py:97-140: same batches for all candidates).
+        Default metric is the fused adanet_loss; any head metric (e.g.
+        "accuracy" with a MAXIMIZE objective) may drive selection.
         Round-robin: each rank evaluates the candidates it owns; results
         are merged by adanet_losses()-style all-gather in the caller."""
         # Accumulate per-candidate losses as DEVICE tensors: one host sync
@@ -793,8 +813,12 @@ class _Iteration(object):
                 for i, spec in enumerate(self.ensemble_specs):
                     if spec.ensemble is None:
                         continue
-                    loss = self._ensemble_adanet_loss(spec, frozen_out,
-                                                      labels, train=False)
+                    if metric_name == "adanet_loss":
+                        loss = self._ensemble_adanet_loss(
+                            spec, frozen_out, labels, train=False)
+                    else:
+                        loss = self._ensemble_eval_metric(
+                            spec, frozen_out, labels, metric_name)
                     if loss is not None:
                         sums[i] = sums[i] + (loss.detach() if isinstance(
                             loss, torch.Tensor) else loss)

@@ -641,3 +641,24 @@ def test_custom_metric_fn(model_dir, synthetic_classification):
     assert "custom_error" in res
     assert res["custom_error"] == pytest.approx(1.0 - res["accuracy"],
                                                 abs=1e-6)
+
+
+def test_evaluator_custom_metric_maximize(model_dir,
+                                          synthetic_classification):
+    """Evaluator(metric_name='accuracy', objective=MAXIMIZE) drives
+    selection by head accuracy instead of adanet_loss."""
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        evaluator=adanet_amd.Evaluator(input_fn=input_fn, steps=2,
+                                       metric_name="accuracy",
+                                       objective="maximize"))
+    est.train(input_fn, max_steps=20)
+    assert est.iteration_number == 2
+    # unknown metric raises loudly
+    est2 = _make_estimator(
+        os.path.join(model_dir, "bad"), input_fn,
+        evaluator=adanet_amd.Evaluator(input_fn=input_fn, steps=1,
+                                       metric_name="not_a_metric"))
+    with pytest.raises(ValueError):
+        est2.train(input_fn, max_steps=10)
