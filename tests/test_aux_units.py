@@ -90,3 +90,32 @@ def test_public_api_surface():
                                  MultiClassHead, MultiHead, RegressionHead)
     from adanet_amd.ops.conv import (HipConv1x1,  # noqa: F401
                                      HipConvNxN, HipDepthwiseConv2d)
+
+
+def test_runconfig_env_parsing(monkeypatch):
+    from adanet_amd.config import RunConfig
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    monkeypatch.setenv("RANK", "3")
+    c = RunConfig(tf_random_seed=5)
+    assert c.world_size == 8 and c.rank == 3
+    assert c.random_seed == 5  # reference kwarg name preserved
+    monkeypatch.delenv("WORLD_SIZE")
+    monkeypatch.delenv("RANK")
+    c2 = RunConfig()
+    assert c2.world_size == 1 and c2.rank == 0
+
+
+def test_train_manager_persistence(tmp_path):
+    """TrainManager round-trips 'done' state through its JSON dir
+    (reference iteration.py:40-118 restart semantics)."""
+    from adanet_amd.core.iteration import _TrainManager
+    tm = _TrainManager(str(tmp_path), 2, is_chief=True)
+    assert tm.should_train("cand_a")
+    tm.request_stop("cand_a", "NaN loss")
+    tm2 = _TrainManager(str(tmp_path), 2, is_chief=True)  # fresh process
+    assert not tm2.should_train("cand_a")
+    assert tm2.stopped["cand_a"] == "NaN loss"
+    assert tm2.should_train("cand_b")
+    assert not tm2.is_over(["cand_a", "cand_b"])
+    tm2.request_stop("cand_b")
+    assert tm2.is_over(["cand_a", "cand_b"])
