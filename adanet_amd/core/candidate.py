@@ -30,12 +30,16 @@ class _Candidate(object):
 
     def update(self, adanet_loss: float):
         """assign_moving_average with zero-debias-free semantics
-        (reference candidate.py:117-129)."""
+        (reference candidate.py:117-129). NaN PROPAGATES exactly like the
+        reference's EMA arithmetic: once a candidate diverges its EMA stays
+        NaN (selection maps it to -inf forever) — a later finite loss must
+        NOT resurrect it."""
         v = float(adanet_loss)
-        if math.isnan(v):
+        if math.isnan(v) or (self._ema is not None
+                             and math.isnan(self._ema)):
             self._ema = float("nan")
             return
-        if self._ema is None or math.isnan(self._ema):
+        if self._ema is None:
             self._ema = v
         else:
             self._ema = self._decay * self._ema + (1.0 - self._decay) * v

@@ -161,3 +161,24 @@ def test_device_loader_cpu_passthrough():
     out = list(loader())
     assert len(out) == 3
     assert torch.equal(out[0][0], torch.ones(2, 3))
+
+
+def test_candidate_nan_poisons_ema():
+    """A NaN loss permanently poisons a candidate's EMA so selection maps
+    it to -inf (reference _NanLossHook + NaN->-inf semantics)."""
+    from types import SimpleNamespace
+    from adanet_amd.core.candidate import _Candidate
+    c = _Candidate(SimpleNamespace(name="x"), adanet_loss_decay=0.9)
+    c.update(1.0)
+    c.update(float("nan"))
+    c.update(0.5)  # cannot recover
+    import math
+    assert math.isnan(c.adanet_loss)
+
+
+def test_mean_accumulator_weighted():
+    from adanet_amd.core.eval_metrics import _MeanAccumulator
+    a = _MeanAccumulator()
+    a.update(1.0, n=3)
+    a.update(5.0, n=1)
+    assert a.value == 2.0
