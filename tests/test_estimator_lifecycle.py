@@ -662,3 +662,22 @@ def test_evaluator_custom_metric_maximize(model_dir,
                                        metric_name="not_a_metric"))
     with pytest.raises(ValueError):
         est2.train(input_fn, max_steps=10)
+
+
+def test_input_exhaustion_ends_training(model_dir, synthetic_classification):
+    """max_iteration_steps=None + finite input: training consumes the
+    input and returns mid-iteration (reference OutOfRange semantics —
+    no bookkeeping runs for the incomplete iteration)."""
+    X, Y, input_fn = synthetic_classification
+
+    def finite_input_fn():
+        return iter([(X[:32], Y[:32])] * 7)
+
+    est = _make_estimator(model_dir, finite_input_fn,
+                          max_iteration_steps=None)
+    est.train(finite_input_fn)
+    assert est.global_step == 7
+    assert est.iteration_number == 0
+    # resume with more data completes nothing new until budget exists
+    est.train(finite_input_fn)
+    assert est.global_step == 14
