@@ -207,3 +207,44 @@ def test_dnn_linear_combined_in_pool(tmp_path):
     assert est.iteration_number == 2
     res = est.evaluate(input_fn, steps=2)
     assert res["accuracy"] >= 0.0
+
+
+def test_private_input_exhaustion_stops_candidate_only(tmp_path):
+    """A bagging candidate whose private input_fn exhausts stops with
+    'OutOfRange' while the rest of the pool trains to the iteration
+    budget (reference _SecondaryTrainOpRunnerHook end-of-input)."""
+    import glob
+    import json
+    from adanet_amd.models.canned import DNNEstimator, LinearEstimator
+    torch.manual_seed(0)
+    X = torch.randn(64, 8)
+    Y = (X.sum(1) > 0).long()
+
+    def input_fn():
+        def gen():
+            while True:
+                yield X, Y
+        return gen()
+
+    def short_private_fn():
+        return iter([(X, Y)] * 3)
+
+    head = MultiClassHead(2)
+    md = str(tmp_path / "bag")
+    est = AutoEnsembleEstimator(
+        head=head,
+        candidate_pool={
+            "short": AutoEnsembleSubestimator(
+                LinearEstimator(head=head),
+                train_input_fn=short_private_fn),
+            "long": DNNEstimator(head=head, hidden_units=[8]),
+        },
+        max_iteration_steps=10, model_dir=md,
+        config=adanet_amd.RunConfig(tf_random_seed=0))
+    est.train(input_fn, max_steps=10)
+    reasons = {
+        os.path.basename(f): json.load(open(f))["reason"]
+        for f in glob.glob(os.path.join(md, "train_manager", "t0", "*.json"))
+    }
+    assert reasons["t0_short.json"] == "OutOfRange"
+    assert reasons["t0_long.json"] == "Training is over."
