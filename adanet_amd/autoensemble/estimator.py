@@ -14,6 +14,10 @@ from adanet_amd.core.estimator import Estimator
 
 
 class AutoEnsembleEstimator(Estimator):
+    """adanet.Estimator over a pool of arbitrary models: each entry of
+    `candidate_pool` (dict/list/callable of canned estimators or
+    AutoEnsembleSubestimators) becomes a candidate Builder per iteration
+    (reference autoensemble/estimator.py:28-220)."""
 
     def __init__(self, head, candidate_pool, max_iteration_steps,
                  logits_fn=None, last_layer_fn=None, ensemblers=None,

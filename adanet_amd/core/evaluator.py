@@ -22,6 +22,10 @@ class Objective(object):
 
 
 class Evaluator(object):
+    """Scores every candidate ensemble on `steps` shared batches of
+    `input_fn` and selects argmin/argmax of `metric_name` (default
+    adanet_loss; any head metric works — e.g. accuracy with MAXIMIZE).
+    Reference adanet/core/evaluator.py:31-140."""
 
     def __init__(self, input_fn, steps: Optional[int] = None,
                  metric_name: str = "adanet_loss",

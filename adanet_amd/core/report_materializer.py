@@ -15,6 +15,9 @@ from adanet_amd.subnetwork.report import MaterializedReport, Report
 
 
 class ReportMaterializer(object):
+    """Materializes each candidate's `subnetwork.Report` over `steps`
+    batches of `input_fn` into python `MaterializedReport`s for
+    report-driven Generators (reference report_materializer.py:30-74)."""
 
     def __init__(self, input_fn, steps: Optional[int] = None):
         self._input_fn = input_fn
