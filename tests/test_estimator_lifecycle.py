@@ -681,3 +681,21 @@ def test_input_exhaustion_ends_training(model_dir, synthetic_classification):
     # resume with more data completes nothing new until budget exists
     est.train(finite_input_fn)
     assert est.global_step == 14
+
+
+def test_debug_mode_raises_on_nonfinite_input(model_dir):
+    """debug=True validates every input batch (reference debug mode,
+    estimator_test.py:3081 family)."""
+    X = torch.randn(32, 8)
+    X[3, 4] = float("inf")
+    Y = torch.zeros(32, dtype=torch.long)
+
+    def input_fn():
+        def gen():
+            while True:
+                yield X, Y
+        return gen()
+
+    est = _make_estimator(model_dir, input_fn, debug=True)
+    with pytest.raises(ValueError):
+        est.train(input_fn, max_steps=5)
