@@ -21,6 +21,7 @@ from adanet_amd.autoensemble.common import AutoEnsembleSubestimator
 from adanet_amd.autoensemble.estimator import AutoEnsembleEstimator
 from adanet_amd.config import RunConfig
 from adanet_amd.core.estimator import Estimator
+from adanet_amd.core.estimator import NanLossDuringTrainingError
 from adanet_amd.core.evaluator import Evaluator
 from adanet_amd.core.report_materializer import ReportMaterializer
 from adanet_amd.core.summary import Summary
@@ -37,6 +38,7 @@ __all__ = [
     "Estimator",
     "Evaluator",
     "MixtureWeightType",
+    "NanLossDuringTrainingError",
     "ReportMaterializer",
     "RunConfig",
     "Subnetwork",

@@ -55,6 +55,15 @@ from adanet_amd.subnetwork.generator import Generator, Subnetwork
 log = logging.getLogger("adanet_amd")
 
 
+class NanLossDuringTrainingError(RuntimeError):
+    """Raised when the selected best candidate's adanet_loss is NaN.
+
+    The reference surfaces divergence the same way: selection maps NaN to
+    -inf so the diverged candidate wins (adanet/core/iteration.py:1040-1046)
+    and tf.estimator's NaN-loss hook then raises NanLossDuringTrainingError.
+    """
+
+
 class _roctx(object):
     """rocTX range markers around engine phases (the reference
     only had ProfilerHook in a test; here ranges are first-class, shown in
@@ -97,6 +106,67 @@ def _to_device(features, labels, device, dtype=torch.bfloat16):
         else:
             labels = labels.to(device, non_blocking=True)
     return features, labels
+
+
+def _merge_candidate_losses(gathered_maps, n, objective):
+    """Merge per-rank {index: value} evaluator results into one loss list.
+
+    Unevaluated candidates (built on no rank) stay +inf so they can never
+    win argmin — under MAXIMIZE only values a rank actually produced are
+    negated (an inf placeholder negated to -inf would beat every real
+    candidate). NaN stays NaN so selection surfaces divergence.
+    """
+    losses = [float("inf")] * n
+    evaluated = set()
+    for d in gathered_maps:
+        for i, v in d.items():
+            losses[i] = v
+            evaluated.add(i)
+    if objective == "maximize":
+        losses = [
+            -v if (i in evaluated and not math.isnan(v)) else v
+            for i, v in enumerate(losses)
+        ]
+    return losses
+
+
+def _colocation_groups(builders, candidates):
+    """Map builder name -> placement group index, with builders that share
+    any ensemble candidate unioned into one group.
+
+    Round-robin placement assigns OWNERSHIP per group (not per builder), so
+    every candidate's new subnetworks are built on a single rank. With
+    single-builder candidates (Solo/Grow, the defaults) each group is one
+    builder and placement is unchanged; AllStrategy-style candidates collapse
+    their members into one group (correctness over parallelism).
+    """
+    idx = {b.name: i for i, b in enumerate(builders)}
+    parent = list(range(len(builders)))
+
+    def find(x):
+        while parent[x] != x:
+            parent[x] = parent[parent[x]]
+            x = parent[x]
+        return x
+
+    for cand in candidates:
+        members = [
+            idx[b.name] for b in cand.subnetwork_builders if b.name in idx
+        ]
+        for a, b in zip(members, members[1:]):
+            ra, rb = find(a), find(b)
+            if ra != rb:
+                parent[max(ra, rb)] = min(ra, rb)
+    roots = []
+    root_index = {}
+    group_of = {}
+    for b in builders:
+        r = find(idx[b.name])
+        if r not in root_index:
+            root_index[r] = len(roots)
+            roots.append(r)
+        group_of[b.name] = root_index[r]
+    return group_of
 
 
 class Estimator(object):
@@ -629,14 +699,42 @@ class Estimator(object):
 
         train_manager = _TrainManager(self._model_dir, t,
                                       is_chief=comm.is_chief())
+
+        # --- ensemble candidates are generated FIRST (deterministic across
+        # ranks) so placement can co-locate all of a candidate's new builders
+        # on one rank: a multi-builder candidate (AllStrategy, custom
+        # Candidates) whose members were owned by different ranks would be
+        # built on NO rank and silently never train. Builders that share a
+        # candidate are union-found into one placement group.
+        prev_builder_handles = []
+        if prev_ensemble is not None:
+            prev_arch_members = _Architecture.deserialize(
+                self._architectures[t - 1]).subnetworks
+            prev_builder_handles = [
+                _FrozenBuilderHandle("t{}_{}".format(it_num, bname), it_num,
+                                     bname)
+                for it_num, bname in prev_arch_members
+            ]
+        strategy_candidates = [
+            (strategy,
+             strategy.generate_ensemble_candidates(
+                 builders, prev_builder_handles or None))
+            for strategy in self._ensemble_strategies
+        ]
+        group_of = _colocation_groups(builders, [
+            c for _, cands in strategy_candidates for c in cands
+        ])
+        n_groups = max(group_of.values()) + 1 if group_of else 0
+
         # --- subnetwork specs (placement-gated: reference iteration.py:629) ---
         sub_specs: List[_SubnetworkSpec] = []
         builder_hooks = []
         n = len(builders)
         for i, b in enumerate(builders):
             name = "t{}_{}".format(t, b.name)
-            owner = self._placement.subnetwork_owner(n, i)
-            build_here = self._placement.should_build_subnetwork(n, i)
+            gi = group_of[b.name]
+            owner = self._placement.subnetwork_owner(n_groups, gi)
+            build_here = self._placement.should_build_subnetwork(n_groups, gi)
             summary = self._make_summary("subnetwork", b.name, t,
                                          self._enable_subnetwork_summaries)
             if build_here:
@@ -697,15 +795,6 @@ class Estimator(object):
                                     summary=summary))
 
         # --- ensemble candidates ---
-        prev_builder_handles = []
-        if prev_ensemble is not None:
-            prev_arch_members = _Architecture.deserialize(
-                self._architectures[t - 1]).subnetworks
-            prev_builder_handles = [
-                _FrozenBuilderHandle("t{}_{}".format(it_num, bname), it_num,
-                                     bname)
-                for it_num, bname in prev_arch_members
-            ]
         ens_specs: List[_EnsembleSpec] = []
         if prev_ensemble is not None:
             # Previous-best ensemble as candidate 0 (reference
@@ -734,9 +823,7 @@ class Estimator(object):
                         self._enable_ensemble_summaries)))
 
         spec_by_builder = {s.builder.name: s for s in sub_specs}
-        for strategy in self._ensemble_strategies:
-            candidates = strategy.generate_ensemble_candidates(
-                builders, prev_builder_handles or None)
+        for strategy, candidates in strategy_candidates:
             for cand in candidates:
                 for ensembler in self._ensemblers:
                     ens_specs.append(
@@ -868,14 +955,8 @@ class Estimator(object):
                 i: v for i, v in enumerate(local)
                 if iteration.ensemble_specs[i].ensemble is not None
             })
-            losses = [float("inf")] * len(local)
-            for d in merged:
-                for i, v in d.items():
-                    losses[i] = v
-            if self._evaluator.objective == "maximize":
-                losses = [
-                    -v if not math.isnan(v) else v for v in losses
-                ]
+            losses = _merge_candidate_losses(merged, len(local),
+                                             self._evaluator.objective)
             best_index = iteration.best_candidate_index(losses=losses)
         else:
             losses = iteration.adanet_losses()
@@ -897,6 +978,14 @@ class Estimator(object):
         best_index = comm.broadcast_object(best_index, src=0) if (
             comm.is_initialized()) else best_index
         chosen = iteration.ensemble_specs[best_index]
+        # A NaN winner means the selected ensemble diverged (selection maps
+        # NaN -> -inf so divergence surfaces, reference iteration.py:1040-46;
+        # tf.estimator then raises NanLossDuringTrainingError). Fail loudly
+        # instead of silently freezing a garbage model.
+        if best_index < len(losses) and math.isnan(losses[best_index]):
+            raise NanLossDuringTrainingError(
+                "Iteration {}: selected candidate {!r} has NaN adanet_loss "
+                "(training diverged).".format(t, chosen.name))
         # Per-candidate eval summaries (the analog of _EvalMetricSaverHook's
         # per-candidate eval dirs, reference estimator.py:150-233).
         for i, spec in enumerate(iteration.ensemble_specs):
@@ -1009,16 +1098,11 @@ class Estimator(object):
             # frozen members' states already stored from earlier iterations
         ens_sd = None
         if chosen.ensemble is not None:
-            ens_sd = {}
-            for k, v in chosen.ensemble.state_dict().items():
-                # Persist only the mixture parameters (member modules are
-                # stored separately per frozen_states).
-                if "weighted_subnetworks" in k and ".subnetwork" in k:
-                    continue
-                ens_sd[k] = v.detach().cpu()
-            full = {k: v.detach().cpu()
-                    for k, v in chosen.ensemble.state_dict().items()}
-            ens_sd = full
+            # Mixture parameters only: Subnetwork is a plain dataclass (not a
+            # registered submodule), so the ensemble state dict never contains
+            # member weights — those live in frozen_states, write-once.
+            ens_sd = {k: v.detach().cpu()
+                      for k, v in chosen.ensemble.state_dict().items()}
         if comm.is_initialized():
             ens_sd = comm.broadcast_object(ens_sd, src=owner)
         self._best_ensemble_state = ens_sd

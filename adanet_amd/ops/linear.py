@@ -206,10 +206,18 @@ class _LinearFn(torch.autograd.Function):
                 dw_out = wg if direct_w else torch.empty(
                     (N, K), device=dz.device, dtype=torch.bfloat16)
                 # overwrite epilogue when the optimizer skipped the arena
-                # zero (single-write params, ops/optim.py zero_grad).
-                epi = 0 if not direct_w else (
-                    0 if getattr(wparam, "_adanet_grad_overwrite", False)
-                    else 2)
+                # zero (single-write params, ops/optim.py zero_grad) — but
+                # only on the FIRST dW write this step: a module invoked
+                # twice per step (weight sharing in a custom Builder) must
+                # accumulate its second write or it would silently clobber
+                # the first. The counter resets in zero_grad.
+                if direct_w and getattr(wparam, "_adanet_grad_overwrite",
+                                        False):
+                    writes = getattr(wparam, "_adanet_bwd_writes", 0)
+                    wparam._adanet_bwd_writes = writes + 1
+                    epi = 0 if writes == 0 else 2
+                else:
+                    epi = 0 if not direct_w else 2
                 if (B % 32 == 0 and dz.stride(0) % 8 == 0
                         and x.stride(0) % 8 == 0):
                     ext.gemm_tr_bf16(dz, x, dw_out, None, epi, 1, 1)

@@ -121,6 +121,7 @@ class _ArenaMixin(object):
                     a["flat_g"].zero_()
                 for p, off, k in a["views"]:
                     p._adanet_grad_overwrite = skip
+                    p._adanet_bwd_writes = 0
                     if p.grad is None or p.grad.data_ptr() != a[
                             "flat_g"][off:off + k].data_ptr():
                         p.grad = a["flat_g"][off:off + k].view_as(p.data)

@@ -182,3 +182,49 @@ def test_mean_accumulator_weighted():
     a.update(1.0, n=3)
     a.update(5.0, n=1)
     assert a.value == 2.0
+
+
+def test_merge_candidate_losses_maximize_skips_placeholders():
+    """MAXIMIZE must not negate the +inf placeholders of candidates no rank
+    evaluated — an unbuilt candidate could otherwise win argmin as -inf."""
+    from adanet_amd.core.estimator import _merge_candidate_losses
+    merged = _merge_candidate_losses([{0: 0.9}, {2: 0.7}], 4, "maximize")
+    assert merged[0] == -0.9
+    assert merged[1] == float("inf")  # unevaluated: stays +inf
+    assert merged[2] == -0.7
+    assert merged[3] == float("inf")
+    # NaN stays NaN under maximize (divergence surfaces in selection).
+    merged = _merge_candidate_losses([{1: float("nan")}], 2, "maximize")
+    assert math.isnan(merged[1])
+    # minimize: plain merge.
+    merged = _merge_candidate_losses([{0: 0.5}], 2, "minimize")
+    assert merged == [0.5, float("inf")]
+
+
+def test_colocation_groups_union_by_candidate():
+    """Builders sharing an ensemble candidate collapse into one placement
+    group so round-robin never splits a candidate across ranks."""
+    from adanet_amd.core.estimator import _colocation_groups
+
+    class _B:
+        def __init__(self, name):
+            self.name = name
+
+    class _C:
+        def __init__(self, builders):
+            self.subnetwork_builders = builders
+
+    b = [_B("a"), _B("b"), _B("c"), _B("d")]
+    # Grow-style: each candidate has one new builder -> identity groups.
+    groups = _colocation_groups(b, [_C([b[0]]), _C([b[1]]), _C([b[2]]),
+                                    _C([b[3]])])
+    assert sorted(set(groups.values())) == [0, 1, 2, 3]
+    # All-style: one candidate spans b and c -> they share a group.
+    groups = _colocation_groups(b, [_C([b[0]]), _C([b[1], b[2]]),
+                                    _C([b[3]])])
+    assert groups["b"] == groups["c"]
+    assert len(set(groups.values())) == 3
+    # Transitive union: (a,b) + (b,c) -> {a,b,c} one group.
+    groups = _colocation_groups(b, [_C([b[0], b[1]]), _C([b[1], b[2]])])
+    assert groups["a"] == groups["b"] == groups["c"]
+    assert groups["d"] != groups["a"]

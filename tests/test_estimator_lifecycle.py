@@ -210,24 +210,22 @@ class _GoodBuilder(Builder):
         return Subnetwork(module=_M(), complexity=1.0)
 
 
-def test_nan_candidate_does_not_crash_and_good_candidate_wins(
-        model_dir, synthetic_classification):
-    """A NaN candidate must not break training when an Evaluator selects on
-    eval data; selection via np.argmin surfaces NaN (reference semantics)
-    but the Evaluator path tolerates it (estimator.py:386-439 disables the
-    asserts). Here: the NaN candidate's EMA poisons, the evaluator (which
-    sees NaN too) is not used, and selection WITHOUT evaluator picks the
-    NaN candidate (NaN -> -inf, reference iteration.py:1040-1046)."""
+def test_nan_winner_raises_nan_loss_error(model_dir,
+                                          synthetic_classification):
+    """Selection maps NaN -> -inf so a diverged candidate WINS argmin
+    (reference iteration.py:1040-1046) — and the engine must then fail
+    loudly, like tf.estimator's NanLossDuringTrainingError, instead of
+    silently freezing a garbage model."""
+    import pytest
     X, Y, input_fn = synthetic_classification
     est = _make_estimator(
         model_dir, input_fn,
         subnetwork_generator=SimpleGenerator(
             [_NanBuilder(), _GoodBuilder()]))
-    est.train(input_fn, max_steps=10)
-    arch = json.loads(
-        open(os.path.join(model_dir, "architecture-0.json")).read())
-    # NaN -> -inf wins argmin: divergence is surfaced in the architecture.
-    assert arch["subnetworks"][0]["builder_name"] == "nan"
+    with pytest.raises(adanet_amd.NanLossDuringTrainingError):
+        est.train(input_fn, max_steps=10)
+    # Divergence surfaced before any architecture was frozen.
+    assert not os.path.exists(os.path.join(model_dir, "architecture-0.json"))
 
 
 def test_evaluator_selects_on_eval_loss(model_dir, synthetic_classification):
