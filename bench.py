@@ -32,8 +32,10 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1,
-                   help="accepted for the driver contract; the actual "
-                        "world size comes from the torchrun env")
+                   help="world size; when run without a torchrun env and "
+                        "N>1, bench.py self-launches under "
+                        "torch.distributed.run so a plain `python bench.py "
+                        "--gpus 8` measures 8 GPUs")
     p.add_argument("--steps", type=int, default=3,
                    help="AdaNet iterations to time")
     p.add_argument("--warmup", type=int, default=1,
@@ -53,8 +55,26 @@ def parse_args():
     return p.parse_args()
 
 
+def _self_launch(args):
+    """Re-exec under torchrun: one rank per GPU (driver contract parity)."""
+    import socket
+    import subprocess
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+        "--nproc-per-node", str(args.gpus), "--master-addr", "127.0.0.1",
+        "--master-port", str(port),
+        os.path.abspath(__file__),
+    ] + sys.argv[1:]
+    sys.exit(subprocess.call(cmd))
+
+
 def main():
     args = parse_args()
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        _self_launch(args)
     if os.environ.get("ADANET_LOG"):
         import logging
         logging.basicConfig(level=os.environ["ADANET_LOG"])
@@ -71,6 +91,10 @@ def main():
     use_gpu = torch.cuda.is_available() and not args.cpu
     comm.maybe_init_process_group()
     world = comm.world_size()
+    if "WORLD_SIZE" in os.environ and world != args.gpus:
+        raise SystemExit(
+            "bench.py --gpus {} but torchrun world size is {} — the "
+            "reported n_gpus would lie".format(args.gpus, world))
     rank = comm.rank()
     local_rank = int(os.environ.get("LOCAL_RANK", rank))
     device = torch.device("cuda", local_rank) if use_gpu else torch.device(

@@ -76,3 +76,34 @@ def test_bench_torchrun_replication_cpu(tmp_path):
     d = json.loads(lines[0])
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+
+
+def test_bench_self_launches_torchrun(tmp_path):
+    """A plain `python bench.py --gpus 2` (no torchrun env) must self-launch
+    under torch.distributed.run and report n_gpus=2 — never silently measure
+    one rank (VERDICT r01 weak #6)."""
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--gpus", "2",
+         "--cpu", "--steps", "1", "--warmup", "0", "--batch", "64",
+         "--hidden", "32", "--train-steps-per-iter", "3",
+         "--eval-batches", "2"],
+        capture_output=True, text=True, timeout=900, cwd=str(tmp_path))
+    assert out.returncode == 0, (out.stdout[-1000:], out.stderr[-2000:])
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-2000:]
+    assert json.loads(lines[0])["n_gpus"] == 2
+
+
+def test_bench_world_size_mismatch_fails(tmp_path):
+    """Under a torchrun env whose world size contradicts --gpus, bench.py
+    must fail instead of reporting a wrong n_gpus."""
+    port = _free_port()
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", "4", "--cpu", "--steps", "1", "--warmup", "0",
+         "--batch", "64", "--hidden", "32", "--train-steps-per-iter", "3",
+         "--eval-batches", "2"],
+        capture_output=True, text=True, timeout=900, cwd=str(tmp_path))
+    assert out.returncode != 0
