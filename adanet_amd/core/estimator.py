@@ -41,8 +41,9 @@ import torch
 from adanet_amd.config import RunConfig
 from adanet_amd.core.architecture import _Architecture
 from adanet_amd.core.evaluator import Evaluator
-from adanet_amd.core.iteration import (_EnsembleSpec, _Iteration,
-                                       _SubnetworkSpec, _TrainManager)
+from adanet_amd.core.iteration import (_EnsembleSpec, _FrozenLogitCache,
+                                       _Iteration, _SubnetworkSpec,
+                                       _TrainManager)
 from adanet_amd.core.report_accessor import _ReportAccessor
 from adanet_amd.core.summary import _ScopedSummary
 from adanet_amd.distributed import comm
@@ -284,6 +285,10 @@ class Estimator(object):
         self._best_ensemble_state: Optional[dict] = None
         self._replay_indices: List[int] = []
         self._current_iteration: Optional[_Iteration] = None
+        # Cross-iteration HBM cache of frozen-member outputs on resident
+        # batches: frozen weights are immutable, so only the newly frozen
+        # winner is ever a miss (keeps per-iteration cost O(new members)).
+        self._frozen_logit_cache = _FrozenLogitCache()
         self._phase_secs = {"build": 0.0, "train": 0.0, "bookkeeping": 0.0,
                             "checkpoint": 0.0}
         self._restore_checkpoint()
@@ -842,7 +847,8 @@ class Estimator(object):
             replicate_ensemble_in_training=(
                 self._replicate_ensemble_in_training),
             to_device=lambda f, l: _to_device(f, l, self._device),
-            use_graphs=self._use_hip_graphs)
+            use_graphs=self._use_hip_graphs,
+            frozen_logit_cache=self._frozen_logit_cache)
         iteration.builder_hooks = builder_hooks
         self._restore_iteration_state(iteration)
         self._current_iteration = iteration

@@ -45,6 +45,49 @@ log = logging.getLogger("adanet_amd")
 _LOSS_FLUSH_STEPS = 16
 
 
+class _FrozenLogitCache(object):
+    """HBM-resident cache of frozen-member outputs, keyed
+    (batch_cache_key, member_name) -> (last_layer | None, logits).
+
+    Frozen member weights are immutable once frozen, so entries stay valid
+    ACROSS AdaNet iterations — only the newly frozen winner of iteration
+    t-1 is a miss at iteration t. This is what flattens the per-member
+    cost curve: without it every iteration recomputes all J members on
+    every resident batch (O(J) forwards per iteration, the dominant growth
+    term the round-1 driver measured at large ensembles). Owned by the
+    Estimator; _Iteration holds a reference. Insertion stops at `cap`
+    bytes (ADANET_FROZEN_CACHE_BYTES, default 32 GiB of the 288 GB HBM).
+    """
+
+    def __init__(self, cap_bytes: Optional[int] = None):
+        if cap_bytes is None:
+            cap_bytes = int(
+                os.environ.get("ADANET_FROZEN_CACHE_BYTES",
+                               32 * 1024 ** 3))
+        self.cap = cap_bytes
+        self.bytes = 0
+        self._d: Dict = {}
+
+    def get(self, key):
+        return self._d.get(key)
+
+    def put(self, key, last, logits):
+        sz = logits.numel() * logits.element_size() + (
+            last.numel() * last.element_size() if last is not None else 0)
+        old = self._d.get(key)
+        if old is not None:
+            self.bytes -= (old[1].numel() * old[1].element_size() +
+                           (old[0].numel() * old[0].element_size()
+                            if old[0] is not None else 0))
+        if self.bytes + sz > self.cap:
+            return
+        self._d[key] = (last, logits)
+        self.bytes += sz
+
+    def __len__(self):
+        return len(self._d)
+
+
 class _TrainManager(object):
     """Persists per-spec "done training" state as JSON files.
 
@@ -150,7 +193,8 @@ class _Iteration(object):
                  device: torch.device, placement, use_streams: bool = True,
                  replicate_ensemble_in_training: bool = False,
                  to_device: Optional[Callable] = None,
-                 use_graphs: bool = True):
+                 use_graphs: bool = True,
+                 frozen_logit_cache: Optional[_FrozenLogitCache] = None):
         self.number = number
         self.head = head
         self.subnetwork_specs: List[_SubnetworkSpec] = list(subnetwork_specs)
@@ -187,8 +231,11 @@ class _Iteration(object):
         self._nan_scalar = None
         self._frozen_event = None
         self.builder_hooks = []  # TrainOpSpec hooks collected by the engine
-        # HBM frozen-logit cache (per-iteration lifetime).
-        self._frozen_cache: Dict = {}
+        # HBM frozen-logit cache: shared across iterations when the engine
+        # passes its own (frozen member outputs are immutable forever).
+        self._frozen_cache = (frozen_logit_cache
+                              if frozen_logit_cache is not None
+                              else _FrozenLogitCache())
         self._frozen_static = None
         self._need_last = None
         # hipGraph state
@@ -256,48 +303,69 @@ class _Iteration(object):
                      and not (training
                               and self.replicate_ensemble_in_training))
         keep_last = self._need_frozen_last_layer()
-        if cacheable and key in self._frozen_cache:
-            cached = self._frozen_cache[key]
-            srcs, dsts = [], []
-            for name, (last_c, logits_c) in cached.items():
-                last_s, logits_s = self._frozen_static[name]
-                srcs.append(logits_c)
-                dsts.append(logits_s)
-                if keep_last and last_c is not None:
-                    srcs.append(last_c)
-                    dsts.append(last_s)
-            if srcs and srcs[0].is_cuda and all(
-                    s.dtype == torch.bfloat16 and s.is_contiguous()
-                    and d.is_contiguous() for s, d in zip(srcs, dsts)):
-                # One batched-copy launch instead of J-1 copyBuffer calls.
-                from adanet_amd.ops import _extension
-                _extension.require().multi_copy_bf16(srcs, dsts)
-            else:
-                for s, d in zip(srcs, dsts):
-                    d.copy_(s)
+        if not cacheable:
+            fresh = self.compute_frozen_outputs(features, training=training)
+            outs = {n: (last if keep_last else None, logits)
+                    for n, (last, logits) in fresh.items()}
+            self._stage_frozen(outs)
             return self._frozen_static
-        fresh = self.compute_frozen_outputs(features, training=training)
+        # Per-member cross-iteration cache: only members not yet seen on
+        # this batch (i.e. the newly frozen winner) run a forward.
+        outs = {}
+        missing = []
+        for name in self.frozen_subnetworks:
+            ent = self._frozen_cache.get((key, name))
+            if ent is None or (keep_last and ent[0] is None):
+                missing.append(name)
+            else:
+                outs[name] = ent
+        if missing:
+            with torch.no_grad():
+                for name in missing:
+                    sub = self.frozen_subnetworks[name]
+                    sub.module.eval()
+                    last, logits = sub(features)
+                    ent = (last.detach().clone() if keep_last else None,
+                           logits.detach().clone())
+                    self._frozen_cache.put((key, name), *ent)
+                    outs[name] = ent
+        self._stage_frozen(outs)
+        return self._frozen_static
+
+    def _stage_frozen(self, outs: Dict):
+        """Copy per-member (last|None, logits) into the STATIC buffers a
+        captured hipGraph reads; (re)allocates on first use / shape change."""
         if self._frozen_static is None or any(
-                self._frozen_static[n][1].shape != o[1].shape
-                for n, o in fresh.items()):
+                n not in self._frozen_static
+                or self._frozen_static[n][1].shape != o[1].shape
+                or (o[0] is not None and self._frozen_static[n][0] is None)
+                for n, o in outs.items()):
             self._frozen_static = {
-                n: (last.clone(), logits.clone())
-                for n, (last, logits) in fresh.items()
+                n: [last.clone() if last is not None else None,
+                    logits.clone()]
+                for n, (last, logits) in outs.items()
             }
             # A captured graph reads the OLD static buffers — invalidate it
             # (batch-shape change mid-iteration).
             self._graph = None
+            return
+        srcs, dsts = [], []
+        for name, (last_c, logits_c) in outs.items():
+            last_s, logits_s = self._frozen_static[name]
+            srcs.append(logits_c)
+            dsts.append(logits_s)
+            if last_c is not None and last_s is not None:
+                srcs.append(last_c)
+                dsts.append(last_s)
+        if srcs and srcs[0].is_cuda and all(
+                s.dtype == torch.bfloat16 and s.is_contiguous()
+                and d.is_contiguous() for s, d in zip(srcs, dsts)):
+            # One batched-copy launch instead of J-1 copyBuffer calls.
+            from adanet_amd.ops import _extension
+            _extension.require().multi_copy_bf16(srcs, dsts)
         else:
-            for n, (last, logits) in fresh.items():
-                last_s, logits_s = self._frozen_static[n]
-                logits_s.copy_(logits)
-                last_s.copy_(last)
-        if cacheable:
-            self._frozen_cache[key] = {
-                n: ((last.clone() if keep_last else None), logits.clone())
-                for n, (last, logits) in fresh.items()
-            }
-        return self._frozen_static
+            for s, d in zip(srcs, dsts):
+                d.copy_(s)
 
     def train_step(self, features, labels) -> None:
         """One lockstep training step for every still-active spec

@@ -383,6 +383,73 @@ def test_frozen_logit_cache_correctness(model_dir, synthetic_classification):
             r2["architecture/adanet/ensembles"])
 
 
+def test_frozen_logit_cache_is_cross_iteration(model_dir,
+                                               synthetic_classification):
+    """A frozen member's forward runs ONCE per resident batch over the
+    WHOLE run, not once per iteration: the cache outlives the iteration
+    (per-iteration cost stays O(new members) as the ensemble grows)."""
+    X, Y, input_fn = synthetic_classification
+    forward_counts = {}
+
+    class _CountingBuilder(_GoodBuilder):
+
+        def build_subnetwork(self, features, logits_dimension, training,
+                             previous_ensemble=None):
+            sub = super().build_subnetwork(features, logits_dimension,
+                                           training, previous_ensemble)
+            mod = sub.module
+            orig_forward = mod.forward
+            bname = self.name
+
+            def counted(x, _orig=orig_forward, _b=bname):
+                if not mod.training and not torch.is_grad_enabled():
+                    forward_counts[_b] = forward_counts.get(_b, 0) + 1
+                return _orig(x)
+
+            mod.forward = counted
+            return sub
+
+    n_batches = 4
+
+    def keyed_input_fn():
+        def gen():
+            g = torch.Generator().manual_seed(3)
+            batches = []
+            for i in range(n_batches):
+                idx = torch.randint(0, X.shape[0], (32,), generator=g)
+                xb = X[idx].clone()
+                xb.adanet_cache_key = ("kb", i)
+                batches.append((xb, Y[idx]))
+            i = 0
+            while True:
+                yield batches[i % n_batches]
+                i += 1
+
+        return gen()
+
+    est = _make_estimator(
+        model_dir, keyed_input_fn,
+        subnetwork_generator=SimpleGenerator([_CountingBuilder("g0")]),
+        max_iteration_steps=n_batches, force_grow=True)
+    baseline = None
+    # 4 iterations: the member frozen at iteration 0 must NOT be re-run
+    # at iterations 2 and 3 (its logits replay from the HBM cache).
+    for it in range(4):
+        forward_counts.clear()
+        est.train(keyed_input_fn, steps=n_batches)
+        if it == 1:
+            # first iteration with a frozen member: populates the cache
+            baseline = dict(forward_counts)
+            assert baseline, "expected frozen-member eval forwards"
+        if it >= 2:
+            # all resident batches already cached for old members; only
+            # the NEWLY frozen winner of iteration it-1 misses.
+            total = sum(forward_counts.values())
+            assert total <= sum(baseline.values()), (
+                "frozen forwards grew with ensemble size: %r (baseline %r)"
+                % (forward_counts, baseline))
+
+
 def test_builder_prune_previous_ensemble(model_dir,
                                          synthetic_classification):
     """Legacy pruning hook (reference ensemble_builder.py:371-395): a
