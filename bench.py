@@ -52,6 +52,9 @@ def parse_args():
     p.add_argument("--lr", type=float, default=0.05)
     p.add_argument("--model-dir", default=None)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
+    p.add_argument("--per-iter-json", default=None,
+                   help="write a per-iteration wall-clock curve (ms vs "
+                        "ensemble size + winner arch) to this path, rank 0")
     return p.parse_args()
 
 
@@ -189,13 +192,38 @@ def main():
             torch.cuda.synchronize(device)
         comm.barrier()
 
+    curve = []
+
+    def record_iter(dt_ms, timed):
+        t = est.iteration_number - 1
+        entry = {"iteration": t, "ms": round(dt_ms, 2), "timed": timed}
+        try:
+            arch = json.load(
+                open(os.path.join(model_dir,
+                                  "architecture-%d.json" % t)))
+            entry["winner"] = arch["subnetworks"][-1]["builder_name"]
+            entry["ensemble_size"] = len(arch["subnetworks"])
+        except Exception:
+            pass
+        entry["phase_secs"] = {k: round(v, 3)
+                               for k, v in est._phase_secs.items()}
+        curve.append(entry)
+
     for _ in range(args.warmup):
+        ti = time.perf_counter()
         one_iteration()
+        if args.per_iter_json:
+            sync()
+            record_iter((time.perf_counter() - ti) * 1000.0, False)
 
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
+        ti = time.perf_counter()
         one_iteration()
+        if args.per_iter_json:
+            sync()
+            record_iter((time.perf_counter() - ti) * 1000.0, True)
     sync()
     elapsed = time.perf_counter() - t0
 
@@ -241,6 +269,9 @@ def main():
             },
         }
         print(json.dumps(out))
+        if args.per_iter_json:
+            with open(args.per_iter_json, "w") as f:
+                json.dump({"curve": curve}, f, indent=1)
     if args.model_dir is None:
         shutil.rmtree(model_dir, ignore_errors=True)
     if comm.is_initialized():
