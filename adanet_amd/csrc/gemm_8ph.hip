@@ -1,0 +1,293 @@
+// 8-phase deep-pipelined bf16 MFMA GEMM for gfx950: C[M,N] = A[M,K]*B[N,K]^T.
+//
+// The round-1 kernel (gemm.hip) uses the 2-barrier-per-K-step structure whose
+// measured ceiling is ~900 TF: __syncthreads() drains vmcnt(0) so the staged
+// tile pipeline empties at every barrier. This kernel implements the CDNA4
+// guide's 256^2 8-phase schedule (measured 1563 TF @4096^3 in the guide's
+// m201) re-derived for this codebase:
+//
+//   * BK=64 K-tiles, double-buffered; each tile split into 4 half-tiles
+//     (A rows [0,BM/2), [BM/2,BM); B likewise) staged by block-wide
+//     global_load_lds rounds (512 threads x 16 B = 64 rows of 128 B/round).
+//   * 4 compute phases per K-tile; phase p computes m-reps {p*FM/4..} x all
+//     n-reps x 2 k-steps. B fragments are ds_read ONCE per K-tile (phase 0)
+//     and held in VGPRs; A fragments are read per phase.
+//   * raw s_barrier (NOT __syncthreads) so the global_load_lds queue is
+//     never drained at barriers; an explicit counted s_waitcnt vmcnt(N)
+//     runs ONCE per K-tile (phase 3), leaving the two youngest half-tile
+//     prefetches in flight across the tile boundary.
+//   * prefetch schedule (provable invariant, see below): during tile t,
+//     phases 0/1 issue A-halves of tile t+1 (other LDS buffer); phases 2/3
+//     issue B-halves of tile t+2 (SAME buffer as t: B slots are dead after
+//     t's phase 0 because B fragments live in registers).
+//       issue order: ... B0(t) B1(t) A0(t) A1(t) B0(t+1) B1(t+1) ...
+//       at tile t-1 phase 3: s_waitcnt vmcnt(2*LB) leaves exactly the
+//       B(t+1) pair outstanding => every half of tile t has landed, and
+//       each wave's wait + the phase barrier publishes all waves' staging.
+//   * LDS st-swizzle: 16-byte block index XOR (row&7) kills the 128 B
+//     row-stride bank conflict on fragment ds_read_b128 (16-way -> 2-way;
+//     2-way is free on CDNA4). global_load_lds writes LDS linearly, so the
+//     swizzle is applied by permuting each lane's GLOBAL source block
+//     (involution) and reading LDS with the same XOR.
+//   * s_setprio(1) around the MFMA cluster (guide T5: +21-25% on this
+//     structure), XCD-aware bijective tile swizzle (T1).
+//
+// Reference dependency (K1): tf.layers.dense/tf.matmul hot path,
+// adanet/examples/simple_dnn.py:74-86. SAFE=true template variant keeps
+// __syncthreads-style drains for A/B correctness bisection.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include "common.h"
+
+namespace {
+
+// Stage one half-tile (HROWS x 64 bf16) into LDS with the block's 8 waves.
+// Linear LDS dest (global_load_lds constraint); swizzled global source.
+// HROWS in {64, 128} -> 1 or 2 rounds; each lane issues HROWS/64 loads.
+template <int HROWS>
+__device__ __forceinline__ void stage_half(
+    const bf16_t* __restrict__ G, int64_t ld, int rows0, int max_row, int k0,
+    bf16_t* __restrict__ lds_half, int wid, int lane) {
+#pragma unroll
+  for (int r = 0; r < HROWS / 64; ++r) {
+    const int row_in_half = r * 64 + wid * 8 + (lane >> 3);
+    int grow = rows0 + row_in_half;
+    grow = grow < max_row ? grow : max_row - 1;  // clamp; masked on C-store
+    const int src_blk = (lane & 7) ^ ((lane >> 3) & 7);  // st-swizzle
+    const bf16_t* gp = G + (int64_t)grow * ld + k0 + src_blk * 8;
+    bf16_t* lp = lds_half + (r * 64 + wid * 8) * 64;  // wave-uniform base
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+// Swizzled LDS fragment read: logical (row, 8-element block) of a [ROWS][64]
+// bf16 tile whose 16 B blocks were XOR-permuted by (row&7) at staging.
+__device__ __forceinline__ s16x8 lds_frag(const bf16_t* __restrict__ lds,
+                                          int row, int blk8) {
+  const int pblk = blk8 ^ (row & 7);
+  return *(const s16x8*)&lds[row * 64 + pblk * 8];
+}
+
+#define S_BARRIER() __builtin_amdgcn_s_barrier()
+
+// s_waitcnt vmcnt(N) with a compile-time literal (asm strings can't expand
+// constexpr values; dispatch over the small set of counts this kernel uses).
+template <int N>
+__device__ __forceinline__ void vmcnt_wait() {
+  static_assert(N == 0 || N == 2 || N == 4, "unsupported vmcnt");
+  if constexpr (N == 0) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else if constexpr (N == 2) {
+    asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  } else if constexpr (N == 4) {
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  }
+}
+
+// BM x BN output tile, 8 waves (2 M x 4 N), BK=64, 4 phases per K-tile.
+// SAFE: drain-everything barriers (correctness bisection baseline).
+template <int BM, int BN, bool SAFE = false, bool SETPRIO = true>
+__global__ __launch_bounds__(512, 2) void gemm_nt8_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
+    int ntiles) {
+  constexpr int FM = BM / (2 * 16);   // m-fragments per wave
+  constexpr int FN = BN / (4 * 16);   // n-fragments per wave
+  constexpr int MG = FM / 4;          // m-reps per phase
+  static_assert(FM >= 4, "need >=1 m-rep per phase");
+  constexpr int HA = BM / 2;          // A half-tile rows
+  constexpr int HB = BN / 2;          // B half-tile rows
+  constexpr int LB = HB / 64;         // loads/lane per B half-tile
+  __shared__ bf16_t As[2][BM * 64];
+  __shared__ bf16_t Bs[2][BN * 64];
+
+  // Bijective XCD-aware swizzle (8 XCDs).
+  const int nwg = mtiles * ntiles;
+  const int orig = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
+                 (orig >> 3);
+  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  const int row0 = tile_m * BM, col0 = tile_n * BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wm = wid >> 2, wn = wid & 3;  // 2 x 4 wave grid
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;  // K-tiles (host guarantees K % 64 == 0)
+
+  // ---- prologue: tile 0 fully + B(1); leave B(1) pair in flight ----
+  stage_half<HA>(A, lda, row0, M, 0, &As[0][0], wid, lane);
+  stage_half<HA>(A, lda, row0 + HA, M, 0, &As[0][HA * 64], wid, lane);
+  stage_half<HB>(B, ldb, col0, N, 0, &Bs[0][0], wid, lane);
+  stage_half<HB>(B, ldb, col0 + HB, N, 0, &Bs[0][HB * 64], wid, lane);
+  if (NT > 1) {
+    stage_half<HB>(B, ldb, col0, N, 64, &Bs[1][0], wid, lane);
+    stage_half<HB>(B, ldb, col0 + HB, N, 64, &Bs[1][HB * 64], wid, lane);
+    if (SAFE) vmcnt_wait<0>(); else vmcnt_wait<2 * LB>();
+  } else {
+    vmcnt_wait<0>();
+  }
+  S_BARRIER();
+
+  const int arow_base = wm * (FM * 16) + (lane & 15);
+  const int brow_base = wn * (FN * 16) + (lane & 15);
+  const int kblk = (lane >> 4);  // 8-element k-block within a 32-k-step
+
+  for (int t = 0; t < NT; ++t) {
+    const int buf = t & 1;
+    const bf16_t* at = &As[buf][0];
+    const bf16_t* bt = &Bs[buf][0];
+    s16x8 bfrag[FN][2];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      // --- ds-load this phase's register subtile ---
+      s16x8 afrag[MG][2];
+#pragma unroll
+      for (int g = 0; g < MG; ++g) {
+        const int m = p * MG + g;
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          afrag[g][kk] =
+              lds_frag(at, arow_base + m * 16, kk * 4 + kblk);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int n = 0; n < FN; ++n)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            bfrag[n][kk] =
+                lds_frag(bt, brow_base + n * 16, kk * 4 + kblk);
+      }
+      // --- issue this phase's half-tile prefetch ---
+      if (p == 0 && t + 1 < NT) {
+        stage_half<HA>(A, lda, row0, M, (t + 1) * 64, &As[buf ^ 1][0], wid,
+                       lane);
+      } else if (p == 1 && t + 1 < NT) {
+        stage_half<HA>(A, lda, row0 + HA, M, (t + 1) * 64,
+                       &As[buf ^ 1][HA * 64], wid, lane);
+      } else if (p == 2 && t + 2 < NT) {
+        stage_half<HB>(B, ldb, col0, N, (t + 2) * 64, &Bs[buf][0], wid,
+                       lane);
+      } else if (p == 3 && t + 2 < NT) {
+        stage_half<HB>(B, ldb, col0 + HB, N, (t + 2) * 64,
+                       &Bs[buf][HB * 64], wid, lane);
+      }
+      if (p == 3) {
+        // Once per K-tile: everything except the B(t+2) pair must land.
+        if (SAFE) vmcnt_wait<0>(); else vmcnt_wait<2 * LB>();
+      }
+      if (SAFE) {
+        __syncthreads();
+      } else {
+        S_BARRIER();
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
+      // --- MFMA cluster ---
+      if (SETPRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int g = 0; g < MG; ++g) {
+        const int m = p * MG + g;
+#pragma unroll
+        for (int n = 0; n < FN; ++n)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[g][kk], bfrag[n][kk], acc[m][n], 0, 0, 0);
+      }
+      if (SETPRIO) __builtin_amdgcn_s_setprio(0);
+      if (SAFE) {
+        __syncthreads();
+      } else {
+        S_BARRIER();
+      }
+    }
+  }
+
+  // ---- epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + reg ----
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr] + bv;
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        bf16_t* cp = &C[(int64_t)row * ldc + col];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// Host entry: launches the 8-phase kernel for a given tile config.
+// variant: 0 = 256x256, 1 = 256x128, 2 = 128x256, 3 = 128x128,
+//          +10 = SAFE (drain) twin, +20 = no-setprio twin.
+void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+              const c10::optional<at::Tensor>& bias, int64_t act,
+              int64_t variant) {
+  TORCH_CHECK(A.is_cuda() && B.is_cuda() && C.is_cuda(), "gemm8: GPU only");
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16, "gemm8: bf16 inputs");
+  const int M = (int)A.size(0), K = (int)A.size(1), N = (int)B.size(0);
+  TORCH_CHECK(B.size(1) == K && C.size(0) == M && C.size(1) == N,
+              "gemm8: shape mismatch");
+  TORCH_CHECK(K % 64 == 0, "gemm8: K % 64 != 0");
+  const int64_t lda = A.stride(0), ldb = B.stride(0), ldc = C.stride(0);
+  TORCH_CHECK(A.stride(1) == 1 && B.stride(1) == 1 && C.stride(1) == 1 &&
+              lda % 8 == 0 && ldb % 8 == 0, "gemm8: need 16B-aligned rows");
+  const float* bias_ptr = nullptr;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->scalar_type() == at::kFloat && bias->numel() == N,
+                "gemm8: bias must be fp32[N]");
+    bias_ptr = bias->data_ptr<float>();
+  }
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+
+#define LAUNCH_8PH(BM, BN, SAFE, PRIO)                                        \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_nt8_kernel<BM, BN, SAFE, PRIO>), dim3(mt * nt),  \
+                       dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M,   \
+                       N, K, lda, ldb, ldc, (int)act, mt, nt);                \
+  } while (0)
+
+  switch (variant) {
+    case 0: LAUNCH_8PH(256, 256, false, true); break;
+    case 1: LAUNCH_8PH(256, 128, false, true); break;
+    case 2: LAUNCH_8PH(128, 256, false, true); break;
+    case 3: LAUNCH_8PH(128, 128, false, true); break;
+    case 10: LAUNCH_8PH(256, 256, true, true); break;
+    case 11: LAUNCH_8PH(256, 128, true, true); break;
+    case 12: LAUNCH_8PH(128, 256, true, true); break;
+    case 13: LAUNCH_8PH(128, 128, true, true); break;
+    case 20: LAUNCH_8PH(256, 256, false, false); break;
+    case 23: LAUNCH_8PH(128, 128, false, false); break;
+    default: TORCH_CHECK(false, "gemm8: unknown variant");
+  }
+#undef LAUNCH_8PH
+  HIP_CHECK_KERNEL();
+}

@@ -16,6 +16,7 @@ from torch.utils import cpp_extension  # noqa: E402
 SRC = [
     "adanet_amd/csrc/binding.cpp",
     "adanet_amd/csrc/gemm.hip",
+    "adanet_amd/csrc/gemm_8ph.hip",
     "adanet_amd/csrc/gemm_tn.hip",
     "adanet_amd/csrc/transpose.hip",
     "adanet_amd/csrc/softmax_xent.hip",
