@@ -65,10 +65,23 @@ __device__ __forceinline__ void stage_half(
 
 // Swizzled LDS fragment read: logical (row, 8-element block) of a [ROWS][64]
 // bf16 tile whose 16 B blocks were XOR-permuted by (row&7) at staging.
-__device__ __forceinline__ s16x8 lds_frag(const bf16_t* __restrict__ lds,
-                                          int row, int blk8) {
+//
+// The read is INLINE ASM on purpose: a C++ LDS load after a pending
+// global_load_lds makes the compiler insert s_waitcnt vmcnt(0) before it
+// (it cannot prove the in-flight DMA writes a different LDS region), which
+// drains the prefetch pipeline at every phase — measured 927 vs the
+// counted-vmcnt schedule this kernel is built around. The schedule's own
+// vmcnt/barrier invariant is what guarantees the data has landed. Callers
+// MUST wait lgkmcnt(0) + sched_barrier(0) before consuming the result
+// (asm ds_read data arrives asynchronously; the compiler doesn't know).
+__device__ __forceinline__ s16x8 lds_frag(const bf16_t* lds, int row,
+                                          int blk8) {
   const int pblk = blk8 ^ (row & 7);
-  return *(const s16x8*)&lds[row * 64 + pblk * 8];
+  const uint32_t addr = (uint32_t)(uintptr_t)(
+      const __attribute__((address_space(3))) void*)&lds[row * 64 + pblk * 8];
+  s16x8 out;
+  asm volatile("ds_read_b128 %0, %1" : "=v"(out) : "v"(addr));
+  return out;
 }
 
 #define S_BARRIER() __builtin_amdgcn_s_barrier()
@@ -191,8 +204,12 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8_kernel(
         __syncthreads();
       } else {
         S_BARRIER();
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       }
+      // asm ds_read results land only after lgkmcnt; the sched_barrier
+      // keeps the compiler from hoisting the (register-only) MFMAs above
+      // the wait (guide rule 18).
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
       // --- MFMA cluster ---
       if (SETPRIO) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
