@@ -90,11 +90,15 @@ __device__ __forceinline__ s16x8 lds_frag(const bf16_t* lds, int row,
 // constexpr values; dispatch over the small set of counts this kernel uses).
 template <int N>
 __device__ __forceinline__ void vmcnt_wait() {
-  static_assert(N == 0 || N == 2 || N == 4, "unsupported vmcnt");
+  static_assert(N >= 0 && N <= 4, "unsupported vmcnt");
   if constexpr (N == 0) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  } else if constexpr (N == 1) {
+    asm volatile("s_waitcnt vmcnt(1)" ::: "memory");
   } else if constexpr (N == 2) {
     asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  } else if constexpr (N == 3) {
+    asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
   } else if constexpr (N == 4) {
     asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   }
@@ -255,6 +259,207 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8_kernel(
   }
 }
 
+// s_waitcnt lgkmcnt(N) with a compile-time literal.
+template <int N>
+__device__ __forceinline__ void lgkm_wait() {
+  static_assert(N == 0 || N == 2 || N == 4, "unsupported lgkmcnt");
+  if constexpr (N == 0) {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  } else if constexpr (N == 2) {
+    asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");
+  } else if constexpr (N == 4) {
+    asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+  }
+}
+
+// Register-pipelined variant: fragment ds_reads are issued one phase AHEAD
+// of their MFMA (counted lgkmcnt instead of a full drain), one barrier per
+// phase instead of two, and every read targets the CURRENT K-tile:
+//   ph0: issue B(all) + A(m-groups 0,1); lgkm leaves group 1 in flight
+//   ph1: issue A(group 2);               lgkm leaves group 2
+//   ph2: issue A(group 3);               lgkm leaves group 3
+//   ph3: no reads; lgkm(0); counted vmcnt BEFORE the barrier so the whole
+//        next tile is landed when its ph0 reads issue.
+// Prefetch is unchanged (A(t+1) at ph0/1, B(t+2) at ph2/3).
+template <int BM, int BN, bool SETPRIO = false>
+__global__ __launch_bounds__(512, 2) void gemm_nt8p_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
+    int ntiles) {
+  constexpr int FM = BM / (2 * 16);
+  constexpr int FN = BN / (4 * 16);
+  constexpr int MG = FM / 4;  // m-reps per phase
+  static_assert(FM >= 4, "need >=1 m-rep per phase");
+  constexpr int HA = BM / 2;
+  constexpr int HB = BN / 2;
+  constexpr int LB = HB / 64;  // loads/lane per B half-tile
+  __shared__ bf16_t As[2][BM * 64];
+  __shared__ bf16_t Bs[2][BN * 64];
+
+  const int nwg = mtiles * ntiles;
+  const int orig = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
+                 (orig >> 3);
+  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  const int row0 = tile_m * BM, col0 = tile_n * BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wm = wid >> 2, wn = wid & 3;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+
+  stage_half<HA>(A, lda, row0, M, 0, &As[0][0], wid, lane);
+  stage_half<HA>(A, lda, row0 + HA, M, 0, &As[0][HA * 64], wid, lane);
+  stage_half<HB>(B, ldb, col0, N, 0, &Bs[0][0], wid, lane);
+  stage_half<HB>(B, ldb, col0 + HB, N, 0, &Bs[0][HB * 64], wid, lane);
+  if (NT > 1) {
+    stage_half<HB>(B, ldb, col0, N, 64, &Bs[1][0], wid, lane);
+    stage_half<HB>(B, ldb, col0 + HB, N, 64, &Bs[1][HB * 64], wid, lane);
+    vmcnt_wait<2 * LB>();
+  } else {
+    vmcnt_wait<0>();
+  }
+  S_BARRIER();
+
+  const int arow_base = wm * (FM * 16) + (lane & 15);
+  const int brow_base = wn * (FN * 16) + (lane & 15);
+  const int kblk = (lane >> 4);
+
+  s16x8 bfrag[FN][2];
+  s16x8 afrag[2][MG][2];  // [phase parity][m-rep in group][k-step]
+
+  for (int t = 0; t < NT; ++t) {
+    const int buf = t & 1;
+    const bf16_t* at = &As[buf][0];
+    const bf16_t* bt = &Bs[buf][0];
+    const bool pre_b0 = t + 2 < NT;
+
+    // ---- ph0: issue B(all) + A groups 0,1; compute group 0 ----
+#pragma unroll
+    for (int n = 0; n < FN; ++n)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        bfrag[n][kk] = lds_frag(bt, brow_base + n * 16, kk * 4 + kblk);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        afrag[0][g][kk] =
+            lds_frag(at, arow_base + g * 16, kk * 4 + kblk);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        afrag[1][g][kk] =
+            lds_frag(at, arow_base + (MG + g) * 16, kk * 4 + kblk);
+    if (t + 1 < NT)
+      stage_half<HA>(A, lda, row0, M, (t + 1) * 64, &As[buf ^ 1][0], wid,
+                     lane);
+    lgkm_wait<2 * MG>();  // group 1 still in flight
+    __builtin_amdgcn_sched_barrier(0);
+    if (SETPRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int n = 0; n < FN; ++n)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[0][g][kk], bfrag[n][kk], acc[g][n], 0, 0, 0);
+    if (SETPRIO) __builtin_amdgcn_s_setprio(0);
+    S_BARRIER();
+
+    // ---- ph1/ph2: issue group p+1; compute group p ----
+#pragma unroll
+    for (int p = 1; p < 3; ++p) {
+      const int cur = p & 1, nxt = cur ^ 1;
+#pragma unroll
+      for (int g = 0; g < MG; ++g)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          afrag[nxt][g][kk] = lds_frag(
+              at, arow_base + ((p + 1) * MG + g) * 16, kk * 4 + kblk);
+      if (p == 1) {
+        if (t + 1 < NT)
+          stage_half<HA>(A, lda, row0 + HA, M, (t + 1) * 64,
+                         &As[buf ^ 1][HA * 64], wid, lane);
+      } else {
+        if (pre_b0)
+          stage_half<HB>(B, ldb, col0, N, (t + 2) * 64, &Bs[buf][0], wid,
+                         lane);
+      }
+      lgkm_wait<2 * MG>();
+      __builtin_amdgcn_sched_barrier(0);
+      if (SETPRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int g = 0; g < MG; ++g)
+#pragma unroll
+        for (int n = 0; n < FN; ++n)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[p * MG + g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[cur][g][kk], bfrag[n][kk], acc[p * MG + g][n], 0, 0,
+                0);
+      if (SETPRIO) __builtin_amdgcn_s_setprio(0);
+      S_BARRIER();
+    }
+
+    // ---- ph3: no reads; land the whole next tile; compute group 3 ----
+    if (pre_b0) {
+      vmcnt_wait<LB>();  // leave only B0(t+2) in flight
+    } else {
+      vmcnt_wait<0>();
+    }
+    if (t + 2 < NT)
+      stage_half<HB>(B, ldb, col0 + HB, N, (t + 2) * 64, &Bs[buf][HB * 64],
+                     wid, lane);
+    lgkm_wait<0>();
+    __builtin_amdgcn_sched_barrier(0);
+    if (SETPRIO) __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int n = 0; n < FN; ++n)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[3 * MG + g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[1][g][kk], bfrag[n][kk], acc[3 * MG + g][n], 0, 0, 0);
+    if (SETPRIO) __builtin_amdgcn_s_setprio(0);
+    S_BARRIER();
+  }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr] + bv;
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        bf16_t* cp = &C[(int64_t)row * ldc + col];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 // Host entry: launches the 8-phase kernel for a given tile config.
@@ -303,6 +508,19 @@ void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     case 13: LAUNCH_8PH(128, 128, true, true); break;
     case 20: LAUNCH_8PH(256, 256, false, false); break;
     case 23: LAUNCH_8PH(128, 128, false, false); break;
+#define LAUNCH_8PHP(BM, BN, PRIO)                                             \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_nt8p_kernel<BM, BN, PRIO>), dim3(mt * nt),       \
+                       dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M,   \
+                       N, K, lda, ldb, ldc, (int)act, mt, nt);                \
+  } while (0)
+    case 30: LAUNCH_8PHP(256, 256, false); break;
+    case 31: LAUNCH_8PHP(256, 256, true); break;
+    case 32: LAUNCH_8PHP(256, 128, false); break;
+    case 33: LAUNCH_8PHP(128, 256, false); break;
+    case 34: LAUNCH_8PHP(128, 128, false); break;
+#undef LAUNCH_8PHP
     default: TORCH_CHECK(false, "gemm8: unknown variant");
   }
 #undef LAUNCH_8PH
