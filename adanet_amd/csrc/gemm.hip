@@ -29,6 +29,11 @@
 
 #define THREADS 256
 
+// 8-phase deep-pipelined path (gemm_8ph.hip); returns false -> fall back.
+bool gemm_nt8_try(const bf16_t* a, const bf16_t* b, bf16_t* c,
+                  const float* bias_ptr, int M, int N, int K, int64_t lda,
+                  int64_t ldb, int64_t ldc, int act, hipStream_t stream);
+
 typedef s16x8 frag_ab;
 
 // Stage a [ROWS x KSTEP] bf16 tile into LDS: each global_load_lds covers
@@ -295,6 +300,11 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL(gemm_nt_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
                        ldc, (int)act);
+    HIP_CHECK_KERNEL();
+    return;
+  }
+  if (fast && gemm_nt8_try(a, b, c, bias_ptr, M, N, K, lda, ldb, ldc,
+                           (int)act, stream.stream())) {
     HIP_CHECK_KERNEL();
     return;
   }

@@ -462,6 +462,35 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8p_kernel(
 
 }  // namespace
 
+// Production dispatch hook, called from gemm.hip's gemm_nt_bf16 fast path.
+// Returns false when the 8-phase kernels don't apply (caller falls back).
+// Thresholds from measured A/B (profiles/gemm8_r02c.json, MI355X):
+//   * 256^2 noprio wins at big grids (1036-1167 TF vs old 860-914 @4k/8k),
+//   * 128^2 8-phase wins at 2048-class grids (552 vs 473 TF @2048^3),
+//   * setprio measured NEGATIVE on this lockstep schedule (-17%): off.
+bool gemm_nt8_try(const bf16_t* a, const bf16_t* b, bf16_t* c,
+                  const float* bias_ptr, int M, int N, int K, int64_t lda,
+                  int64_t ldb, int64_t ldc, int act, hipStream_t stream) {
+  if (K % 64 != 0 || K < 128 || lda % 8 || ldb % 8) return false;
+  const int64_t t256 = (int64_t)((M + 255) / 256) * ((N + 255) / 256);
+  const int64_t t128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  if (t256 >= 200) {
+    const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+    hipLaunchKernelGGL((gemm_nt8_kernel<256, 256, false, false>),
+                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
+                       bias_ptr, M, N, K, lda, ldb, ldc, act, mt, nt);
+    return true;
+  }
+  if (t128 >= 256) {
+    const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+    hipLaunchKernelGGL((gemm_nt8_kernel<128, 128, false, false>),
+                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
+                       bias_ptr, M, N, K, lda, ldb, ldc, act, mt, nt);
+    return true;
+  }
+  return false;
+}
+
 // Host entry: launches the 8-phase kernel for a given tile config.
 // variant: 0 = 256x256, 1 = 256x128, 2 = 128x256, 3 = 128x128,
 //          +10 = SAFE (drain) twin, +20 = no-setprio twin.
