@@ -11,6 +11,9 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
 void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
               const c10::optional<at::Tensor>& bias, int64_t act,
               int64_t variant);
+void gemm_x8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+             const c10::optional<at::Tensor>& bias, int64_t act,
+             int64_t trans_a, int64_t trans_b, int64_t variant);
 void probe_tr16_layout(at::Tensor& out);
 void gemm_tr_probe(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                    int64_t trans_a, int64_t trans_b, int64_t variant);
@@ -94,6 +97,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_nt_bf16_probe", &gemm_nt_bf16_probe);
   m.def("gemm_nt8", &gemm_nt8,
         "8-phase deep-pipelined GEMM (256^2-class tiles)");
+  m.def("gemm_x8", &gemm_x8,
+        "pipelined GEMM with K-major (transposed) operand staging");
   m.def("probe_tr16_layout", &probe_tr16_layout);
   m.def("gemm_tr_probe", &gemm_tr_probe);
   m.def("gemm_tr_bf16", &gemm_tr_bf16);

@@ -462,6 +462,357 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8p_kernel(
 
 }  // namespace
 
+// ---------------------------------------------------------------------
+// Generalized pipelined kernel: same 4-phase schedule as gemm_nt8p_kernel,
+// additionally templated over the wave grid (2 x WN) and per-operand
+// K-major staging (TRA/TRB) using the ds_read_b64_tr_b16 hardware
+// transpose (semantics measured in gemm_tn.hip round 1). This brings the
+// counted-vmcnt pipeline to the backward dX (trans_b) / dW (tt) GEMMs,
+// which the round-2 bench profile shows at ~40% of step GPU time on the
+// drain-bound 2-phase structure (profiles/prof_r02b).
+// ---------------------------------------------------------------------
+
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_g;
+
+__device__ __forceinline__ u32x2_g tr8_b16_read(const bf16_t* p) {
+  const unsigned off = (unsigned)(uintptr_t)(
+      const __attribute__((address_space(3))) bf16_t*)p;
+  u32x2_g out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(out) : "v"(off));
+  return out;
+}
+
+namespace {
+
+// Linear stage of one half-tile generalized over wave count.
+template <int HROWS, int NW>
+__device__ __forceinline__ void stage_half_g(
+    const bf16_t* __restrict__ G, int64_t ld, int rows0, int max_row, int k0,
+    bf16_t* __restrict__ lds_half, int wid, int lane) {
+#pragma unroll
+  for (int r = 0; r < HROWS / (NW * 8); ++r) {
+    const int row_in_half = r * (NW * 8) + wid * 8 + (lane >> 3);
+    int grow = rows0 + row_in_half;
+    grow = grow < max_row ? grow : max_row - 1;
+    const int src_blk = (lane & 7) ^ ((lane >> 3) & 7);
+    const bf16_t* gp = G + (int64_t)grow * ld + k0 + src_blk * 8;
+    bf16_t* lp = lds_half + (r * (NW * 8) + wid * 8) * 64;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+// K-major stage of one half-tile (HCOLS output-cols x 64 k) from G[k][col]:
+// layout [2 kslab][HCOLS/16][32][16], one 1 KiB DMA per [32][16] subtile.
+template <int HCOLS, int NW>
+__device__ __forceinline__ void stage_half_tr8(
+    const bf16_t* __restrict__ G, int64_t ld, int col0, int max_col, int k0,
+    int max_k, bf16_t* __restrict__ lds_half, int wid, int lane) {
+  constexpr int SUBT = HCOLS / 16;
+#pragma unroll
+  for (int slab = 0; slab < 2; ++slab) {
+#pragma unroll
+    for (int st = wid; st < SUBT; st += NW) {
+      const int k = k0 + slab * 32 + (lane >> 1);
+      int col = col0 + st * 16 + (lane & 1) * 8;
+      if (col + 8 > max_col) col = max(0, (max_col - 8) & ~7);
+      const bf16_t* gp = G + (int64_t)min(k, max_k - 1) * ld + col;
+      bf16_t* lp = lds_half + slab * HCOLS * 32 + st * 512;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)gp,
+          (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+    }
+  }
+}
+
+// Fragment from a K-major-staged half: two hardware-transpose reads of the
+// [32][16] subtile (measured cooperative-transpose semantics, gemm_tn.hip).
+template <int HCOLS>
+__device__ __forceinline__ s16x8 frag_tr8(const bf16_t* lds_half,
+                                          int colblock, int kk, int lane) {
+  const bf16_t* sub = lds_half + kk * HCOLS * 32 + colblock * 512;
+  const int row = 8 * (lane >> 4) + ((lane >> 2) & 3);
+  const int col4 = 4 * (lane & 3);
+  u32x2_g r0 = tr8_b16_read(sub + row * 16 + col4);
+  u32x2_g r1 = tr8_b16_read(sub + (row + 4) * 16 + col4);
+  union {
+    s16x8 f;
+    struct { u32x2_g lo, hi; } u;
+  } pack;
+  pack.u.lo = r0;
+  pack.u.hi = r1;
+  return pack.f;
+}
+
+template <int N>
+__device__ __forceinline__ void lgkm_wait_g() {
+  static_assert(N == 0 || N == 2 || N == 4 || N == 8, "lgkm count");
+  if constexpr (N == 0) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt lgkmcnt(2)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt lgkmcnt(4)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");
+}
+
+// Pipelined 4-phase kernel, generalized: 2 x WN wave grid, per-operand
+// K-major staging. Schedule identical to gemm_nt8p_kernel (see above).
+template <int BM, int BN, int WN, bool TRA, bool TRB>
+__global__ __launch_bounds__(2 * WN * 64, 2) void gemm_x8_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
+    int ntiles) {
+  constexpr int NW = 2 * WN;
+  constexpr int FM = BM / (2 * 16);
+  constexpr int FN = BN / (WN * 16);
+  constexpr int MG = FM / 4;
+  static_assert(FM >= 4, "need >=1 m-rep per phase");
+  constexpr int HA = BM / 2;
+  constexpr int HB = BN / 2;
+  constexpr int LB = HB / (8 * NW) > 0 ? HB / (8 * NW) : 1;  // loads/lane/half
+  __shared__ bf16_t As[2][BM * 64];
+  __shared__ bf16_t Bs[2][BN * 64];
+
+  const int nwg = mtiles * ntiles;
+  const int orig = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
+                 (orig >> 3);
+  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  const int row0 = tile_m * BM, col0 = tile_n * BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wm = wid / WN, wn = wid % WN;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+
+#define STAGE_A8(buf, half, k0)                                               \
+  do {                                                                        \
+    if (TRA)                                                                  \
+      stage_half_tr8<HA, NW>(A, lda, row0 + (half)*HA, M, k0, K,              \
+                             &As[buf][(half)*HA * 64], wid, lane);            \
+    else                                                                      \
+      stage_half_g<HA, NW>(A, lda, row0 + (half)*HA, M, k0,                   \
+                           &As[buf][(half)*HA * 64], wid, lane);              \
+  } while (0)
+#define STAGE_B8(buf, half, k0)                                               \
+  do {                                                                        \
+    if (TRB)                                                                  \
+      stage_half_tr8<HB, NW>(B, ldb, col0 + (half)*HB, N, k0, K,              \
+                             &Bs[buf][(half)*HB * 64], wid, lane);            \
+    else                                                                      \
+      stage_half_g<HB, NW>(B, ldb, col0 + (half)*HB, N, k0,                   \
+                           &Bs[buf][(half)*HB * 64], wid, lane);              \
+  } while (0)
+
+  STAGE_A8(0, 0, 0);
+  STAGE_A8(0, 1, 0);
+  STAGE_B8(0, 0, 0);
+  STAGE_B8(0, 1, 0);
+  if (NT > 1) {
+    STAGE_B8(1, 0, 64);
+    STAGE_B8(1, 1, 64);
+    vmcnt_wait<2 * LB>();
+  } else {
+    vmcnt_wait<0>();
+  }
+  S_BARRIER();
+
+  const int arow_base = wm * (FM * 16) + (lane & 15);
+  const int brow_base = wn * (FN * 16) + (lane & 15);
+  const int kblk = (lane >> 4);
+
+  // Fragment loaders: operand half selected by the wave's fixed span.
+  // A: wave wm's rows live in half wm (WARPS_M == 2). B: half = cols
+  // span / HB. For K-major operands the fragment col-block indexes into
+  // the wave's half-local subtile array.
+#define LOAD_A8(dst, at_, m)                                                  \
+  do {                                                                        \
+    if (TRA) {                                                                \
+      const int cb = (wm * FM + (m)) % (HA / 16);                             \
+      const bf16_t* hbase = (at_) + ((wm * FM + (m)) / (HA / 16)) * HA * 64;  \
+      (dst)[0] = frag_tr8<HA>(hbase, cb, 0, lane);                            \
+      (dst)[1] = frag_tr8<HA>(hbase, cb, 1, lane);                            \
+    } else {                                                                  \
+      (dst)[0] = lds_frag(at_, arow_base + (m)*16, kblk);                     \
+      (dst)[1] = lds_frag(at_, arow_base + (m)*16, 4 + kblk);                 \
+    }                                                                         \
+  } while (0)
+#define LOAD_B8(dst, bt_, n)                                                  \
+  do {                                                                        \
+    if (TRB) {                                                                \
+      const int cb = (wn * FN + (n)) % (HB / 16);                             \
+      const bf16_t* hbase = (bt_) + ((wn * FN + (n)) / (HB / 16)) * HB * 64;  \
+      (dst)[0] = frag_tr8<HB>(hbase, cb, 0, lane);                            \
+      (dst)[1] = frag_tr8<HB>(hbase, cb, 1, lane);                            \
+    } else {                                                                  \
+      (dst)[0] = lds_frag(bt_, brow_base + (n)*16, kblk);                     \
+      (dst)[1] = lds_frag(bt_, brow_base + (n)*16, 4 + kblk);                 \
+    }                                                                         \
+  } while (0)
+
+  s16x8 bfrag[FN][2];
+  s16x8 afrag[2][MG][2];
+  constexpr int AG_READS = MG * 2 * (TRA ? 2 : 1);  // lgkm units per A group
+
+  for (int t = 0; t < NT; ++t) {
+    const int buf = t & 1;
+    const bf16_t* at = &As[buf][0];
+    const bf16_t* bt = &Bs[buf][0];
+    const bool pre_b0 = t + 2 < NT;
+
+    // ph0
+#pragma unroll
+    for (int n = 0; n < FN; ++n) LOAD_B8(bfrag[n], bt, n);
+#pragma unroll
+    for (int g = 0; g < MG; ++g) LOAD_A8(afrag[0][g], at, g);
+#pragma unroll
+    for (int g = 0; g < MG; ++g) LOAD_A8(afrag[1][g], at, MG + g);
+    if (t + 1 < NT) STAGE_A8(buf ^ 1, 0, (t + 1) * 64);
+    lgkm_wait_g<AG_READS>();
+    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int n = 0; n < FN; ++n)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[0][g][kk], bfrag[n][kk], acc[g][n], 0, 0, 0);
+    S_BARRIER();
+
+    // ph1, ph2
+#pragma unroll
+    for (int p = 1; p < 3; ++p) {
+      const int cur = p & 1, nxt = cur ^ 1;
+#pragma unroll
+      for (int g = 0; g < MG; ++g)
+        LOAD_A8(afrag[nxt][g], at, (p + 1) * MG + g);
+      if (p == 1) {
+        if (t + 1 < NT) STAGE_A8(buf ^ 1, 1, (t + 1) * 64);
+      } else {
+        if (pre_b0) STAGE_B8(buf, 0, (t + 2) * 64);
+      }
+      lgkm_wait_g<AG_READS>();
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int g = 0; g < MG; ++g)
+#pragma unroll
+        for (int n = 0; n < FN; ++n)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[p * MG + g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[cur][g][kk], bfrag[n][kk], acc[p * MG + g][n], 0, 0,
+                0);
+      S_BARRIER();
+    }
+
+    // ph3
+    if (pre_b0) {
+      vmcnt_wait<LB>();
+    } else {
+      vmcnt_wait<0>();
+    }
+    if (t + 2 < NT) STAGE_B8(buf, 1, (t + 2) * 64);
+    lgkm_wait_g<0>();
+    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+    for (int g = 0; g < MG; ++g)
+#pragma unroll
+      for (int n = 0; n < FN; ++n)
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk)
+          acc[3 * MG + g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[1][g][kk], bfrag[n][kk], acc[3 * MG + g][n], 0, 0, 0);
+    S_BARRIER();
+  }
+#undef STAGE_A8
+#undef STAGE_B8
+#undef LOAD_A8
+#undef LOAD_B8
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr] + bv;
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        bf16_t* cp = &C[(int64_t)row * ldc + col];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// Probe/production entry for the generalized pipelined kernel.
+// A is [M,K] ([K,M] when trans_a); B is [N,K] ([K,N] when trans_b).
+// variant: 0 = 128^2 8-wave, 1 = 128^2 4-wave, 2 = 256^2 8-wave,
+//          3 = 256x128 8-wave, 4 = 128x256 8-wave.
+void gemm_x8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+             const c10::optional<at::Tensor>& bias, int64_t act,
+             int64_t trans_a, int64_t trans_b, int64_t variant) {
+  const int M = (int)C.size(0), N = (int)C.size(1);
+  const int K = (int)(trans_a ? A.size(0) : A.size(1));
+  TORCH_CHECK((int)(trans_b ? B.size(1) : B.size(0)) == N &&
+              (int)(trans_b ? B.size(0) : B.size(1)) == K &&
+              (int)(trans_a ? A.size(1) : A.size(0)) == M,
+              "gemm_x8: shape mismatch");
+  TORCH_CHECK(K % 64 == 0, "gemm_x8: K % 64");
+  const int64_t lda = A.stride(0), ldb = B.stride(0), ldc = C.stride(0);
+  TORCH_CHECK(lda % 8 == 0 && ldb % 8 == 0, "gemm_x8: 16B-aligned rows");
+  const float* bias_ptr = nullptr;
+  if (bias.has_value() && bias->defined()) bias_ptr = bias->data_ptr<float>();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* a = (const bf16_t*)A.data_ptr();
+  const bf16_t* b = (const bf16_t*)B.data_ptr();
+  bf16_t* c = (bf16_t*)C.data_ptr();
+
+#define LX8(BM, BN, WNW, TA, TB)                                              \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_x8_kernel<BM, BN, WNW, TA, TB>), dim3(mt * nt),  \
+                       dim3(2 * WNW * 64), 0, stream.stream(), a, b, c,       \
+                       bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);   \
+  } while (0)
+#define LX8_T(BM, BN, WNW)                                                    \
+  do {                                                                        \
+    if (trans_a && trans_b) LX8(BM, BN, WNW, true, true);                     \
+    else if (trans_b) LX8(BM, BN, WNW, false, true);                          \
+    else if (trans_a) LX8(BM, BN, WNW, true, false);                          \
+    else LX8(BM, BN, WNW, false, false);                                      \
+  } while (0)
+  switch (variant) {
+    case 0: LX8_T(128, 128, 4); break;
+    case 1: LX8_T(128, 128, 2); break;
+    case 2: LX8_T(256, 256, 4); break;
+    case 3: LX8_T(256, 128, 4); break;
+    case 4: LX8_T(128, 256, 4); break;
+    default: TORCH_CHECK(false, "gemm_x8: unknown variant");
+  }
+#undef LX8
+#undef LX8_T
+  HIP_CHECK_KERNEL();
+}
+
 // Production dispatch hook, called from gemm.hip's gemm_nt_bf16 fast path.
 // Returns false when the 8-phase kernels don't apply (caller falls back).
 // Thresholds from measured A/B (profiles/gemm8_r02c.json, MI355X):
