@@ -75,7 +75,7 @@ def main():
             for (ta, tb) in ((True, True), (False, True), (True, False),
                              (False, False)):
                 for (M, N, K) in ((256, 256, 128), (512, 384, 192),
-                                  (300, 200, 128)):
+                                  (304, 200, 128)):
                     ok, rel = refcheck(M, N, K, ta, tb, variant)
                     results["refcheck"].append(
                         {"MNK": [M, N, K], "ta": ta, "tb": tb, "v": variant,
