@@ -24,7 +24,6 @@ def make(M, N, K, ta, tb, seed):
     Bf = torch.randn(N, K, device=dev) / 8
     A = (Af.t().contiguous() if ta else Af).to(torch.bfloat16)
     B = (Bf.t().contiguous() if tb else Bf).to(torch.bfloat16)
-    ref = A.float().t() @ B.float() if ta else None
     return A, B, Af, Bf
 
 
