@@ -26,6 +26,13 @@
 typedef s16x8 frag_ab;
 typedef __attribute__((ext_vector_type(2))) unsigned int u32x2;
 
+// Pipelined counted-vmcnt path (gemm_8ph.hip) — measured winner for the
+// trans_b-only (dX) mid shapes: +9-20% over the 2-phase structure at
+// 2048-class grids (profiles/gemm_x8_r02a.json).
+void gemm_x8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
+             const c10::optional<at::Tensor>& bias, int64_t act,
+             int64_t trans_a, int64_t trans_b, int64_t variant);
+
 typedef __attribute__((address_space(3))) const bf16_t* lds_cptr_t;
 
 __device__ __forceinline__ u32x2 tr_b16_read(const bf16_t* lds_half_addr) {
@@ -374,6 +381,11 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   const int64_t b128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
   const int64_t t_256x128 = (int64_t)((M + 255) / 256) * ((N + 127) / 128);
   const int64_t t_128x256 = (int64_t)((M + 127) / 128) * ((N + 255) / 256);
+  if (!trans_a && trans_b && K % 64 == 0 && K <= 2048 && b128 >= 256 &&
+      b128 < 1024) {
+    gemm_x8(A, B, C, bias, act, trans_a, trans_b, /*128^2 8-wave*/ 0);
+    return;
+  }
 #define LTR(BM, BN, FM, FN, MW, WGM, WGN)                                     \
   do {                                                                        \
     const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
