@@ -63,6 +63,13 @@ class _ScopedSummary(Summary):
         self._step = 0
         self._lock = threading.Lock()
         self._file = None
+        self._tb = None
+        # TensorBoard event files alongside the JSONL (the reference's UI
+        # is TensorBoard — docs/source/tensorboard.md); per-scope dirs so
+        # same-named tags overlay across candidates. ADANET_TB_EVENTS=0
+        # turns the tfevents stream off.
+        self._tb_enabled = os.environ.get("ADANET_TB_EVENTS", "1") not in (
+            "0", "false")
         if not self._skip:
             parts = [logdir]
             if namespace:
@@ -72,6 +79,12 @@ class _ScopedSummary(Summary):
             self._dir = os.path.join(*parts)
             os.makedirs(self._dir, exist_ok=True)
             self._path = os.path.join(self._dir, "events.jsonl")
+
+    def _tb_writer(self):
+        if self._tb is None and self._tb_enabled and not self._skip:
+            from adanet_amd.core.tb_writer import TBEventWriter
+            self._tb = TBEventWriter(self._dir)
+        return self._tb if self._tb_enabled else None
 
     @property
     def scope(self):
@@ -95,7 +108,13 @@ class _ScopedSummary(Summary):
                 f.write(json.dumps(rec) + "\n")
 
     def scalar(self, name, tensor, family=None):
-        self._write("scalar", name, _to_scalar(tensor), family)
+        v = _to_scalar(tensor)
+        self._write("scalar", name, v, family)
+        tb = None if self._skip else self._tb_writer()
+        if tb is not None:
+            tag = "%s/%s" % (family, name) if family else name
+            with self._lock:
+                tb.scalar(tag, v, self._step)
 
     def histogram(self, name, tensor, family=None):
         if self._skip:
@@ -110,6 +129,12 @@ class _ScopedSummary(Summary):
             "count": int(t.numel()),
         }
         self._write("histogram", name, payload, family)
+        tb = self._tb_writer()
+        if tb is not None:
+            tag = "%s/%s" % (family, name) if family else name
+            sample = t[:16384].cpu().tolist()
+            with self._lock:
+                tb.histogram(tag, sample, self._step)
 
     def image(self, name, tensor, family=None):
         if self._skip:
@@ -124,6 +149,11 @@ class _ScopedSummary(Summary):
 
     def text(self, name, value, family=None):
         self._write("text", name, str(value), family)
+        tb = None if self._skip else self._tb_writer()
+        if tb is not None:
+            tag = "%s/%s" % (family, name) if family else name
+            with self._lock:
+                tb.text(tag, str(value), self._step)
 
 
 def read_events(path_or_dir: str):

@@ -228,3 +228,60 @@ def test_colocation_groups_union_by_candidate():
     groups = _colocation_groups(b, [_C([b[0], b[1]]), _C([b[1], b[2]])])
     assert groups["a"] == groups["b"] == groups["c"]
     assert groups["d"] != groups["a"]
+
+
+def test_tb_writer_crc32c_known_vector():
+    from adanet_amd.core.tb_writer import crc32c
+    # RFC 3720 Castagnoli test vector
+    assert crc32c(b"123456789") == 0xE3069283
+    assert crc32c(b"") == 0
+
+
+def test_tb_writer_roundtrip(tmp_path):
+    """tfevents framing + protobuf encode must parse back CRC-clean: first
+    record is the brain.Event:2 version stamp, then scalars/histograms/text
+    with exact tags, steps and values (reference _ScopedSummaryV2 writes
+    real event files, adanet/core/summary.py:375-637)."""
+    from adanet_amd.core.tb_writer import TBEventWriter, read_tfevents
+    w = TBEventWriter(str(tmp_path))
+    w.scalar("loss", 0.5, step=3)
+    w.scalar("eval/adanet_loss", -1.25, step=10)
+    w.histogram("weights", [0.0, 1.0, 2.0, 2.0], step=4)
+    w.text("architecture/adanet/ensembles", '{"x":1}', step=7)
+    events = read_tfevents(w.path)
+    assert events[0]["file_version"] == "brain.Event:2"
+    assert events[1]["values"][0]["tag"] == "loss"
+    assert events[1]["values"][0]["simple_value"] == pytest.approx(0.5)
+    assert events[1]["step"] == 3
+    assert events[2]["values"][0]["tag"] == "eval/adanet_loss"
+    assert events[2]["values"][0]["simple_value"] == pytest.approx(-1.25)
+    h = events[3]["values"][0]["histo"]
+    assert h["min"] == 0.0 and h["max"] == 2.0 and h["num"] == 4
+    assert h["sum"] == 5.0 and h["sum_squares"] == 9.0
+    assert events[4]["values"][0]["text"] == '{"x":1}'
+
+
+def test_scoped_summary_writes_tfevents_per_scope(tmp_path):
+    """Each candidate scope gets its OWN event dir with the SAME tag names
+    so TensorBoard overlays the charts (reference summary.py:262-296)."""
+    from adanet_amd.core.tb_writer import read_tfevents
+    import glob as _glob
+    scopes = {}
+    for scope in ("cand_a", "cand_b"):
+        s = _ScopedSummary(str(tmp_path), scope=scope, namespace="t0_ens")
+        s.set_step(5)
+        s.scalar("adanet_loss", 0.25 if scope == "cand_a" else 0.75)
+        scopes[scope] = s
+    for scope, want in (("cand_a", 0.25), ("cand_b", 0.75)):
+        files = _glob.glob(
+            os.path.join(str(tmp_path), "t0_ens", scope,
+                         "events.out.tfevents.*"))
+        assert len(files) == 1, files
+        events = read_tfevents(files[0])
+        vals = [v for e in events for v in e["values"]]
+        assert vals[0]["tag"] == "adanet_loss"
+        assert vals[0]["simple_value"] == pytest.approx(want)
+        assert events[1]["step"] == 5
+    # the JSONL stream stays (both backends active)
+    assert read_events(
+        os.path.join(str(tmp_path), "t0_ens", "cand_a"))
