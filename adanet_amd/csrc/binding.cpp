@@ -90,6 +90,23 @@ void depthwise_bwd_dw(const at::Tensor& x, const at::Tensor& dy,
                       at::Tensor& dw, int64_t stride, int64_t pad);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred, at::Tensor& correct);
+void pool3_fwd(const at::Tensor& x, at::Tensor& y,
+               const c10::optional<at::Tensor>& argmax, int64_t stride,
+               int64_t is_max);
+void pool3_bwd(const at::Tensor& dy, const c10::optional<at::Tensor>& argmax,
+               at::Tensor& dx, int64_t stride, int64_t is_max);
+void batchnorm_stats(const at::Tensor& x, at::Tensor& mean, at::Tensor& rstd,
+                     const c10::optional<at::Tensor>& running_mean,
+                     const c10::optional<at::Tensor>& running_var,
+                     double eps, double momentum);
+void batchnorm_norm(const at::Tensor& x, at::Tensor& y,
+                    const at::Tensor& mean, const at::Tensor& rstd,
+                    const c10::optional<at::Tensor>& gamma,
+                    const c10::optional<at::Tensor>& beta);
+void batchnorm_bwd(const at::Tensor& x, const at::Tensor& dy, at::Tensor& dx,
+                   const at::Tensor& mean, const at::Tensor& rstd,
+                   const c10::optional<at::Tensor>& gamma, at::Tensor& sdy,
+                   at::Tensor& sdyx);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "adanet_amd gfx950 (MI355X/CDNA4) kernels";
@@ -126,6 +143,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);
   m.def("binary_histogram", &binary_histogram);
+  m.def("pool3_fwd", &pool3_fwd);
+  m.def("pool3_bwd", &pool3_bwd);
+  m.def("batchnorm_stats", &batchnorm_stats);
+  m.def("batchnorm_norm", &batchnorm_norm);
+  m.def("batchnorm_bwd", &batchnorm_bwd);
   m.def("depthwise_fwd", &depthwise_fwd);
   m.def("depthwise_bwd_dx", &depthwise_bwd_dx);
   m.def("depthwise_bwd_dw", &depthwise_bwd_dw);

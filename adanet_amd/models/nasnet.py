@@ -23,14 +23,17 @@ from adanet_amd.ops.linear import HipLinear
 
 
 class BNfp32(nn.Module):
-    """BatchNorm with fp32 statistics/affine under bf16 activations."""
+    """BatchNorm with fp32 statistics/affine under bf16 activations —
+    hand-written CDNA4 kernels on GPU (csrc/batchnorm.hip, K8), fp32
+    F.batch_norm on CPU (inside HipBatchNorm2d)."""
 
     def __init__(self, c, momentum=0.1, eps=1e-3):
         super().__init__()
-        self.bn = nn.BatchNorm2d(c, momentum=momentum, eps=eps)
+        from adanet_amd.ops.batchnorm import HipBatchNorm2d
+        self.bn = HipBatchNorm2d(c, momentum=momentum, eps=eps)
 
     def forward(self, x):
-        return self.bn(x.float()).to(x.dtype)
+        return self.bn(x)
 
 
 def drop_path(x, keep_prob: float, training: bool):
@@ -114,11 +117,9 @@ class _Pool(nn.Module):
 
     def __init__(self, kind, stride):
         super().__init__()
-        if kind == "avg":
-            self.pool = nn.AvgPool2d(3, stride=stride, padding=1,
-                                     count_include_pad=False)
-        else:
-            self.pool = nn.MaxPool2d(3, stride=stride, padding=1)
+        from adanet_amd.ops.conv import HipPool2d
+        # Native 3x3 pooling (csrc/pool.hip) — no MIOpen on the cell path.
+        self.pool = HipPool2d(kind, stride)
 
     def forward(self, x):
         return self.pool(x)

@@ -20,6 +20,7 @@ from __future__ import annotations
 from typing import Optional
 
 import torch
+import torch.nn.functional as F
 from torch import nn
 
 from adanet_amd.ops import _extension
@@ -256,3 +257,61 @@ class HipConvNxN(nn.Module):
         return "in=%d, out=%d, k=%d, s=%d (im2col+MFMA when Co%%32==0)" % (
             self.in_channels, self.out_channels, self.kernel_size,
             self.stride)
+
+
+class _Pool3Fn(torch.autograd.Function):
+    """3x3 pooling on the native kernel (csrc/pool.hip): avg is
+    count_include_pad=False, max saves the window argmax for an exact
+    deterministic gather backward."""
+
+    @staticmethod
+    def forward(ctx, x, stride, is_max):
+        ext = _extension.require()
+        x = x.contiguous()
+        N, C, H, W = x.shape
+        OH = (H + 2 - 3) // stride + 1
+        OW = (W + 2 - 3) // stride + 1
+        y = torch.empty((N, C, OH, OW), device=x.device, dtype=x.dtype)
+        argmax = None
+        if is_max:
+            argmax = torch.empty((N, C, OH, OW), device=x.device,
+                                 dtype=torch.uint8)
+        ext.pool3_fwd(x, y, argmax, stride, int(is_max))
+        ctx.save_for_backward(argmax) if argmax is not None else \
+            ctx.save_for_backward()
+        ctx.meta = (x.shape, stride, is_max)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _extension.require()
+        shape, stride, is_max = ctx.meta
+        argmax = ctx.saved_tensors[0] if ctx.saved_tensors else None
+        dy = dy.contiguous()
+        dx = torch.empty(shape, device=dy.device, dtype=dy.dtype)
+        ext.pool3_bwd(dy, argmax, dx, stride, int(is_max))
+        return dx, None, None
+
+
+class HipPool2d(nn.Module):
+    """3x3 avg/max pool, stride 1 or 2, pad 1 — native CDNA4 kernels on
+    GPU (the last MIOpen ops on the NASNet cell hot path, K9 remainder);
+    torch pooling on CPU."""
+
+    def __init__(self, kind: str, stride: int):
+        super().__init__()
+        assert kind in ("avg", "max")
+        self.kind = kind
+        self.stride = stride
+
+    def forward(self, x):
+        if x.is_cuda:
+            return _Pool3Fn.apply(x.to(torch.bfloat16), self.stride,
+                                  self.kind == "max")
+        if self.kind == "avg":
+            return F.avg_pool2d(x, 3, self.stride, 1,
+                                count_include_pad=False)
+        return F.max_pool2d(x, 3, self.stride, 1)
+
+    def extra_repr(self):
+        return "%s3x3 s%d" % (self.kind, self.stride)

@@ -23,6 +23,8 @@ SRC = [
     "adanet_amd/csrc/mixer.hip",
     "adanet_amd/csrc/optim.hip",
     "adanet_amd/csrc/layernorm.hip",
+    "adanet_amd/csrc/batchnorm.hip",
+    "adanet_amd/csrc/pool.hip",
     "adanet_amd/csrc/elementwise.hip",
     "adanet_amd/csrc/reduce.hip",
     "adanet_amd/csrc/depthwise.hip",

@@ -652,3 +652,89 @@ def test_conv_nxn_unfold_gemm_vs_fp32():
                                       None, 2, 0)
     assert (y2.float() - ref2).abs().mean() / (
         ref2.abs().mean() + 1e-3) < 0.02
+
+
+# ------------------------------------------------------------ batchnorm K8
+def test_batchnorm_train_fwd_bwd_vs_fp32():
+    """Native BN (csrc/batchnorm.hip) vs torch fp32 BatchNorm2d: forward,
+    running stats, and dx/dgamma/dbeta in train mode."""
+    from adanet_amd.ops.batchnorm import HipBatchNorm2d
+    torch.manual_seed(11)
+    N, C, H, W = 8, 32, 16, 16
+    x = torch.randn(N, C, H, W, device=DEV)
+    bn = HipBatchNorm2d(C, momentum=0.1, eps=1e-3).to(DEV)
+    ref = torch.nn.BatchNorm2d(C, momentum=0.1, eps=1e-3).to(DEV)
+    with torch.no_grad():
+        ref.weight.copy_(torch.rand(C) + 0.5)
+        ref.bias.copy_(torch.randn(C) * 0.1)
+        bn.weight.copy_(ref.weight)
+        bn.bias.copy_(ref.bias)
+    xb = x.to(torch.bfloat16).requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    y = bn(xb)
+    yr = ref(xf)
+    assert (y.float() - yr).abs().max().item() < 0.05
+    assert torch.allclose(bn.running_mean, ref.running_mean, atol=1e-2)
+    assert torch.allclose(bn.running_var, ref.running_var, atol=1e-2)
+    g = torch.randn_like(yr)
+    y.backward(g.to(torch.bfloat16))
+    yr.backward(g)
+    assert (xb.grad.float() - xf.grad).abs().max().item() < 0.05
+    assert torch.allclose(bn.weight.grad, ref.weight.grad, atol=0.3,
+                          rtol=0.05)
+    assert torch.allclose(bn.bias.grad, ref.bias.grad, atol=0.3, rtol=0.05)
+
+
+def test_batchnorm_eval_mode_uses_running_stats():
+    from adanet_amd.ops.batchnorm import HipBatchNorm2d
+    torch.manual_seed(12)
+    C = 16
+    bn = HipBatchNorm2d(C, eps=1e-3).to(DEV)
+    with torch.no_grad():
+        bn.running_mean.copy_(torch.randn(C) * 0.3)
+        bn.running_var.copy_(torch.rand(C) + 0.5)
+        bn.weight.copy_(torch.rand(C) + 0.5)
+        bn.bias.copy_(torch.randn(C) * 0.1)
+    bn.eval()
+    x = torch.randn(4, C, 8, 8, device=DEV)
+    y = bn(x.to(torch.bfloat16))
+    ref = torch.nn.functional.batch_norm(
+        x, bn.running_mean, bn.running_var, bn.weight, bn.bias, False,
+        0.1, 1e-3)
+    assert (y.float() - ref).abs().max().item() < 0.05
+
+
+def test_batchnorm_odd_hw_scalar_path():
+    """HW % 8 != 0 exercises the scalar stats loop."""
+    from adanet_amd.ops.batchnorm import HipBatchNorm2d
+    torch.manual_seed(13)
+    x = torch.randn(4, 8, 5, 3, device=DEV)
+    bn = HipBatchNorm2d(8, eps=1e-3).to(DEV)
+    y = bn(x.to(torch.bfloat16))
+    ref = torch.nn.functional.batch_norm(
+        x, None, None, bn.weight, bn.bias, True, 0.1, 1e-3)
+    assert (y.float() - ref).abs().max().item() < 0.05
+
+
+# --------------------------------------------------------------- pool3 K9
+@pytest.mark.parametrize("kind,stride", [("avg", 1), ("avg", 2),
+                                         ("max", 1), ("max", 2)])
+def test_pool3_fwd_bwd_vs_fp32(kind, stride):
+    from adanet_amd.ops.conv import HipPool2d
+    torch.manual_seed(21)
+    x = torch.randn(4, 12, 15, 15, device=DEV)   # odd HW: edge windows
+    xb = x.to(torch.bfloat16).requires_grad_(True)
+    xf = x.clone().requires_grad_(True)
+    pool = HipPool2d(kind, stride)
+    y = pool(xb)
+    if kind == "avg":
+        yr = torch.nn.functional.avg_pool2d(xf, 3, stride, 1,
+                                            count_include_pad=False)
+    else:
+        yr = torch.nn.functional.max_pool2d(xf, 3, stride, 1)
+    assert y.shape == yr.shape
+    assert (y.float() - yr).abs().max().item() < 0.03
+    g = torch.randn_like(yr)
+    y.backward(g.to(torch.bfloat16))
+    yr.backward(g)
+    assert (xb.grad.float() - xf.grad).abs().max().item() < 0.03
