@@ -86,6 +86,14 @@ class HipBatchNorm2d(nn.Module):
             "num_batches_tracked", torch.zeros((), dtype=torch.long))
 
     def forward(self, x):
+        # dtype pinning: stats/affine are fp32 by design; a blanket
+        # module.to(bf16) without restore_fp32_params must not break BN.
+        if self.running_mean.dtype != torch.float32:
+            self.running_mean = self.running_mean.float()
+            self.running_var = self.running_var.float()
+        if self.weight is not None and self.weight.dtype != torch.float32:
+            self.weight.data = self.weight.data.float()
+            self.bias.data = self.bias.data.float()
         if x.is_cuda:
             if self.training:
                 self.num_batches_tracked += 1

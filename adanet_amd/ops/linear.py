@@ -91,7 +91,9 @@ def restore_fp32_params(module) -> None:
         if isinstance(m, HipLayerNorm) and m.weight is not None:
             m.weight.data = m.weight.data.float()
             m.bias.data = m.bias.data.float()
-        if isinstance(m, torch.nn.modules.batchnorm._BatchNorm):
+        from adanet_amd.ops.batchnorm import HipBatchNorm2d
+        if isinstance(m, (torch.nn.modules.batchnorm._BatchNorm,
+                          HipBatchNorm2d)):
             # BN statistics/affine stay fp32 under bf16 activations
             # (models/nasnet.py BNfp32 computes in fp32).
             if m.weight is not None:
