@@ -724,7 +724,9 @@ def test_pool3_fwd_bwd_vs_fp32(kind, stride):
     torch.manual_seed(21)
     x = torch.randn(4, 12, 15, 15, device=DEV)   # odd HW: edge windows
     xb = x.to(torch.bfloat16).requires_grad_(True)
-    xf = x.clone().requires_grad_(True)
+    # reference on the bf16-ROUNDED values: max-pool tie-breaks must see
+    # the same inputs or the argmax (and thus the routed gradient) differs
+    xf = x.to(torch.bfloat16).float().requires_grad_(True)
     pool = HipPool2d(kind, stride)
     y = pool(xb)
     if kind == "avg":
