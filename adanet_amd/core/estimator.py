@@ -1097,9 +1097,10 @@ class Estimator(object):
                 else:
                     sd = None
                 if comm.is_initialized():
-                    sd = comm.broadcast_object(sd, src=spec.owner_rank if not
-                                               self._placement.data_parallel
-                                               else 0)
+                    # flat tensor broadcast from the owner (not pickles).
+                    sd = comm.broadcast_named_tensors(
+                        sd, src=spec.owner_rank if not
+                        self._placement.data_parallel else 0)
                 self._frozen_states[self._frozen_key(t, name)] = sd
             # frozen members' states already stored from earlier iterations
         ens_sd = None
@@ -1110,7 +1111,7 @@ class Estimator(object):
             ens_sd = {k: v.detach().cpu()
                       for k, v in chosen.ensemble.state_dict().items()}
         if comm.is_initialized():
-            ens_sd = comm.broadcast_object(ens_sd, src=owner)
+            ens_sd = comm.broadcast_named_tensors(ens_sd, src=owner)
         self._best_ensemble_state = ens_sd
 
     # ------------------------------------------------------------------

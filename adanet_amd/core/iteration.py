@@ -568,14 +568,7 @@ class _Iteration(object):
                         if (self.placement is not None
                                 and self.placement.data_parallel
                                 and comm.world_size() > 1):
-                            arenas = getattr(spec.optimizer, "_arenas", None)
-                            if arenas:
-                                # Zero-copy flat buckets over xGMI.
-                                comm.allreduce_buffers(
-                                    spec.optimizer.flat_grad_buffers())
-                            else:
-                                comm.allreduce_gradients(
-                                    list(spec.subnetwork.module.parameters()))
+                            self._dp_allreduce(spec)
                         spec.optimizer.step()
                     losses_row[spec.name] = loss.detach()
                 else:
@@ -617,6 +610,41 @@ class _Iteration(object):
                 if spec.stream is not None:
                     cur.wait_stream(spec.stream)
         return losses_row
+
+    def _dp_allreduce(self, spec):
+        """Gradient all-reduce for one candidate (ReplicationStrategy).
+
+        On GPU the collective runs on a dedicated comm stream gated by an
+        event on the candidate's compute stream, so the xGMI ring of
+        candidate i overlaps candidate i+1's forward/backward enqueued
+        right after (the round-1 structure reduced strictly AFTER backward
+        on the compute stream — zero overlap). All ranks enqueue
+        collectives in identical spec order, so RCCL ordering holds.
+        """
+        arenas = getattr(spec.optimizer, "_arenas", None)
+        bufs = (spec.optimizer.flat_grad_buffers() if arenas else None)
+        if self.device.type == "cuda":
+            if not hasattr(self, "_comm_stream") or self._comm_stream is None:
+                self._comm_stream = torch.cuda.Stream(device=self.device)
+            cur = torch.cuda.current_stream(self.device)
+            ev = torch.cuda.Event()
+            ev.record(cur)
+            self._comm_stream.wait_event(ev)
+            with torch.cuda.stream(self._comm_stream):
+                if bufs is not None:
+                    comm.allreduce_buffers(bufs)
+                else:
+                    comm.allreduce_gradients(
+                        list(spec.subnetwork.module.parameters()))
+            # optimizer.step (enqueued next on `cur`) consumes the reduced
+            # grads: device-side join, no host sync.
+            cur.wait_stream(self._comm_stream)
+        else:
+            if bufs is not None:
+                comm.allreduce_buffers(bufs)
+            else:
+                comm.allreduce_gradients(
+                    list(spec.subnetwork.module.parameters()))
 
     def _subnetwork_loss(self, spec, logits, labels, features, frozen_out):
         """Head loss, or the builder's custom loss hook when provided

@@ -112,6 +112,64 @@ def allreduce_gradients(params: Sequence[torch.nn.Parameter]):
         off += n
 
 
+def broadcast_named_tensors(sd: Optional[dict], src: int = 0):
+    """Broadcasts a {name: tensor} state dict from `src` as ONE flat tensor
+    broadcast (+ a tiny shape-meta object), not a pickled blob.
+
+    Non-src ranks pass None and get back a CPU dict with src's contents.
+    Replaces object-pickle winner-weight broadcasts (a 4096-wide candidate
+    is tens of MB: pickling + eager deserialization is host-bound, a flat
+    tensor broadcast rides the xGMI/RCCL data plane directly).
+    """
+    if not is_initialized():
+        return sd
+    me = rank()
+    if me == src:
+        items = sorted((k, v) for k, v in (sd or {}).items()
+                       if torch.is_tensor(v))
+        meta = [(k, tuple(v.shape), str(v.dtype).replace("torch.", ""))
+                for k, v in items]
+        nontensor = {k: v for k, v in (sd or {}).items()
+                     if not torch.is_tensor(v)}
+        header = {"meta": meta, "none": sd is None, "extra": nontensor}
+    else:
+        header = None
+    header = broadcast_object(header, src=src)
+    if header["none"]:
+        return None
+    meta = header["meta"]
+    use_cuda = dist.get_backend() == "nccl"
+    dev = torch.device("cuda", torch.cuda.current_device()) if use_cuda \
+        else torch.device("cpu")
+    out = dict(header["extra"])
+    if not meta:
+        return out if me != src else sd
+    # one flat buffer per dtype (broadcast can't mix dtypes)
+    by_dtype = {}
+    for k, shape, dt in meta:
+        by_dtype.setdefault(dt, []).append((k, shape))
+    for dt, entries in sorted(by_dtype.items()):
+        dtype = getattr(torch, dt)
+        total = sum(int(torch.tensor(s).prod()) if s else 1
+                    for _, s in entries)
+        if me == src:
+            flat = torch.cat([
+                (sd[k]).reshape(-1).to(device=dev, dtype=dtype)
+                for k, _ in entries])
+        else:
+            flat = torch.empty(total, device=dev, dtype=dtype)
+        dist.broadcast(flat, src=src)
+        if me != src:
+            off = 0
+            for k, shape in entries:
+                n = 1
+                for d in shape:
+                    n *= d
+                out[k] = flat[off:off + n].reshape(shape).cpu()
+                off += n
+    return sd if me == src else out
+
+
 def broadcast_object(obj, src: int = 0):
     if not is_initialized():
         return obj

@@ -104,6 +104,11 @@ def _worker(rank, world, port, model_dir, placement_name, q,
     (2, "round_robin", False),
     (3, "round_robin", False),  # more ranks than candidates at t=0
     (2, "round_robin", True),   # mid-iteration checkpoint + resume
+    # reference exercises up to 5 workers (estimator_distributed_test.py:
+    # 198-277); one node of MI355X is 8 GPUs — cover both widths on gloo.
+    (4, "round_robin", False),
+    (4, "replication", False),
+    (8, "round_robin", False),
 ])
 def test_multi_rank_agreement(tmp_path, world, placement_name, resume):
     model_dir = str(tmp_path / "model")
@@ -196,3 +201,180 @@ def test_placement_truth_tables():
 
     rr2.config = _Cfg2()
     assert not rr2.should_train_subnetworks(2)
+
+
+def _crash_worker(rank, world, port, model_dir, phase, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import adanet_amd
+        from adanet_amd.distributed import RoundRobinStrategy
+        from adanet_amd.head import MultiClassHead
+        from adanet_amd.models import simple_dnn
+
+        torch.manual_seed(0)
+        N, D, C = 256, 8, 4
+        X = torch.randn(N, D)
+        W = torch.randn(D, C)
+        Y = (X @ W).argmax(dim=1)
+
+        def input_fn():
+            def gen():
+                g = torch.Generator().manual_seed(100 + rank)
+                while True:
+                    idx = torch.randint(0, N, (32,), generator=g)
+                    yield X[idx], Y[idx]
+
+            return gen()
+
+        est = adanet_amd.Estimator(
+            head=MultiClassHead(C),
+            subnetwork_generator=simple_dnn.Generator(layer_size=8),
+            max_iteration_steps=6,
+            model_dir=model_dir,
+            worker_wait_timeout_secs=20,
+            config=adanet_amd.RunConfig(tf_random_seed=42),
+            experimental_placement_strategy=RoundRobinStrategy(),
+            use_streams=False,
+        )
+        if phase == "crash":
+            # Both ranks reach the mid-iteration-1 checkpoint at step 9...
+            est.train(input_fn, steps=9)
+            assert est.iteration_number == 1
+            if rank == 1:
+                q.put((rank, "died-as-planned", est.global_step))
+                os._exit(77)  # hard worker death (reference kills procs)
+            # ... then the survivor hits the iteration-end collective and
+            # must FAIL LOUDLY (process-group timeout), not hang forever.
+            try:
+                est.train(input_fn, steps=3)
+                q.put((rank, "ERROR: survivor did not detect dead worker",
+                       est.global_step))
+            except Exception:
+                q.put((rank, "detected-dead-worker", est.global_step))
+        else:  # phase == "resume": fresh world restarts from the checkpoint
+            assert est.global_step == 9, est.global_step
+            assert est.iteration_number == 1
+            est.train(input_fn, max_steps=12)
+            q.put((rank, "resumed-ok", est.iteration_number))
+            torch.distributed.destroy_process_group()
+    except Exception:
+        import traceback
+        q.put((rank, "EXC:" + traceback.format_exc(), None))
+
+
+def test_killed_worker_then_restart_resumes(tmp_path):
+    """A worker dying mid-iteration must (a) not corrupt state — the last
+    all-rank checkpoint is intact, (b) surface as a loud failure on the
+    survivor, and (c) a restarted world resumes from the checkpoint and
+    completes (reference kill/exit-code handling,
+    estimator_distributed_test.py:131-195)."""
+    model_dir = str(tmp_path / "model")
+    os.makedirs(model_dir, exist_ok=True)
+    ctx = mp.get_context("spawn")
+
+    q = ctx.SimpleQueue()
+    port = _free_port()
+    procs = [ctx.Process(target=_crash_worker,
+                         args=(r, 2, port, model_dir, "crash", q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    msgs = {}
+    for _ in range(2):
+        rank, msg, extra = _get(q)
+        msgs[rank] = (msg, extra)
+    for p in procs:
+        p.join(timeout=120)
+    assert msgs[1][0] == "died-as-planned"
+    assert msgs[0][0] == "detected-dead-worker", msgs[0]
+
+    # restart a fresh 2-rank world on the same model_dir
+    q2 = ctx.SimpleQueue()
+    port2 = _free_port()
+    procs = [ctx.Process(target=_crash_worker,
+                         args=(r, 2, port2, model_dir, "resume", q2))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        rank, msg, it = _get(q2)
+        assert msg == "resumed-ok", (rank, msg)
+        assert it == 2
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert os.path.exists(os.path.join(model_dir, "architecture-1.json"))
+
+
+def _allstrategy_worker(rank, world, port, model_dir, q):
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import adanet_amd
+        from adanet_amd.distributed import RoundRobinStrategy
+        from adanet_amd.ensemble import AllStrategy
+        from adanet_amd.head import MultiClassHead
+        from adanet_amd.models import simple_dnn
+
+        torch.manual_seed(0)
+        N, D, C = 128, 8, 4
+        X = torch.randn(N, D)
+        Y = (X @ torch.randn(D, C)).argmax(dim=1)
+
+        def input_fn():
+            def gen():
+                g = torch.Generator().manual_seed(7 + rank)
+                while True:
+                    idx = torch.randint(0, N, (32,), generator=g)
+                    yield X[idx], Y[idx]
+
+            return gen()
+
+        est = adanet_amd.Estimator(
+            head=MultiClassHead(C),
+            subnetwork_generator=simple_dnn.Generator(layer_size=8),
+            max_iteration_steps=4,
+            ensemble_strategies=[AllStrategy()],
+            model_dir=model_dir,
+            config=adanet_amd.RunConfig(tf_random_seed=42),
+            experimental_placement_strategy=RoundRobinStrategy(),
+            use_streams=False,
+        )
+        est.train(input_fn, steps=4)
+        arch = json.loads(est._architectures[0])
+        q.put((rank, None, len(arch["subnetworks"])))
+        torch.distributed.destroy_process_group()
+    except Exception:
+        import traceback
+        q.put((rank, traceback.format_exc(), None))
+
+
+def test_allstrategy_candidate_not_split_across_ranks(tmp_path):
+    """A multi-builder candidate (AllStrategy) under round-robin world=2:
+    the co-location groups must put ALL its builders on one rank so the
+    candidate actually trains — previously it was built on NO rank and
+    silently reported inf loss (round-1 advisor finding)."""
+    model_dir = str(tmp_path / "model")
+    os.makedirs(model_dir, exist_ok=True)
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_allstrategy_worker,
+                         args=(r, 2, port, model_dir, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for _ in range(2):
+        rank, err, n_members = _get(q)
+        assert err is None, "rank %s failed:\n%s" % (rank, err)
+        # AllStrategy candidate = both simple_dnn builders
+        assert n_members == 2
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
