@@ -5,7 +5,8 @@
 #include <torch/extension.h>
 
 void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
-                  const c10::optional<at::Tensor>& bias, int64_t act);
+                  const c10::optional<at::Tensor>& bias, int64_t act,
+                  double dropout_p, const c10::optional<at::Tensor>& seed);
 void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
                         at::Tensor& C, int64_t variant);
 void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
@@ -74,9 +75,10 @@ void dropout_fwd(const at::Tensor& x, at::Tensor& y, double p,
                  const at::Tensor& seed);
 void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
                  const at::Tensor& seed);
-void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx);
+void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx,
+              double scale);
 void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
-                     at::Tensor& dz, at::Tensor& db);
+                     at::Tensor& dz, at::Tensor& db, double scale);
 void multi_copy_bf16(const std::vector<at::Tensor>& srcs,
                      const std::vector<at::Tensor>& dsts);
 void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum);
@@ -110,7 +112,11 @@ void batchnorm_bwd(const at::Tensor& x, const at::Tensor& dy, at::Tensor& dx,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "adanet_amd gfx950 (MI355X/CDNA4) kernels";
-  m.def("gemm_nt_bf16", &gemm_nt_bf16, "C[M,N] = A[M,K] @ B[N,K]^T, fused bias/relu");
+  m.def("gemm_nt_bf16", &gemm_nt_bf16,
+        "C[M,N] = A[M,K] @ B[N,K]^T, fused bias/relu/dropout",
+        py::arg("A"), py::arg("B"), py::arg("C"), py::arg("bias"),
+        py::arg("act"), py::arg("dropout_p") = 0.0,
+        py::arg("seed") = c10::nullopt);
   m.def("gemm_nt_bf16_probe", &gemm_nt_bf16_probe);
   m.def("gemm_nt8", &gemm_nt8,
         "8-phase deep-pipelined GEMM (256^2-class tiles)");
@@ -136,8 +142,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
-  m.def("relu_bwd", &relu_bwd);
-  m.def("relu_bwd_colsum", &relu_bwd_colsum);
+  m.def("relu_bwd", &relu_bwd, py::arg("dy"), py::arg("y"), py::arg("dx"),
+        py::arg("scale") = 1.0);
+  m.def("relu_bwd_colsum", &relu_bwd_colsum, py::arg("dy"), py::arg("y"),
+        py::arg("dz"), py::arg("db"), py::arg("scale") = 1.0);
   m.def("multi_copy_bf16", &multi_copy_bf16);
   m.def("colsum_bf16", &colsum_bf16, py::arg("x"), py::arg("out"),
         py::arg("accum") = 0);

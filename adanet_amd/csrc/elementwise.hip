@@ -44,11 +44,13 @@ __global__ __launch_bounds__(256) void dropout_bwd_kernel(
 // dX = dY * (Y > 0): Y is the fused GEMM+ReLU output.
 __global__ __launch_bounds__(256) void relu_bwd_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
-    bf16_t* __restrict__ dx, int64_t n) {
+    bf16_t* __restrict__ dx, int64_t n, float scale) {
   const int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
-    dx[i] = bf2f(y[i]) > 0.f ? dy[i] : f2bf(0.f);
+    // scale = 1/(1-p) under the fused relu+dropout epilogue: y>0 identifies
+    // kept-and-positive elements, so dropout backward is a scaled mask.
+    dx[i] = bf2f(y[i]) > 0.f ? f2bf(bf2f(dy[i]) * scale) : f2bf(0.f);
   }
 }
 
@@ -62,7 +64,7 @@ __global__ __launch_bounds__(256) void relu_bwd_kernel(
 __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
     bf16_t* __restrict__ dz, float* __restrict__ db, int B, int C,
-    int rows_per_block) {
+    int rows_per_block, float scale) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const int r0 = blockIdx.y * rows_per_block;
@@ -70,7 +72,7 @@ __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
   float acc = 0.f;
   for (int b = r0; b < r1; ++b) {
     const int64_t i = (int64_t)b * C + c;
-    const float g = bf2f(y[i]) > 0.f ? bf2f(dy[i]) : 0.f;
+    const float g = bf2f(y[i]) > 0.f ? bf2f(dy[i]) * scale : 0.f;
     dz[i] = f2bf(g);
     acc += g;
   }
@@ -168,18 +170,20 @@ void dropout_bwd(const at::Tensor& dy, at::Tensor& dx, double p,
   HIP_CHECK_KERNEL();
 }
 
-void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx) {
+void relu_bwd(const at::Tensor& dy, const at::Tensor& y, at::Tensor& dx,
+              double scale) {
   const int64_t n = dy.numel();
   if (n == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(relu_bwd_kernel, dim3(ew_grid(n)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)dy.data_ptr(),
-                     (const bf16_t*)y.data_ptr(), (bf16_t*)dx.data_ptr(), n);
+                     (const bf16_t*)y.data_ptr(), (bf16_t*)dx.data_ptr(), n,
+                     (float)scale);
   HIP_CHECK_KERNEL();
 }
 
 void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
-                     at::Tensor& dz, at::Tensor& db) {
+                     at::Tensor& dz, at::Tensor& db, double scale) {
   TORCH_CHECK(dy.is_contiguous() && y.is_contiguous() && dz.is_contiguous(),
               "relu_bwd_colsum: contiguous tensors required");
   TORCH_CHECK(db.scalar_type() == at::kFloat, "relu_bwd_colsum: fp32 db");
@@ -194,6 +198,7 @@ void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
                      dim3((unsigned)stripes, (unsigned)row_chunks), dim3(256),
                      0, stream.stream(), (const bf16_t*)dy.data_ptr(),
                      (const bf16_t*)y.data_ptr(), (bf16_t*)dz.data_ptr(),
-                     db.data_ptr<float>(), B, C, rows_per_block);
+                     db.data_ptr<float>(), B, C, rows_per_block,
+                     (float)scale);
   HIP_CHECK_KERNEL();
 }

@@ -33,6 +33,11 @@
 bool gemm_nt8_try(const bf16_t* a, const bf16_t* b, bf16_t* c,
                   const float* bias_ptr, int M, int N, int K, int64_t lda,
                   int64_t ldb, int64_t ldc, int act, hipStream_t stream);
+bool gemm_nt8_try_dropout(const bf16_t* a, const bf16_t* b, bf16_t* c,
+                          const float* bias_ptr, int M, int N, int K,
+                          int64_t lda, int64_t ldb, int64_t ldc,
+                          const long long* seed, uint32_t pthresh,
+                          float inv_keep, hipStream_t stream);
 
 typedef s16x8 frag_ab;
 
@@ -72,7 +77,9 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int lda, int ldb, int ldc, int act, int mtiles, int ntiles,
-    float* __restrict__ C32 = nullptr, int ksplit = 1) {
+    float* __restrict__ C32 = nullptr, int ksplit = 1,
+    const long long* __restrict__ seed = nullptr, uint32_t pthresh = 0,
+    float inv_keep = 1.f) {
   __shared__ bf16_t As[2][BM * KSTEP];
   __shared__ bf16_t Bs[2][BN * KSTEP];
 
@@ -184,6 +191,12 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
         } else {
           float v = acc[i][j][rr] + bv;
           if (act == 1) v = v > 0.f ? v : 0.f;
+          if (act == 3) {  // fused relu + dropout (stateless counter RNG)
+            v = v > 0.f ? v : 0.f;
+            const uint32_t r =
+                hash_rng((uint64_t)seed[0], (uint64_t)row * N + col);
+            v = (r >= pthresh) ? v * inv_keep : 0.f;
+          }
           // act==2: accumulate into C (direct-to-arena gradient writes).
           bf16_t* cp = &C[(int64_t)row * ldc + col];
           if (act == 2) v += bf2f(*cp);
@@ -219,7 +232,9 @@ struct bfx8 {
 __global__ __launch_bounds__(256) void gemm_nt_gemv_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
-    int K, int lda, int ldb, int ldc, int act) {
+    int K, int lda, int ldb, int ldc, int act,
+    const long long* __restrict__ seed = nullptr, uint32_t pthresh = 0,
+    float inv_keep = 1.f) {
   const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
   const int n = blockIdx.x * 4 + wid;
   if (n >= N) return;
@@ -240,6 +255,12 @@ __global__ __launch_bounds__(256) void gemm_nt_gemv_kernel(
     if (lane == 0) {
       if (bias) v += bias[n];
       if (act == 1) v = v > 0.f ? v : 0.f;
+      if (act == 3) {
+        v = v > 0.f ? v : 0.f;
+        const uint32_t r =
+            hash_rng((uint64_t)seed[0], (uint64_t)m * N + n);
+        v = (r >= pthresh) ? v * inv_keep : 0.f;
+      }
       bf16_t* cp = &C[(int64_t)m * ldc + n];
       if (act == 2) v += bf2f(*cp);
       *cp = f2bf(v);
@@ -254,7 +275,11 @@ __global__ void gemm_nt_generic_kernel(const bf16_t* __restrict__ A,
                                        bf16_t* __restrict__ C,
                                        const float* __restrict__ bias, int M,
                                        int N, int K, int lda, int ldb, int ldc,
-                                       int act) {
+                                       int act,
+                                       const long long* __restrict__ seed
+                                       = nullptr,
+                                       uint32_t pthresh = 0,
+                                       float inv_keep = 1.f) {
   const int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t total = (int64_t)M * N;
   for (int64_t p = idx; p < total; p += (int64_t)gridDim.x * blockDim.x) {
@@ -264,13 +289,19 @@ __global__ void gemm_nt_generic_kernel(const bf16_t* __restrict__ A,
       acc += bf2f(A[(int64_t)m * lda + k]) * bf2f(B[(int64_t)n * ldb + k]);
     if (bias) acc += bias[n];
     if (act == 1) acc = acc > 0.f ? acc : 0.f;
+    if (act == 3) {
+      acc = acc > 0.f ? acc : 0.f;
+      const uint32_t r = hash_rng((uint64_t)seed[0], (uint64_t)m * N + n);
+      acc = (r >= pthresh) ? acc * inv_keep : 0.f;
+    }
     if (act == 2) acc += bf2f(C[(int64_t)m * ldc + n]);
     C[(int64_t)m * ldc + n] = f2bf(acc);
   }
 }
 
 void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
-                  const c10::optional<at::Tensor>& bias, int64_t act) {
+                  const c10::optional<at::Tensor>& bias, int64_t act,
+                  double dropout_p, const c10::optional<at::Tensor>& seed) {
   TORCH_CHECK(A.is_cuda() && B.is_cuda() && C.is_cuda(), "gemm: need GPU tensors");
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 && B.scalar_type() == at::kBFloat16,
               "gemm: bf16 inputs required");
@@ -295,16 +326,38 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   const bf16_t* b = (const bf16_t*)B.data_ptr();
   bf16_t* c = (bf16_t*)C.data_ptr();
 
+  // act==3: fused relu+dropout epilogue (stateless counter RNG keyed on a
+  // DEVICE seed snapshot -> hipGraph replays draw fresh masks).
+  const long long* seed_ptr = nullptr;
+  uint32_t pthresh = 0;
+  float inv_keep = 1.f;
+  if (act == 3) {
+    TORCH_CHECK(seed.has_value() && seed->defined() && seed->is_cuda() &&
+                seed->scalar_type() == at::kLong,
+                "gemm: act=3 needs a device int64 seed tensor");
+    TORCH_CHECK(dropout_p > 0.0 && dropout_p < 1.0, "gemm: bad dropout_p");
+    seed_ptr = (const long long*)seed->data_ptr();
+    pthresh = (uint32_t)(dropout_p * 4294967296.0);
+    inv_keep = (float)(1.0 / (1.0 - dropout_p));
+  }
+
   const bool fast = (K % 32 == 0) && (lda % 8 == 0) && (ldb % 8 == 0);
   if (fast && M <= 8) {
     hipLaunchKernelGGL(gemm_nt_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
-                       ldc, (int)act);
+                       ldc, (int)act, seed_ptr, pthresh, inv_keep);
     HIP_CHECK_KERNEL();
     return;
   }
-  if (fast && gemm_nt8_try(a, b, c, bias_ptr, M, N, K, lda, ldb, ldc,
-                           (int)act, stream.stream())) {
+  if (fast && act != 3 &&
+      gemm_nt8_try(a, b, c, bias_ptr, M, N, K, lda, ldb, ldc,
+                   (int)act, stream.stream())) {
+    HIP_CHECK_KERNEL();
+    return;
+  }
+  if (fast && act == 3 &&
+      gemm_nt8_try_dropout(a, b, c, bias_ptr, M, N, K, lda, ldb, ldc,
+                           seed_ptr, pthresh, inv_keep, stream.stream())) {
     HIP_CHECK_KERNEL();
     return;
   }
@@ -321,7 +374,7 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL((gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32>),         \
                        dim3(mt * nt), dim3(THREADS), 0, stream.stream(), a,  \
                        b, c, bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, \
-                       nt);                                                   \
+                       nt, nullptr, 1, seed_ptr, pthresh, inv_keep);          \
   } while (0)
 #define LAUNCH_CFG_W(BM, BN, FM, FN, MW, WGM, WGN, GM)                        \
   do {                                                                        \
@@ -329,7 +382,8 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL(                                                       \
         (gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, 32, false, WGM, WGN, GM>),   \
         dim3(mt * nt), dim3(WGM * WGN * 64), 0, stream.stream(), a, b, c,    \
-        bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                  \
+        bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt, nullptr, 1,       \
+        seed_ptr, pthresh, inv_keep);                                         \
   } while (0)
     const int64_t t_256x128 =
         (int64_t)((M + 255) / 256) * ((N + 127) / 128);
@@ -359,7 +413,7 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     const int blocks = (int)std::min<int64_t>((total + 255) / 256, 2048);
     hipLaunchKernelGGL(gemm_nt_generic_kernel, dim3(blocks), dim3(256), 0,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
-                       ldc, (int)act);
+                       ldc, (int)act, seed_ptr, pthresh, inv_keep);
   }
   HIP_CHECK_KERNEL();
 }

@@ -111,7 +111,8 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
-    int ntiles) {
+    int ntiles, const long long* __restrict__ seed = nullptr,
+    uint32_t pthresh = 0, float inv_keep = 1.f) {
   constexpr int FM = BM / (2 * 16);   // m-fragments per wave
   constexpr int FN = BN / (4 * 16);   // n-fragments per wave
   constexpr int MG = FM / 4;          // m-reps per phase
@@ -251,6 +252,12 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8_kernel(
         if (row >= M) continue;
         float v = acc[i][j][rr] + bv;
         if (act == 1) v = v > 0.f ? v : 0.f;
+        if (act == 3) {  // fused relu+dropout (stateless counter RNG)
+          v = v > 0.f ? v : 0.f;
+          const uint32_t r =
+              hash_rng((uint64_t)seed[0], (uint64_t)row * N + col);
+          v = (r >= pthresh) ? v * inv_keep : 0.f;
+        }
         bf16_t* cp = &C[(int64_t)row * ldc + col];
         if (act == 2) v += bf2f(*cp);
         *cp = f2bf(v);
@@ -837,6 +844,34 @@ bool gemm_nt8_try(const bf16_t* a, const bf16_t* b, bf16_t* c,
     hipLaunchKernelGGL((gemm_nt8_kernel<128, 128, false, false>),
                        dim3(mt * nt), dim3(512), 0, stream, a, b, c,
                        bias_ptr, M, N, K, lda, ldb, ldc, act, mt, nt);
+    return true;
+  }
+  return false;
+}
+
+// Dropout twin of gemm_nt8_try: same tile thresholds, act==3 epilogue.
+bool gemm_nt8_try_dropout(const bf16_t* a, const bf16_t* b, bf16_t* c,
+                          const float* bias_ptr, int M, int N, int K,
+                          int64_t lda, int64_t ldb, int64_t ldc,
+                          const long long* seed, uint32_t pthresh,
+                          float inv_keep, hipStream_t stream) {
+  if (K % 64 != 0 || K < 128 || lda % 8 || ldb % 8) return false;
+  const int64_t t256 = (int64_t)((M + 255) / 256) * ((N + 255) / 256);
+  const int64_t t128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
+  if (t256 >= 200) {
+    const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+    hipLaunchKernelGGL((gemm_nt8_kernel<256, 256, false, false>),
+                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
+                       bias_ptr, M, N, K, lda, ldb, ldc, 3, mt, nt, seed,
+                       pthresh, inv_keep);
+    return true;
+  }
+  if (t128 >= 256) {
+    const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+    hipLaunchKernelGGL((gemm_nt8_kernel<128, 128, false, false>),
+                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
+                       bias_ptr, M, N, K, lda, ldb, ldc, 3, mt, nt, seed,
+                       pthresh, inv_keep);
     return true;
   }
   return false;

@@ -40,9 +40,12 @@ class _DNNModule(SubnetworkModule):
         layers = []
         d = in_dim
         for _ in range(num_layers):
-            layers.append(HipLinear(d, layer_size, activation="relu"))
-            if dropout > 0:
-                layers.append(HipDropout(dropout))
+            # dropout is FUSED into the GEMM's relu epilogue (gemm.hip
+            # act=3, stateless counter RNG): the paper's relu->dropout
+            # layer (reference simple_dnn.py:77-81) costs zero extra
+            # kernels and stays hipGraph-capturable.
+            layers.append(HipLinear(d, layer_size, activation="relu",
+                                    dropout=dropout))
             d = layer_size
         self.hidden = nn.Sequential(*layers)
         self.logits_layer = HipLinear(d, logits_dim)

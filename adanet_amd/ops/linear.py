@@ -106,13 +106,20 @@ def restore_fp32_params(module) -> None:
 
 def gemm_nt(a: torch.Tensor, b: torch.Tensor,
             bias: Optional[torch.Tensor] = None,
-            activation: Optional[str] = None) -> torch.Tensor:
-    """C[M,N] = A[M,K] @ B[N,K]^T (+bias, +relu). bf16 on GPU, fp32 on CPU."""
+            activation: Optional[str] = None,
+            dropout_p: float = 0.0,
+            seed: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """C[M,N] = A[M,K] @ B[N,K]^T (+bias, +relu[, +fused dropout]).
+    bf16 on GPU, fp32 on CPU."""
     if a.is_cuda:
         ext = _extension.require()
         out = torch.empty((a.shape[0], b.shape[0]), device=a.device,
                           dtype=torch.bfloat16)
-        ext.gemm_nt_bf16(a, b, out, bias, 1 if activation == "relu" else 0)
+        if dropout_p > 0.0 and activation == "relu":
+            ext.gemm_nt_bf16(a, b, out, bias, 3, dropout_p, seed)
+        else:
+            ext.gemm_nt_bf16(a, b, out, bias,
+                             1 if activation == "relu" else 0)
         return out
     out = a.float() @ b.float().t()
     if bias is not None:
@@ -135,10 +142,11 @@ def transpose2d(x: torch.Tensor) -> torch.Tensor:
 class _LinearFn(torch.autograd.Function):
 
     @staticmethod
-    def forward(ctx, x, weight, bias, activation):
+    def forward(ctx, x, weight, bias, activation, dropout_p=0.0, seed=None):
         x = x.contiguous()
-        y = gemm_nt(x, weight, bias, activation)
+        y = gemm_nt(x, weight, bias, activation, dropout_p, seed)
         ctx.activation = activation
+        ctx.dropout_p = dropout_p
         ctx.has_bias = bias is not None
         # Keep the live Parameter objects so backward can reach their .grad
         # views for direct-to-arena writes (saved_tensors unwraps to plain
@@ -165,14 +173,19 @@ class _LinearFn(torch.autograd.Function):
                         and bg.is_contiguous())
             db_done = False
             if ctx.activation == "relu":
+                # Under the fused relu+dropout epilogue, y>0 identifies
+                # kept-and-positive elements: dropout backward is the same
+                # mask with a 1/(1-p) scale — no mask recompute needed.
+                scale = (1.0 / (1.0 - ctx.dropout_p)
+                         if ctx.dropout_p > 0.0 else 1.0)
                 dz = torch.empty_like(dy)
                 if direct_b and dy.is_contiguous() and y.is_contiguous():
                     # Fused mask + bias-grad colsum: one read of dY/Y
                     # produces dz AND accumulates db into the arena view.
-                    ext.relu_bwd_colsum(dy, y, dz, bg)
+                    ext.relu_bwd_colsum(dy, y, dz, bg, scale)
                     db_done = True
                 else:
-                    ext.relu_bwd(dy, y, dz)
+                    ext.relu_bwd(dy, y, dz, scale)
             else:
                 dz = dy
             dx = dw = None
@@ -240,10 +253,12 @@ class _LinearFn(torch.autograd.Function):
             dzf = dy.float()
             if ctx.activation == "relu":
                 dzf = dzf * (y > 0).float() if y is not None else dzf
+                if ctx.dropout_p > 0.0:
+                    dzf = dzf / (1.0 - ctx.dropout_p)
             dx = (dzf @ weight.float()).to(x.dtype) if need_dx else None
             dw = (dzf.t() @ x.float()).to(weight.dtype) if need_dw else None
             db = dzf.sum(dim=0) if ctx.has_bias else None
-        return dx, dw, db, None
+        return dx, dw, db, None, None, None
 
 
 class HipLinear(nn.Module):
@@ -251,14 +266,27 @@ class HipLinear(nn.Module):
 
     def __init__(self, in_features: int, out_features: int, bias: bool = True,
                  activation: Optional[str] = None, device=None,
-                 dtype: torch.dtype = torch.bfloat16):
+                 dtype: torch.dtype = torch.bfloat16,
+                 dropout: float = 0.0):
         super().__init__()
         if activation not in (None, "relu"):
             raise ValueError("activation must be None or 'relu'")
+        if dropout and activation != "relu":
+            raise ValueError("fused dropout requires activation='relu'")
+        if not 0.0 <= dropout < 1.0:
+            raise ValueError("dropout must be in [0, 1)")
         self.in_features = in_features
         self.out_features = out_features
         self.padded_out = _pad8(out_features)
         self.activation = activation
+        # Fused dropout (paper search space: relu->dropout per layer,
+        # reference simple_dnn.py:77-81). Device counter seed so hipGraph
+        # replays draw fresh masks (same scheme as HipDropout).
+        self.dropout = float(dropout)
+        self.register_buffer("_drop_seed",
+                             torch.zeros(1, dtype=torch.int64),
+                             persistent=False)
+        self._drop_seeded = False
         self.weight = nn.Parameter(
             torch.empty((self.padded_out, in_features), device=device,
                         dtype=dtype))
@@ -288,8 +316,23 @@ class HipLinear(nn.Module):
     def forward(self, x):
         if x.dim() > 2:
             x = x.reshape(x.shape[0], -1)
-        y = _LinearFn.apply(x.to(self.weight.dtype), self.weight, self.bias,
-                            self.activation)
+        p = self.dropout if (self.training and self.dropout > 0.0) else 0.0
+        if p > 0.0 and x.is_cuda:
+            if not self._drop_seeded or self._drop_seed.device != x.device:
+                base = int(torch.initial_seed()) & 0x7FFFFFFFFFFF
+                self._drop_seed = torch.tensor(
+                    [base + id(self) % 100003], dtype=torch.int64,
+                    device=x.device)
+                self._drop_seeded = True
+            snapshot = self._drop_seed.clone()      # device op (capturable)
+            self._drop_seed.add_(0x9E3779B9)        # device op (capturable)
+            y = _LinearFn.apply(x.to(self.weight.dtype), self.weight,
+                                self.bias, self.activation, p, snapshot)
+        else:
+            y = _LinearFn.apply(x.to(self.weight.dtype), self.weight,
+                                self.bias, self.activation)
+            if p > 0.0:  # CPU fallback: unfused functional dropout
+                y = torch.nn.functional.dropout(y, p, training=True)
         if self.padded_out != self.out_features:
             y = y[:, :self.out_features]
         return y

@@ -49,6 +49,9 @@ def parse_args():
                         "GPU so round-robin owns 2 candidates per rank)")
     p.add_argument("--placement", choices=["replication", "round_robin"],
                    default="round_robin")
+    p.add_argument("--dropout", type=float, default=0.0,
+                   help="per-layer dropout (fused into the GEMM relu "
+                        "epilogue; the paper space trains with dropout)")
     p.add_argument("--lr", type=float, default=0.05)
     p.add_argument("--model-dir", default=None)
     p.add_argument("--cpu", action="store_true", help="debug on CPU")
@@ -162,7 +165,7 @@ def main():
         layer_size=args.hidden,
         initial_num_layers=1,
         learn_mixture_weights=True,
-        dropout=0.0,
+        dropout=args.dropout,
         seed=77,
         num_restarts=restarts)
 
@@ -261,6 +264,7 @@ def main():
                 "candidates_per_iter": 2 * restarts,
                 "train_steps_per_iter": args.train_steps_per_iter,
                 "eval_batches": args.eval_batches,
+                "dropout": args.dropout,
                 "parallelism": ("dp%d" % world
                                 if args.placement == "replication" else
                                 "round_robin%d" % world),

@@ -741,3 +741,55 @@ def test_pool3_fwd_bwd_vs_fp32(kind, stride):
     yr.backward(g)
     # dx is bf16: up to 9 summed dy values -> one-ulp ~0.06 at |dx|~6
     assert (xb.grad.float() - xf.grad).abs().max().item() < 0.08
+
+
+# ------------------------------------------------- fused dropout epilogue
+def test_fused_gemm_dropout_statistics_and_grads():
+    """act=3 GEMM epilogue: kept fraction ~= 1-p, kept values = relu/(1-p),
+    and backward is the y>0 mask scaled by 1/(1-p) (no mask storage)."""
+    from adanet_amd.ops.linear import HipLinear
+    torch.manual_seed(33)
+    B, D, H, p = 512, 256, 512, 0.3
+    lin = HipLinear(D, H, activation="relu", dropout=p).to(DEV)
+    lin = lin.to(torch.bfloat16)
+    from adanet_amd.ops.linear import restore_fp32_params
+    restore_fp32_params(lin)
+    x = torch.randn(B, D, device=DEV).to(torch.bfloat16).requires_grad_(True)
+    lin.train()
+    y = lin(x)
+    # reference pre-dropout activation
+    with torch.no_grad():
+        ref = torch.relu(x.float() @ lin.weight.float().t() +
+                         lin.bias.float())[:, :H]
+    kept = (y.float() > 0)
+    pos = ref > 1e-3
+    drop_rate = 1.0 - kept[pos].float().mean().item()
+    assert abs(drop_rate - p) < 0.03, drop_rate
+    # kept positives match ref/(1-p)
+    sel = kept & pos
+    ratio = (y.float()[sel] / ref[sel])
+    assert (ratio - 1.0 / (1.0 - p)).abs().max().item() < 0.05
+    # backward: dx through the scaled mask
+    g = torch.randn_like(y)
+    y.backward(g)
+    with torch.no_grad():
+        dz = g.float() * kept.float() / (1.0 - p)
+        dx_ref = dz @ lin.weight.float()
+    assert (x.grad.float() - dx_ref).abs().max().item() < 0.15
+
+
+def test_fused_dropout_fresh_mask_per_step_and_eval_off():
+    from adanet_amd.ops.linear import HipLinear, restore_fp32_params
+    torch.manual_seed(34)
+    lin = HipLinear(64, 128, activation="relu", dropout=0.5).to(DEV)
+    lin = lin.to(torch.bfloat16)
+    restore_fp32_params(lin)
+    x = torch.randn(256, 64, device=DEV).to(torch.bfloat16)
+    lin.train()
+    y1 = lin(x)
+    y2 = lin(x)
+    assert not torch.equal(y1, y2), "mask must differ across steps"
+    lin.eval()
+    y3 = lin(x)
+    ref = torch.relu(x.float() @ lin.weight.float().t() + lin.bias.float())
+    assert (y3.float() - ref).abs().max().item() < 0.1
