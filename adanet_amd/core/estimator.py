@@ -267,8 +267,12 @@ class Estimator(object):
         placement = kwargs.pop("experimental_placement_strategy", None)
         self._placement: PlacementStrategy = placement or ReplicationStrategy()
         self._placement.config = self._config
-        self._use_streams = kwargs.pop("use_streams", True)
-        self._use_hip_graphs = kwargs.pop("use_hip_graphs", True)
+        self._use_streams = kwargs.pop(
+            "use_streams",
+            os.environ.get("ADANET_NO_STREAMS", "") in ("", "0"))
+        self._use_hip_graphs = kwargs.pop(
+            "use_hip_graphs",
+            os.environ.get("ADANET_NO_GRAPHS", "") in ("", "0"))
         if kwargs:
             raise ValueError("Unknown kwargs: %s" % sorted(kwargs))
 

@@ -37,6 +37,8 @@ class _DNNModule(SubnetworkModule):
     def __init__(self, in_dim: int, num_layers: int, layer_size: int,
                  logits_dim: int, dropout: float):
         super().__init__()
+        import os
+        unfused = os.environ.get("ADANET_UNFUSED_DROPOUT", "") not in ("", "0")
         layers = []
         d = in_dim
         for _ in range(num_layers):
@@ -45,7 +47,9 @@ class _DNNModule(SubnetworkModule):
             # layer (reference simple_dnn.py:77-81) costs zero extra
             # kernels and stays hipGraph-capturable.
             layers.append(HipLinear(d, layer_size, activation="relu",
-                                    dropout=dropout))
+                                    dropout=0.0 if unfused else dropout))
+            if unfused and dropout > 0:
+                layers.append(HipDropout(dropout))
             d = layer_size
         self.hidden = nn.Sequential(*layers)
         self.logits_layer = HipLinear(d, logits_dim)

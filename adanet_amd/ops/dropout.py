@@ -48,6 +48,10 @@ class HipDropout(nn.Module):
         # graph replays draw fresh masks. Seeded lazily from torch's RNG.
         self.register_buffer("_seed", torch.zeros(1, dtype=torch.int64),
                              persistent=False)
+        # persistent snapshot buffer: no alloc/free inside graph capture
+        # (see ops/linear.py HipLinear._drop_snap).
+        self.register_buffer("_snap", torch.zeros(1, dtype=torch.int64),
+                             persistent=False)
         self._seeded = False
 
     def forward(self, x):
@@ -59,8 +63,10 @@ class HipDropout(nn.Module):
                 self._seed = torch.tensor(
                     [base + id(self) % 100003], dtype=torch.int64,
                     device=x.device)
+                self._snap = torch.zeros_like(self._seed)
                 self._seeded = True
-            snapshot = self._seed.clone()          # device op (capturable)
+            self._snap.copy_(self._seed)           # device op (capturable)
             self._seed.add_(0x9E3779B9)            # device op (capturable)
+            snapshot = self._snap
             return _DropoutFn.apply(x.to(torch.bfloat16), self.p, snapshot)
         return torch.nn.functional.dropout(x, self.p, training=True)

@@ -286,6 +286,14 @@ class HipLinear(nn.Module):
         self.register_buffer("_drop_seed",
                              torch.zeros(1, dtype=torch.int64),
                              persistent=False)
+        # persistent snapshot target: the per-step seed snapshot must NOT
+        # allocate inside a hipGraph capture (an alloc+free on a candidate
+        # stream mid-capture perturbs the graph pool's cross-stream block
+        # reuse -- measured as the ensemble mixer reading garbage on
+        # replay, benchmarks/nan_inspect2 r02).
+        self.register_buffer("_drop_snap",
+                             torch.zeros(1, dtype=torch.int64),
+                             persistent=False)
         self._drop_seeded = False
         self.weight = nn.Parameter(
             torch.empty((self.padded_out, in_features), device=device,
@@ -323,9 +331,13 @@ class HipLinear(nn.Module):
                 self._drop_seed = torch.tensor(
                     [base + id(self) % 100003], dtype=torch.int64,
                     device=x.device)
+                self._drop_snap = torch.zeros_like(self._drop_seed)
                 self._drop_seeded = True
-            snapshot = self._drop_seed.clone()      # device op (capturable)
-            self._drop_seed.add_(0x9E3779B9)        # device op (capturable)
+            # copy_ into the PERSISTENT snapshot buffer: pure device ops,
+            # zero allocations inside any capture.
+            self._drop_snap.copy_(self._drop_seed)
+            self._drop_seed.add_(0x9E3779B9)
+            snapshot = self._drop_snap
             y = _LinearFn.apply(x.to(self.weight.dtype), self.weight,
                                 self.bias, self.activation, p, snapshot)
         else:
