@@ -65,8 +65,16 @@ class HipDropout(nn.Module):
                     device=x.device)
                 self._snap = torch.zeros_like(self._seed)
                 self._seeded = True
-            self._snap.copy_(self._seed)           # device op (capturable)
-            self._seed.add_(0x9E3779B9)            # device op (capturable)
-            snapshot = self._snap
+            if torch.cuda.is_current_stream_capturing():
+                # No alloc/free inside capture (graph-pool cross-stream
+                # reuse race, see ops/linear.py); single forward per
+                # captured step is the engine's contract.
+                self._snap.copy_(self._seed)
+                snapshot = self._snap
+            else:
+                # eager: a fresh clone so backward sees ITS forward's seed
+                # even when the module runs twice before backward.
+                snapshot = self._seed.clone()
+            self._seed.add_(0x9E3779B9)
             return _DropoutFn.apply(x.to(torch.bfloat16), self.p, snapshot)
         return torch.nn.functional.dropout(x, self.p, training=True)
