@@ -954,6 +954,9 @@ class Estimator(object):
         if self._replay_config is not None:
             replay_index = self._replay_config.get_best_ensemble_index(t)
         if replay_index is not None:
+            # Replay overrides selection ENTIRELY — force_grow never
+            # re-routes a replayed index (reference returns early,
+            # estimator.py:1433-1438).
             best_index = replay_index
             losses = iteration.adanet_losses()
         elif self._evaluator is not None:
@@ -973,9 +976,12 @@ class Estimator(object):
             best_index = iteration.best_candidate_index(losses=losses)
 
         # force_grow (reference :1448-1512): never keep the previous
-        # ensemble when growth is possible.
+        # ensemble when growth is possible (the previous ensemble is
+        # excluded from the argmin and the best GROWING candidate wins).
+        # Never applied to a replayed index (reference early-return).
         chosen = iteration.ensemble_specs[best_index]
-        if (self._force_grow and chosen.is_previous_best
+        if (replay_index is None and self._force_grow
+                and chosen.is_previous_best
                 and len(iteration.ensemble_specs) > 1):
             grow_indices = [
                 i for i, s in enumerate(iteration.ensemble_specs)
@@ -988,14 +994,6 @@ class Estimator(object):
         best_index = comm.broadcast_object(best_index, src=0) if (
             comm.is_initialized()) else best_index
         chosen = iteration.ensemble_specs[best_index]
-        # A NaN winner means the selected ensemble diverged (selection maps
-        # NaN -> -inf so divergence surfaces, reference iteration.py:1040-46;
-        # tf.estimator then raises NanLossDuringTrainingError). Fail loudly
-        # instead of silently freezing a garbage model.
-        if best_index < len(losses) and math.isnan(losses[best_index]):
-            raise NanLossDuringTrainingError(
-                "Iteration {}: selected candidate {!r} has NaN adanet_loss "
-                "(training diverged).".format(t, chosen.name))
         # Per-candidate eval summaries (the analog of _EvalMetricSaverHook's
         # per-candidate eval dirs, reference estimator.py:150-233).
         for i, spec in enumerate(iteration.ensemble_specs):

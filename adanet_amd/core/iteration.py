@@ -859,18 +859,26 @@ class _Iteration(object):
     def best_candidate_index(self,
                              override: Optional[int] = None,
                              losses: Optional[Sequence[float]] = None) -> int:
-        """argmin over adanet losses with NaN -> -inf (NaN candidates WIN so
-        divergence surfaces — reference iteration.py:1011-1047)."""
+        """np.nanargmin over adanet losses: diverged (NaN) candidates LOSE
+        (reference bookkeeping selection, estimator.py:1494-1512 uses
+        np.nanargmin — distinct from the in-graph prediction mux's NaN->-inf
+        at iteration.py:1040-1046). All-NaN raises like np.nanargmin."""
         if override is not None:
             return int(override)
         if len(self.candidates) == 1:
             return 0
         vals = list(losses) if losses is not None else self.adanet_losses()
-        vals = [-float("inf") if math.isnan(v) else v for v in vals]
         best, best_i = None, 0
         for i, v in enumerate(vals):
+            if math.isnan(v):
+                continue
             if best is None or v < best:
                 best, best_i = v, i
+        if best is None:
+            from adanet_amd.core.estimator import NanLossDuringTrainingError
+            raise NanLossDuringTrainingError(
+                "Iteration {}: every candidate's adanet_loss is NaN "
+                "(training diverged).".format(self.number))
         return best_i
 
     # ------------------------------------------------------------------

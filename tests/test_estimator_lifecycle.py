@@ -210,22 +210,48 @@ class _GoodBuilder(Builder):
         return Subnetwork(module=_M(), complexity=1.0)
 
 
-def test_nan_winner_raises_nan_loss_error(model_dir,
-                                          synthetic_classification):
-    """Selection maps NaN -> -inf so a diverged candidate WINS argmin
-    (reference iteration.py:1040-1046) — and the engine must then fail
-    loudly, like tf.estimator's NanLossDuringTrainingError, instead of
-    silently freezing a garbage model."""
-    import pytest
+def test_nan_candidate_loses_selection(model_dir, synthetic_classification):
+    """Reference bookkeeping selection is np.nanargmin (estimator.py:
+    1494-1512): a diverged (NaN) candidate LOSES and the healthy candidate
+    is frozen — divergence does not abort the search."""
     X, Y, input_fn = synthetic_classification
     est = _make_estimator(
         model_dir, input_fn,
         subnetwork_generator=SimpleGenerator(
             [_NanBuilder(), _GoodBuilder()]))
+    est.train(input_fn, max_steps=10)
+    arch = json.loads(
+        open(os.path.join(model_dir, "architecture-0.json")).read())
+    assert arch["subnetworks"][0]["builder_name"] == "good"
+
+
+def test_all_nan_candidates_raise(model_dir, synthetic_classification):
+    """ALL candidates NaN == np.nanargmin's all-NaN error: fail loudly
+    (NanLossDuringTrainingError), never freeze a garbage model."""
+    import pytest
+    X, Y, input_fn = synthetic_classification
+    est = _make_estimator(
+        model_dir, input_fn,
+        subnetwork_generator=SimpleGenerator([_NanBuilder()]),
+        # two candidates (the single-candidate path returns 0 without
+        # inspecting losses, like the reference's early return)
+        ensemble_strategies=None)
+    import adanet_amd.ensemble as ens
+
+    class _NanBuilder2(_NanBuilder):
+
+        @property
+        def name(self):
+            return "nan2"
+
+    est = _make_estimator(
+        str(model_dir) + "_2", input_fn,
+        subnetwork_generator=SimpleGenerator(
+            [_NanBuilder(), _NanBuilder2()]))
     with pytest.raises(adanet_amd.NanLossDuringTrainingError):
         est.train(input_fn, max_steps=10)
-    # Divergence surfaced before any architecture was frozen.
-    assert not os.path.exists(os.path.join(model_dir, "architecture-0.json"))
+    assert not os.path.exists(
+        os.path.join(str(model_dir) + "_2", "architecture-0.json"))
 
 
 def test_evaluator_selects_on_eval_loss(model_dir, synthetic_classification):
@@ -764,3 +790,77 @@ def test_debug_mode_raises_on_nonfinite_input(model_dir):
     est = _make_estimator(model_dir, input_fn, debug=True)
     with pytest.raises(ValueError):
         est.train(input_fn, max_steps=5)
+
+
+@pytest.mark.parametrize("force_grow,use_replay,use_evaluator", [
+    (True, False, False),
+    (True, False, True),
+    (False, False, False),
+    (True, True, False),   # replay overrides force_grow (reference
+    (True, True, True),    # early-return, estimator.py:1433-1438)
+    (False, True, False),
+])
+def test_selection_contract_grid(model_dir, synthetic_classification,
+                                 force_grow, use_replay, use_evaluator):
+    """Reference golden selection behaviors (estimator.py:1415-1517) over
+    {force_grow x replay x evaluator-absent}:
+      * force_grow: the previous ensemble never wins when growth is
+        possible -> member count strictly grows each iteration;
+      * replay pins the winner index and is NEVER re-routed by force_grow
+        (a replayed index 0 at t>0 keeps the previous ensemble even with
+        force_grow=True);
+      * evaluator presence only changes the loss source, not the rules."""
+    X, Y, input_fn = synthetic_classification
+    kwargs = dict(
+        subnetwork_generator=SimpleGenerator(
+            [_GoodBuilder("g0"), _GoodBuilder("g1")]),
+        force_grow=force_grow,
+    )
+    if use_replay:
+        import adanet_amd.replay as replay
+        kwargs["replay_config"] = replay.Config(
+            best_ensemble_indices=[0, 0, 0])
+    if use_evaluator:
+        kwargs["evaluator"] = adanet_amd.Evaluator(input_fn=input_fn,
+                                                   steps=2)
+    est = _make_estimator(model_dir, input_fn, **kwargs)
+    est.train(input_fn, max_steps=30)  # 3 iterations x 10 steps
+    sizes = []
+    for t in range(3):
+        arch = json.loads(open(os.path.join(
+            model_dir, "architecture-%d.json" % t)).read())
+        sizes.append(len(arch["subnetworks"]))
+    if use_replay:
+        # replayed index 0: t0 picks candidate g0 (1 member); t>=1 index 0
+        # is the previous ensemble -> the ensemble NEVER grows, even with
+        # force_grow on.
+        assert sizes == [1, 1, 1], sizes
+    elif force_grow:
+        # growth every iteration, previous ensemble never kept
+        assert sizes == [1, 2, 3], sizes
+    else:
+        # monotone non-shrinking; may plateau when previous wins
+        assert all(b >= a for a, b in zip(sizes, sizes[1:])), sizes
+
+
+def test_best_metrics_mux_with_dict_metrics(model_dir,
+                                            synthetic_classification):
+    """Dict-valued metric_fn flows through the per-candidate stores and the
+    best-mux returns the WINNER's dict (reference eval_metrics.py:306-408
+    best_eval_metrics_tuple semantics)."""
+    X, Y, input_fn = synthetic_classification
+
+    def metric_fn(predictions, features, labels):
+        probs = predictions["probabilities"]
+        return {
+            "custom/max_prob": float(probs.max()),
+            "custom/n_examples": float(probs.shape[0]),
+        }
+
+    est = _make_estimator(model_dir, input_fn, metric_fn=metric_fn)
+    est.train(input_fn, max_steps=10)
+    res = est.evaluate(input_fn, steps=2)
+    assert "custom/max_prob" in res and 0 < res["custom/max_prob"] <= 1.0
+    assert res["custom/n_examples"] > 0
+    assert "best_ensemble_index_0" in res
+    assert "architecture/adanet/ensembles" in res
