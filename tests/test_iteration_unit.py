@@ -102,15 +102,21 @@ def test_iteration_stops_at_max_steps(tmp_path):
     assert [s.step for s in it.subnetwork_specs] == steps_before
 
 
-def test_best_candidate_nan_to_neg_inf(tmp_path):
+def test_best_candidate_nanargmin(tmp_path):
+    """Selection is np.nanargmin (reference estimator.py:1494-1512): NaN
+    candidates lose; all-NaN raises; override bypasses."""
+    import pytest
+    from adanet_amd.core.estimator import NanLossDuringTrainingError
     it, x = _make_iteration(tmp_path, n_builders=3)
     it.candidates[0].update(0.5)
     it.candidates[1].update(float("nan"))
     it.candidates[2].update(0.1)
-    # NaN -> -inf wins argmin (reference iteration.py:1040-1046)
-    assert it.best_candidate_index(losses=[0.5, float("nan"), 0.1]) == 1
+    assert it.best_candidate_index(losses=[0.5, float("nan"), 0.1]) == 2
+    assert it.best_candidate_index(losses=[float("nan"), 0.3, 0.4]) == 1
     assert it.best_candidate_index(losses=[0.5, 0.3, 0.1]) == 2
     assert it.best_candidate_index(override=0) == 0
+    with pytest.raises(NanLossDuringTrainingError):
+        it.best_candidate_index(losses=[float("nan")] * 3)
 
 
 def test_train_manager_persistence(tmp_path):
