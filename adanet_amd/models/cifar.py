@@ -109,42 +109,111 @@ class FakeImageProvider(Provider):
         return self._X, self._Y
 
 
+def _load_cifar_binaries(data_dir: str, n_classes: int, training: bool):
+    """Reads the standard CIFAR binary distribution (the format in
+    cifar-10-binary.tar.gz / cifar-100-binary.tar.gz):
+
+      * CIFAR-10:  data_batch_{1..5}.bin / test_batch.bin, records of
+        1 label byte + 3072 RGB bytes;
+      * CIFAR-100: train.bin / test.bin, records of 2 label bytes
+        (coarse, fine) + 3072 RGB bytes.
+
+    Returns (x [N,3,32,32] float in [0,1], y [N] long) or None if the
+    files are absent. No torchvision / no network required.
+    """
+    import os
+
+    import numpy as np
+    if n_classes == 10:
+        names = (["data_batch_%d.bin" % i for i in range(1, 6)]
+                 if training else ["test_batch.bin"])
+        rec, label_off = 3073, 0
+    else:
+        names = ["train.bin"] if training else ["test.bin"]
+        rec, label_off = 3074, 1  # fine label is the second byte
+    paths = []
+    for n in names:
+        for sub in ("", "cifar-10-batches-bin", "cifar-100-binary"):
+            p = os.path.join(data_dir, sub, n)
+            if os.path.exists(p):
+                paths.append(p)
+                break
+    if len(paths) != len(names):
+        return None
+    xs, ys = [], []
+    for p in paths:
+        raw = np.fromfile(p, dtype=np.uint8)
+        if raw.size % rec:
+            return None
+        raw = raw.reshape(-1, rec)
+        ys.append(raw[:, label_off].astype(np.int64))
+        xs.append(raw[:, rec - 3072:].reshape(-1, 3, 32, 32))
+    x = torch.from_numpy(np.concatenate(xs)).float() / 255.0
+    y = torch.from_numpy(np.concatenate(ys)).long()
+    return x, y
+
+
 class Cifar10Provider(Provider):
-    """CIFAR-10 from pre-downloaded arrays, else synthetic fallback
-    (reference cifar10.py:80-104 offline dummy fallback)."""
+    """CIFAR-10 from pre-downloaded data, else synthetic fallback
+    (reference cifar10.py:80-104 offline dummy fallback). Accepts either
+    the standard binary distribution or a cifar10.npz of
+    {x_train,y_train,x_test,y_test} under ``data_dir`` (also taken from
+    $ADANET_CIFAR_DIR)."""
+
+    _NPZ = "cifar10.npz"
 
     def __init__(self, data_dir: Optional[str] = None, **kwargs):
         super().__init__(n_classes=10, **kwargs)
-        self._arrays = self._try_load(data_dir)
-
-    def _try_load(self, data_dir):
-        if data_dir is None:
-            return None
         import os
+        self._data_dir = data_dir or os.environ.get("ADANET_CIFAR_DIR")
+        self._cache = {}
 
-        import numpy as np
-        path = os.path.join(data_dir, "cifar10.npz")
-        if not os.path.exists(path):
-            return None
-        z = np.load(path)
-        return (torch.from_numpy(z["x_train"]).float().permute(0, 3, 1, 2)
-                / 255.0, torch.from_numpy(z["y_train"]).long().reshape(-1))
+    @property
+    def has_real_data(self) -> bool:
+        return self._load(True) is not None
+
+    def _load(self, training: bool):
+        key = bool(training)
+        if key in self._cache:
+            return self._cache[key]
+        out = None
+        if self._data_dir is not None:
+            import os
+
+            import numpy as np
+            npz = os.path.join(self._data_dir, self._NPZ)
+            if os.path.exists(npz):
+                z = np.load(npz)
+                xk = "x_train" if training else "x_test"
+                yk = "y_train" if training else "y_test"
+                if xk in z:
+                    x = torch.from_numpy(z[xk]).float()
+                    if x.dim() == 4 and x.shape[-1] == 3:  # NHWC -> NCHW
+                        x = x.permute(0, 3, 1, 2).contiguous()
+                    if float(x.max()) > 1.5:
+                        x = x / 255.0
+                    out = (x, torch.from_numpy(z[yk]).long().reshape(-1))
+            if out is None:
+                out = _load_cifar_binaries(self._data_dir, self.n_classes,
+                                           training)
+        self._cache[key] = out
+        return out
 
     def _data(self, training: bool):
-        if self._arrays is not None:
-            return self._arrays
-        fake = FakeImageProvider(n_classes=10, n_examples=512,
+        real = self._load(training)
+        if real is not None:
+            return real
+        fake = FakeImageProvider(n_classes=self.n_classes, n_examples=512,
                                  seed=self.seed)
         return fake._data(training)
 
 
 class Cifar100Provider(Cifar10Provider):
 
+    _NPZ = "cifar100.npz"
+
     def __init__(self, data_dir: Optional[str] = None, **kwargs):
         Provider.__init__(self, n_classes=100, **kwargs)
-        self._arrays = None  # synthetic fallback only without data_dir
-
-    def _data(self, training: bool):
-        fake = FakeImageProvider(n_classes=100, n_examples=512,
-                                 seed=self.seed)
-        return fake._data(training)
+        import os
+        self._data_dir = data_dir or os.environ.get("ADANET_CIFAR_DIR")
+        self._cache = {}
