@@ -217,6 +217,7 @@ __global__ __launch_bounds__(256) void gemm_splitk_epilogue_kernel(
     const int m = (int)(p / N), n = (int)(p % N);
     float v = C32[p] + (bias ? bias[n] : 0.f);
     if (act == 1) v = v > 0.f ? v : 0.f;
+    if (act == 2) v += bf2f(C[(int64_t)m * ldc + n]);
     C[(int64_t)m * ldc + n] = f2bf(v);
   }
 }
@@ -346,6 +347,29 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL(gemm_nt_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
                        ldc, (int)act, seed_ptr, pthresh, inv_keep);
+    HIP_CHECK_KERNEL();
+    return;
+  }
+  // K-dominant tiny-output GEMMs (conv-space pointwise dW: M=N=channels,
+  // K=B*H*W up to ~262k): the tile grid is 1-8 workgroups, so the plain
+  // kernel runs one CU for a ms-scale reduction (measured 49% of the
+  // improve_nas step, profiles/nasprof_summary.txt). Split the K range
+  // over ~512 workgroups with fp32 atomic partials + a tiny epilogue.
+  const int64_t t64 = (int64_t)((M + 63) / 64) * ((N + 63) / 64);
+  if (fast && act != 3 && t64 <= 8 && K >= 16384) {
+    auto C32 = at::zeros({M, N}, A.options().dtype(at::kFloat));
+    int ksplit = (int)std::min<int64_t>(512 / t64, (int64_t)K / 512);
+    if (ksplit < 2) ksplit = 2;
+    const int mt = (M + 63) / 64, nt = (N + 63) / 64;
+    hipLaunchKernelGGL((gemm_nt_bf16_kernel<64, 64, 2, 2, 6, 32, true>),
+                       dim3(mt * nt * ksplit), dim3(THREADS), 0,
+                       stream.stream(), a, b, c, nullptr, M, N, K, lda, ldb,
+                       ldc, (int)act, mt, nt, C32.data_ptr<float>(), ksplit);
+    const int64_t tot = (int64_t)M * N;
+    hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,
+                       dim3((int)std::min<int64_t>((tot + 255) / 256, 2048)),
+                       dim3(256), 0, stream.stream(), C32.data_ptr<float>(),
+                       bias_ptr, c, M, N, ldc, (int)act);
     HIP_CHECK_KERNEL();
     return;
   }
