@@ -107,6 +107,58 @@ __global__ __launch_bounds__(256) void depthwise_bwd_dx_kernel(
 }
 
 // dW: one block per (channel, tap); block-parallel reduction over b,oh,ow.
+// Single-pass dW: each thread reads dy ONCE per output point and
+// accumulates all KS*KS taps in registers (neighboring x reads hit L2),
+// instead of the KS^2 full re-reads of the tap-per-block kernel
+// (measured 14.5% of the improve_nas step at 0.7 TB/s,
+// profiles/nasprof_summary.txt). grid = (C, chunks); one atomicAdd per
+// (block, tap).
+template <int KS>
+__global__ __launch_bounds__(256) void depthwise_bwd_dw_fused_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
+    float* __restrict__ dw, int B, int C, int H, int W, int OH, int OW,
+    int stride, int pad, int64_t chunk) {
+  const int c = blockIdx.x;
+  const int64_t total = (int64_t)B * OH * OW;
+  const int64_t start = (int64_t)blockIdx.y * chunk;
+  const int64_t end = min(total, start + chunk);
+  float acc[KS * KS];
+#pragma unroll
+  for (int t = 0; t < KS * KS; ++t) acc[t] = 0.f;
+  for (int64_t p = start + threadIdx.x; p < end; p += blockDim.x) {
+    const int ow = (int)(p % OW);
+    const int oh = (int)((p / OW) % OH);
+    const int b = (int)(p / ((int64_t)OW * OH));
+    const float dyv =
+        bf2f(dy[((int64_t)b * C + c) * OH * OW + oh * OW + ow]);
+    const bf16_t* xp = x + ((int64_t)b * C + c) * H * W;
+    const int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
+#pragma unroll
+    for (int kh = 0; kh < KS; ++kh) {
+      const int ih = ih0 + kh;
+      if (ih < 0 || ih >= H) continue;
+#pragma unroll
+      for (int kw = 0; kw < KS; ++kw) {
+        const int iw = iw0 + kw;
+        if (iw < 0 || iw >= W) continue;
+        acc[kh * KS + kw] += dyv * bf2f(xp[ih * W + iw]);
+      }
+    }
+  }
+  __shared__ float partial[4];
+#pragma unroll
+  for (int t = 0; t < KS * KS; ++t) {
+    float v = wave_reduce_sum(acc[t]);
+    if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = v;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      atomicAdd(&dw[c * KS * KS + t],
+                partial[0] + partial[1] + partial[2] + partial[3]);
+    }
+    __syncthreads();
+  }
+}
+
 __global__ __launch_bounds__(256) void depthwise_bwd_dw_kernel(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
     float* __restrict__ dw, int B, int C, int H, int W, int OH, int OW,
@@ -193,6 +245,27 @@ void depthwise_bwd_dw(const at::Tensor& x, const at::Tensor& dy,
   const int KS = (int)(std::lround(std::sqrt((double)(dw.numel() / C))));
   TORCH_CHECK(dw.scalar_type() == at::kFloat, "depthwise dW: fp32 out");
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int64_t total = (int64_t)B * OH * OW;
+  // single-pass fused kernel for the NASNet tap sizes; dw must be zeroed
+  // by the caller contract (it is freshly allocated in ops/conv.py).
+  if (KS == 3 || KS == 5 || KS == 7) {
+    const int nchunks =
+        (int)std::max<int64_t>(1, std::min<int64_t>(1024 / C, 64));
+    const int64_t chunk = (total + nchunks - 1) / nchunks;
+    dw.zero_();
+#define LAUNCH_DWF(KSV)                                                       \
+    hipLaunchKernelGGL((depthwise_bwd_dw_fused_kernel<KSV>),                  \
+                       dim3((unsigned)C, (unsigned)nchunks), dim3(256), 0,    \
+                       stream.stream(), (const bf16_t*)x.data_ptr(),          \
+                       (const bf16_t*)dy.data_ptr(), dw.data_ptr<float>(),    \
+                       B, C, H, W, OH, OW, (int)stride, (int)pad, chunk)
+    if (KS == 3) LAUNCH_DWF(3);
+    else if (KS == 5) LAUNCH_DWF(5);
+    else LAUNCH_DWF(7);
+#undef LAUNCH_DWF
+    HIP_CHECK_KERNEL();
+    return;
+  }
   hipLaunchKernelGGL(depthwise_bwd_dw_kernel,
                      dim3((unsigned)C, (unsigned)(KS * KS)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)x.data_ptr(),
