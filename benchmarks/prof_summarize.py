@@ -22,7 +22,7 @@ tot = sum(r[2] for r in rows)
 with open(out_path, "w") as f:
     f.write(f"total GPU time: {tot/1e9:.2f} s over {sum(r[1] for r in rows)} dispatches\n")
     for name, n, dur in rows[:40]:
-        short = re.sub(r'\(.*', '', name)[:84]
+        short = re.sub(r'\(anonymous namespace\)::', '', name); short = re.sub(r'\(.*', '', short)[:84]
         f.write(f"{dur/tot*100:5.1f}%  {dur/1e6:9.1f} ms  n={n:7d}  {short}\n")
     miopen = [(name, n, dur) for name, n, dur in rows
               if 'miopen' in name.lower() or 'Miopen' in name]
