@@ -92,6 +92,10 @@ void depthwise_bwd_dw(const at::Tensor& x, const at::Tensor& dy,
                       at::Tensor& dw, int64_t stride, int64_t pad);
 void argmax_correct(const at::Tensor& logits, const at::Tensor& labels,
                     const c10::optional<at::Tensor>& pred, at::Tensor& correct);
+void im2col_bf16(const at::Tensor& x, at::Tensor& out, int64_t K,
+                 int64_t stride, int64_t pad);
+void col2im_bf16(const at::Tensor& du, at::Tensor& dx, int64_t K,
+                 int64_t stride, int64_t pad);
 void pool3_fwd(const at::Tensor& x, at::Tensor& y,
                const c10::optional<at::Tensor>& argmax, int64_t stride,
                int64_t is_max);
@@ -151,6 +155,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("accum") = 0);
   m.def("argmax_correct", &argmax_correct);
   m.def("binary_histogram", &binary_histogram);
+  m.def("im2col_bf16", &im2col_bf16);
+  m.def("col2im_bf16", &col2im_bf16);
   m.def("pool3_fwd", &pool3_fwd);
   m.def("pool3_bwd", &pool3_bwd);
   m.def("batchnorm_stats", &batchnorm_stats);

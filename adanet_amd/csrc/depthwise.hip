@@ -73,6 +73,46 @@ __global__ __launch_bounds__(256) void depthwise_fwd_tmpl_kernel(
   }
 }
 
+// dX templated fast path (KS, STRIDE compile-time): one (batch, channel)
+// image per blockIdx.y so the taps load once into registers and the loops
+// fully unroll — the runtime-KS kernel below was 13.6% of the improve_nas
+// step at 0.3 TB/s (branchy runtime tap loop, per-element weight reloads).
+template <int KS, int STRIDE>
+__global__ __launch_bounds__(256) void depthwise_bwd_dx_tmpl_kernel(
+    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ w,
+    bf16_t* __restrict__ dx, int C, int H, int W, int OH, int OW, int pad) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const bf16_t* dyp = dy + (int64_t)bc * OH * OW;
+  bf16_t* dxp = dx + (int64_t)bc * H * W;
+  float wr[KS * KS];
+#pragma unroll
+  for (int i = 0; i < KS * KS; ++i) wr[i] = bf2f(w[c * KS * KS + i]);
+  const int total = H * W;
+  for (int p = blockIdx.x * blockDim.x + threadIdx.x; p < total;
+       p += gridDim.x * blockDim.x) {
+    const int iw = p % W;
+    const int ih = p / W;
+    float acc = 0.f;
+#pragma unroll
+    for (int kh = 0; kh < KS; ++kh) {
+      const int num_h = ih + pad - kh;
+      if (num_h < 0 || (STRIDE > 1 && (num_h % STRIDE))) continue;
+      const int oh = num_h / STRIDE;
+      if (oh >= OH) continue;
+#pragma unroll
+      for (int kw = 0; kw < KS; ++kw) {
+        const int num_w = iw + pad - kw;
+        if (num_w < 0 || (STRIDE > 1 && (num_w % STRIDE))) continue;
+        const int ow = num_w / STRIDE;
+        if (ow >= OW) continue;
+        acc += wr[kh * KS + kw] * bf2f(dyp[oh * OW + ow]);
+      }
+    }
+    dxp[p] = f2bf(acc);
+  }
+}
+
 // dX: full correlation with the flipped kernel, honoring stride divisibility.
 __global__ __launch_bounds__(256) void depthwise_bwd_dx_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ w,
@@ -229,6 +269,31 @@ void depthwise_bwd_dx(const at::Tensor& dy, const at::Tensor& w,
   const int OH = (int)dy.size(2), OW = (int)dy.size(3);
   const int KS = (int)w.size(w.dim() - 1);
   auto stream = at::cuda::getCurrentCUDAStream();
+  const int64_t bc = (int64_t)B * C;
+  if ((KS == 3 || KS == 5 || KS == 7) && (stride == 1 || stride == 2) &&
+      bc <= 65535) {
+    const int px_blocks = std::max(1, std::min((H * W + 255) / 256,
+                                               (int)(2048 / bc) + 1));
+    using kern_t = void (*)(const bf16_t*, const bf16_t*, bf16_t*, int, int,
+                            int, int, int, int);
+    kern_t kern;
+    if (stride == 1) {
+      kern = KS == 3 ? depthwise_bwd_dx_tmpl_kernel<3, 1>
+             : KS == 5 ? depthwise_bwd_dx_tmpl_kernel<5, 1>
+                       : depthwise_bwd_dx_tmpl_kernel<7, 1>;
+    } else {
+      kern = KS == 3 ? depthwise_bwd_dx_tmpl_kernel<3, 2>
+             : KS == 5 ? depthwise_bwd_dx_tmpl_kernel<5, 2>
+                       : depthwise_bwd_dx_tmpl_kernel<7, 2>;
+    }
+    hipLaunchKernelGGL(kern, dim3((unsigned)px_blocks, (unsigned)bc),
+                       dim3(256), 0, stream.stream(),
+                       (const bf16_t*)dy.data_ptr(),
+                       (const bf16_t*)w.data_ptr(), (bf16_t*)dx.data_ptr(),
+                       C, H, W, OH, OW, (int)pad);
+    HIP_CHECK_KERNEL();
+    return;
+  }
   hipLaunchKernelGGL(depthwise_bwd_dx_kernel,
                      dim3(dw_grid((int64_t)B * C * H * W)), dim3(256), 0,
                      stream.stream(), (const bf16_t*)dy.data_ptr(),

@@ -198,6 +198,29 @@ class HipDepthwiseConv2d(nn.Module):
             self.channels, self.kernel_size, self.stride, self.padding)
 
 
+class _Im2colFn(torch.autograd.Function):
+    """Batched im2col (one launch for the whole batch, zero-padded K dim)
+    with the exact col2im adjoint for backward (csrc/im2col.hip)."""
+
+    @staticmethod
+    def forward(ctx, x, K, stride, pad, ckk_pad, oh, ow):
+        ext = _extension.require()
+        B = x.shape[0]
+        out = torch.empty((B, ckk_pad, oh, ow), device=x.device,
+                          dtype=x.dtype)
+        ext.im2col_bf16(x, out, K, stride, pad)
+        ctx.meta = (x.shape, K, stride, pad)
+        return out
+
+    @staticmethod
+    def backward(ctx, du):
+        ext = _extension.require()
+        shape, K, stride, pad = ctx.meta
+        dx = torch.empty(shape, device=du.device, dtype=du.dtype)
+        ext.col2im_bf16(du.contiguous(), dx, K, stride, pad)
+        return dx, None, None, None, None, None, None
+
+
 class HipConvNxN(nn.Module):
     """Dense KxK conv as unfold + the batched MFMA GEMM (K9 tail).
 
@@ -237,16 +260,14 @@ class HipConvNxN(nn.Module):
             B = x.shape[0]
             oh = (x.shape[2] + 2 * self.padding - K) // self.stride + 1
             ow = (x.shape[3] + 2 * self.padding - K) // self.stride + 1
-            u = torch.nn.functional.unfold(
-                x, K, padding=self.padding, stride=self.stride)  # [B,ckk,L]
-            if ckk_pad != ckk:
-                u = torch.nn.functional.pad(u, (0, 0, 0, ckk_pad - ckk))
+            # batched im2col with the 32-alignment zero-pad folded in
+            # (torch's F.unfold is one im2col launch PER IMAGE).
+            u = _Im2colFn.apply(x.contiguous(), K, self.stride,
+                                self.padding, ckk_pad, oh, ow)
             w2 = self.weight.reshape(self.out_channels, ckk)
             if ckk_pad != ckk:
                 w2 = torch.nn.functional.pad(w2, (0, ckk_pad - ckk))
-            y = _Conv1x1Fn.apply(
-                u.reshape(B, ckk_pad, oh, ow).contiguous(), w2.contiguous(),
-                self.bias)
+            y = _Conv1x1Fn.apply(u, w2.contiguous(), self.bias)
             return y
         return torch.nn.functional.conv2d(
             x, self.weight.to(x.dtype),

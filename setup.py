@@ -25,6 +25,7 @@ SRC = [
     "adanet_amd/csrc/layernorm.hip",
     "adanet_amd/csrc/batchnorm.hip",
     "adanet_amd/csrc/pool.hip",
+    "adanet_amd/csrc/im2col.hip",
     "adanet_amd/csrc/elementwise.hip",
     "adanet_amd/csrc/reduce.hip",
     "adanet_amd/csrc/depthwise.hip",

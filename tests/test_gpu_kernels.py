@@ -793,3 +793,54 @@ def test_fused_dropout_fresh_mask_per_step_and_eval_off():
     y3 = lin(x)
     ref = torch.relu(x.float() @ lin.weight.float().t() + lin.bias.float())
     assert (y3.float() - ref).abs().max().item() < 0.1
+
+
+# ------------------------------------------------------- batched im2col
+@pytest.mark.parametrize("K,stride,pad,cin", [(3, 1, 1, 7), (3, 2, 1, 16),
+                                              (1, 2, 0, 9), (5, 1, 2, 4)])
+def test_im2col_col2im_vs_unfold(K, stride, pad, cin):
+    ext = _ext()
+    torch.manual_seed(41)
+    B, H, W = 5, 13, 11
+    x = torch.randn(B, cin, H, W, device=DEV).to(torch.bfloat16)
+    oh = (H + 2 * pad - K) // stride + 1
+    ow = (W + 2 * pad - K) // stride + 1
+    ckk = cin * K * K
+    ckk_pad = (ckk + 31) // 32 * 32
+    out = torch.empty((B, ckk_pad, oh, ow), device=DEV, dtype=torch.bfloat16)
+    ext.im2col_bf16(x, out, K, stride, pad)
+    ref = torch.nn.functional.unfold(x.float(), K, padding=pad,
+                                     stride=stride)
+    got = out.float().reshape(B, ckk_pad, -1)
+    assert torch.equal(got[:, :ckk], ref.to(torch.bfloat16).float())
+    assert (got[:, ckk:] == 0).all()
+    # col2im == unfold's adjoint (fold)
+    du = torch.randn(B, ckk_pad, oh, ow, device=DEV).to(torch.bfloat16)
+    dx = torch.empty_like(x)
+    ext.col2im_bf16(du, dx, K, stride, pad)
+    ref_dx = torch.nn.functional.fold(
+        du.float().reshape(B, ckk_pad, -1)[:, :ckk], (H, W), K,
+        padding=pad, stride=stride)
+    assert (dx.float() - ref_dx).abs().max().item() < 0.15
+
+
+def test_conv_nxn_fwd_bwd_vs_torch():
+    from adanet_amd.ops.conv import HipConvNxN
+    torch.manual_seed(42)
+    conv = HipConvNxN(16, 32, 3, stride=1, padding=1, bias=True).to(DEV)
+    conv.weight.data = conv.weight.data.to(torch.bfloat16)
+    x = torch.randn(4, 16, 16, 16, device=DEV)
+    xb = x.to(torch.bfloat16).requires_grad_(True)
+    y = conv(xb)
+    ref = torch.nn.functional.conv2d(
+        x, conv.weight.float(), conv.bias.float(), 1, 1)
+    assert (y.float() - ref).abs().max().item() < 0.2
+    g = torch.randn_like(y)
+    y.backward(g)
+    xf = x.clone().requires_grad_(True)
+    wf = conv.weight.detach().float().requires_grad_(True)
+    torch.nn.functional.conv2d(xf, wf, conv.bias.detach().float(), 1,
+                               1).backward(g.float())
+    assert (xb.grad.float() - xf.grad).abs().max().item() < 0.2
+    assert (conv.weight.grad.float() - wf.grad).abs().max().item() / (
+        wf.grad.abs().max().item() + 1e-3) < 0.05
