@@ -35,6 +35,66 @@ __device__ __forceinline__ float block_reduce(float v, float* scratch,
   return out;
 }
 
+// Chunked partial sums: grid (C, nchunks) so small-C shapes still fill
+// the chip (a C-block grid was 30% of the improve_nas step on 32 CUs,
+// profiles/nasprof3_summary.txt); one atomicAdd pair per block.
+__global__ __launch_bounds__(256) void bn_stats_partial_kernel(
+    const bf16_t* __restrict__ x, float* __restrict__ ssum,
+    float* __restrict__ s2sum, int N, int C, int64_t HW, int nchunks) {
+  __shared__ float scratch[4];
+  const int c = blockIdx.x;
+  const int n0 = (N * blockIdx.y) / nchunks;
+  const int n1 = (N * (blockIdx.y + 1)) / nchunks;
+  const int tid = threadIdx.x;
+  float s = 0.f, s2 = 0.f;
+  for (int n = n0; n < n1; ++n) {
+    const bf16_t* base = x + ((int64_t)n * C + c) * HW;
+    if ((HW & 7) == 0) {
+      for (int64_t i = (int64_t)tid * 8; i < HW; i += 256 * 8) {
+        const bn8 v = *(const bn8*)(base + i);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const float f = bf2f(v.v[k]);
+          s += f;
+          s2 += f * f;
+        }
+      }
+    } else {
+      for (int64_t i = tid; i < HW; i += 256) {
+        const float f = bf2f(base[i]);
+        s += f;
+        s2 += f * f;
+      }
+    }
+  }
+  s = block_reduce(s, scratch, tid);
+  s2 = block_reduce(s2, scratch, tid);
+  if (tid == 0) {
+    atomicAdd(&ssum[c], s);
+    atomicAdd(&s2sum[c], s2);
+  }
+}
+
+__global__ __launch_bounds__(256) void bn_stats_finalize_kernel(
+    const float* __restrict__ ssum, const float* __restrict__ s2sum,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    int N, int C, int64_t HW, float eps, float momentum) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const double M = (double)N * (double)HW;
+  const float mean = (float)(ssum[c] / M);
+  float var = (float)(s2sum[c] / M) - mean * mean;
+  var = var > 0.f ? var : 0.f;
+  mean_out[c] = mean;
+  rstd_out[c] = rsqrtf(var + eps);
+  if (running_mean) {
+    running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+    const float unbiased = M > 1.0 ? var * (float)(M / (M - 1.0)) : var;
+    running_var[c] = (1.f - momentum) * running_var[c] + momentum * unbiased;
+  }
+}
+
 // Per-channel mean/rstd (+ running-stat update). Block c handles channel c.
 __global__ __launch_bounds__(256) void bn_stats_kernel(
     const bf16_t* __restrict__ x, float* __restrict__ mean_out,
@@ -95,6 +155,49 @@ __global__ __launch_bounds__(256) void bn_norm_kernel(
     const float g = gamma ? gamma[c] : 1.f;
     const float b = beta ? beta[c] : 0.f;
     y[p] = f2bf((bf2f(x[p]) - mean[c]) * rstd[c] * g + b);
+  }
+}
+
+// Chunked/vectorized variant of the backward reduce: grid (C, nchunks),
+// atomicAdd per block (sdy/sdyx zeroed by the host).
+__global__ __launch_bounds__(256) void bn_bwd_reduce_chunked_kernel(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
+    const float* __restrict__ mean, const float* __restrict__ rstd,
+    float* __restrict__ sdy, float* __restrict__ sdyx, int N, int C,
+    int64_t HW, int nchunks) {
+  __shared__ float scratch[4];
+  const int c = blockIdx.x;
+  const int n0 = (N * blockIdx.y) / nchunks;
+  const int n1 = (N * (blockIdx.y + 1)) / nchunks;
+  const int tid = threadIdx.x;
+  const float m = mean[c], r = rstd[c];
+  float a = 0.f, b = 0.f;
+  for (int n = n0; n < n1; ++n) {
+    const int64_t off = ((int64_t)n * C + c) * HW;
+    if ((HW & 7) == 0) {
+      for (int64_t i = (int64_t)tid * 8; i < HW; i += 256 * 8) {
+        const bn8 dv = *(const bn8*)(dy + off + i);
+        const bn8 xv = *(const bn8*)(x + off + i);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const float d = bf2f(dv.v[k]);
+          a += d;
+          b += d * (bf2f(xv.v[k]) - m) * r;
+        }
+      }
+    } else {
+      for (int64_t i = tid; i < HW; i += 256) {
+        const float d = bf2f(dy[off + i]);
+        a += d;
+        b += d * (bf2f(x[off + i]) - m) * r;
+      }
+    }
+  }
+  a = block_reduce(a, scratch, tid);
+  b = block_reduce(b, scratch, tid);
+  if (tid == 0) {
+    atomicAdd(&sdy[c], a);
+    atomicAdd(&sdyx[c], b);
   }
 }
 
@@ -165,6 +268,21 @@ void batchnorm_stats(const at::Tensor& x, at::Tensor& mean, at::Tensor& rstd,
     rm = running_mean->data_ptr<float>();
     rv = running_var->data_ptr<float>();
   }
+  const int nchunks = (int)std::max<int64_t>(
+      1, std::min<int64_t>(2048 / C, std::min<int64_t>(N, 64)));
+  if (nchunks > 1) {
+    auto sums = at::zeros({2, C}, x.options().dtype(at::kFloat));
+    float* ssum = sums.data_ptr<float>();
+    hipLaunchKernelGGL(bn_stats_partial_kernel, dim3(C, nchunks), dim3(256),
+                       0, stream.stream(), (const bf16_t*)x.data_ptr(),
+                       ssum, ssum + C, N, C, HW, nchunks);
+    hipLaunchKernelGGL(bn_stats_finalize_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream.stream(), ssum, ssum + C,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), rm,
+                       rv, N, C, HW, (float)eps, (float)momentum);
+    HIP_CHECK_KERNEL();
+    return;
+  }
   hipLaunchKernelGGL(bn_stats_kernel, dim3(C), dim3(256), 0, stream.stream(),
                      (const bf16_t*)x.data_ptr(), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(), rm, rv, N, C, HW, (float)eps,
@@ -199,11 +317,24 @@ void batchnorm_bwd(const at::Tensor& x, const at::Tensor& dy, at::Tensor& dx,
   const int64_t HW = x.size(2) * x.size(3);
   const int64_t total = x.numel();
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0,
-                     stream.stream(), (const bf16_t*)x.data_ptr(),
-                     (const bf16_t*)dy.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), sdy.data_ptr<float>(),
-                     sdyx.data_ptr<float>(), N, C, HW);
+  const int nchunks = (int)std::max<int64_t>(
+      1, std::min<int64_t>(2048 / C, std::min<int64_t>(N, 64)));
+  if (nchunks > 1) {
+    sdy.zero_();
+    sdyx.zero_();
+    hipLaunchKernelGGL(bn_bwd_reduce_chunked_kernel, dim3(C, nchunks),
+                       dim3(256), 0, stream.stream(),
+                       (const bf16_t*)x.data_ptr(),
+                       (const bf16_t*)dy.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), sdy.data_ptr<float>(),
+                       sdyx.data_ptr<float>(), N, C, HW, nchunks);
+  } else {
+    hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0,
+                       stream.stream(), (const bf16_t*)x.data_ptr(),
+                       (const bf16_t*)dy.data_ptr(), mean.data_ptr<float>(),
+                       rstd.data_ptr<float>(), sdy.data_ptr<float>(),
+                       sdyx.data_ptr<float>(), N, C, HW);
+  }
   HIP_CHECK_KERNEL();
   const float invM = 1.f / (float)((double)N * (double)HW);
   hipLaunchKernelGGL(
