@@ -79,35 +79,6 @@ __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
   atomicAdd(&db[c], acc);
 }
 
-// 4-column-per-thread variant (C % 4 == 0): short4 loads/stores bring the
-// per-wave transaction to 512 B (the scalar kernel moved 2 B/lane —
-// measured 1.4 TB/s on the bench shapes, G13).
-__global__ __launch_bounds__(256) void relu_bwd_colsum_v4_kernel(
-    const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
-    bf16_t* __restrict__ dz, float* __restrict__ db, int B, int C,
-    int rows_per_block, float scale) {
-  const int c4 = blockIdx.x * blockDim.x + threadIdx.x;  // column group
-  if (c4 * 4 >= C) return;
-  const int r0 = blockIdx.y * rows_per_block;
-  const int r1 = min(B, r0 + rows_per_block);
-  float acc[4] = {0.f, 0.f, 0.f, 0.f};
-  for (int b = r0; b < r1; ++b) {
-    const int64_t i = (int64_t)b * C + c4 * 4;
-    const s16x4 yv = *(const s16x4*)&y[i];
-    const s16x4 dv = *(const s16x4*)&dy[i];
-    s16x4 out;
-#pragma unroll
-    for (int k = 0; k < 4; ++k) {
-      const float g = bits2f(yv[k]) > 0.f ? bits2f(dv[k]) * scale : 0.f;
-      out[k] = f2bits(g);
-      acc[k] += g;
-    }
-    *(s16x4*)&dz[i] = out;
-  }
-#pragma unroll
-  for (int k = 0; k < 4; ++k) atomicAdd(&db[c4 * 4 + k], acc[k]);
-}
-
 // Batched device-to-device copy: up to 8 (src,dst) pairs per launch
 // (frozen-logit cache -> static graph-input buffers each step was J-1
 // separate ~7.6 us copyBuffer launches; one kernel replaces them).
@@ -219,27 +190,6 @@ void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
   const int B = (int)dy.size(0), C = (int)dy.size(1);
   if (B == 0 || C == 0) return;
   auto stream = at::cuda::getCurrentCUDAStream();
-  // r01 measured an 8-col variant REGRESSING end-to-end; this 4-col one
-  // must survive its own same-box A/B (ADANET_COLSUM_SCALAR=1 reverts).
-  static const bool force_scalar = []() {
-    const char* e = getenv("ADANET_COLSUM_SCALAR");
-    return e && e[0] && e[0] != '0';
-  }();
-  if (C % 4 == 0 && !force_scalar) {
-    const int stripes = (C / 4 + 255) / 256;
-    const int row_chunks =
-        std::max(1, std::min(2048 / stripes, (B + 7) / 8));
-    const int rows_per_block = (B + row_chunks - 1) / row_chunks;
-    hipLaunchKernelGGL(relu_bwd_colsum_v4_kernel,
-                       dim3((unsigned)stripes, (unsigned)row_chunks),
-                       dim3(256), 0, stream.stream(),
-                       (const bf16_t*)dy.data_ptr(),
-                       (const bf16_t*)y.data_ptr(), (bf16_t*)dz.data_ptr(),
-                       db.data_ptr<float>(), B, C, rows_per_block,
-                       (float)scale);
-    HIP_CHECK_KERNEL();
-    return;
-  }
   const int stripes = (C + 255) / 256;
   const int row_chunks =
       std::max(1, std::min(1024 / stripes, (B + 7) / 8));
