@@ -90,7 +90,16 @@ __device__ __forceinline__ s16x8 lds_frag(const bf16_t* lds, int row,
 // constexpr values; dispatch over the small set of counts this kernel uses).
 template <int N>
 __device__ __forceinline__ void vmcnt_wait() {
-  static_assert(N >= 0 && N <= 4, "unsupported vmcnt");
+  static_assert((N >= 0 && N <= 4) || N == 6 || N == 8,
+                "unsupported vmcnt");
+  if constexpr (N == 6) {
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    return;
+  }
+  if constexpr (N == 8) {
+    asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    return;
+  }
   if constexpr (N == 0) {
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   } else if constexpr (N == 1) {
@@ -443,6 +452,197 @@ __global__ __launch_bounds__(512, 2) void gemm_nt8p_kernel(
     if (SETPRIO) __builtin_amdgcn_s_setprio(0);
     S_BARRIER();
   }
+
+  const int c_col_in_frag = lane & 15;
+  const int c_row_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int j = 0; j < FN; ++j) {
+      const int col = col0 + wn * (FN * 16) + j * 16 + c_col_in_frag;
+      if (col >= N) continue;
+      const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
+        if (row >= M) continue;
+        float v = acc[i][j][rr] + bv;
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        bf16_t* cp = &C[(int64_t)row * ldc + col];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// K-slab staged kernel: the deepest pipeline variant. BK=64 is staged as
+// TWO [rows x 32] column slabs per operand; phases {0,1} compute k-slab 0,
+// {2,3} k-slab 1, so consumption is K-progressive and each staged unit
+// has 5-6 phases to land (the half-tile schedule above gives A(t+1) only
+// 2 phases — its vmcnt stalls when loaded-latency HBM exceeds that):
+//
+//   issue at tile t:  ph0: A-slab1(t+1)   ph1: B-slab1(t+1)
+//                     ph2: A-slab0(t+2)   ph3: B-slab0(t+2)
+//   (slab-s LDS homes are free one whole tile early: slab0 reads end at
+//    ph1, so slab0(t+2) can be issued at t ph2.)
+//   waits: vmcnt(4 units) at ph1-end (guards slab1(t) reads at ph2) and
+//   ph3-end (guards slab0(t+1) reads at t+1 ph0) — 4 units stay in
+//   flight across EVERY barrier.
+//
+// Slab rows are 64 B; fragment ds_read_b128 at (row, 16B-block q=lane>>4)
+// would be 8-way bank-conflicted; the involution blk ^= (row>>1)&3
+// spreads the wave across all 32 banks at 2-way (free). Stage-side the
+// same XOR is applied to the per-lane GLOBAL source block
+// ((lane&3) ^ ((lane>>3)&3) — linear LDS dest as global_load_lds needs).
+// ---------------------------------------------------------------------
+
+// Stage one [ROWS x 32] K-slab (rows of 64 B; NW*16 rows per block round).
+template <int ROWS, int NW>
+__device__ __forceinline__ void stage_slab(
+    const bf16_t* __restrict__ G, int64_t ld, int rows0, int max_row, int k0,
+    bf16_t* __restrict__ lds_slab, int wid, int lane) {
+#pragma unroll
+  for (int r = 0; r < ROWS / (NW * 16); ++r) {
+    const int row_in = r * (NW * 16) + wid * 16 + (lane >> 2);
+    int grow = rows0 + row_in;
+    grow = grow < max_row ? grow : max_row - 1;
+    const int src_blk = (lane & 3) ^ ((lane >> 3) & 3);
+    const bf16_t* gp = G + (int64_t)grow * ld + k0 + src_blk * 8;
+    bf16_t* lp = lds_slab + (r * (NW * 16) + wid * 16) * 32;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+__device__ __forceinline__ s16x8 lds_frag_ks(const bf16_t* lds_slab,
+                                             int row, int q) {
+  const int blk = q ^ ((row >> 1) & 3);
+  const uint32_t addr = (uint32_t)(uintptr_t)(
+      const __attribute__((address_space(3))) void*)
+      &lds_slab[row * 32 + blk * 8];
+  s16x8 out;
+  asm volatile("ds_read_b128 %0, %1" : "=v"(out) : "v"(addr));
+  return out;
+}
+
+template <int BM, int BN, int WN>
+__global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
+    int ntiles) {
+  constexpr int NW = 2 * WN;
+  constexpr int FM = BM / 32;          // m-fragments per wave (2 M-waves)
+  constexpr int FN = BN / (WN * 16);
+  constexpr int MH = FM / 2;           // m-reps per phase
+  static_assert(FM >= 2, "need >=1 m-rep per phase pair");
+  constexpr int LPU_A = BM / (NW * 16);  // loads/lane per A slab
+  constexpr int LPU_B = BN / (NW * 16);
+  constexpr int INFLIGHT = 2 * LPU_A + 2 * LPU_B;  // 4 units
+  __shared__ bf16_t As[2][2][BM * 32];  // [dbuf][slab]
+  __shared__ bf16_t Bs[2][2][BN * 32];
+
+  const int nwg = mtiles * ntiles;
+  const int orig = blockIdx.x;
+  const int q = nwg >> 3, r = nwg & 7, xcd = orig & 7;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) +
+                 (orig >> 3);
+  const int tile_m = wg / ntiles, tile_n = wg % ntiles;
+  const int row0 = tile_m * BM, col0 = tile_n * BN;
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6, lane = tid & 63;
+  const int wm = wid / WN, wn = wid % WN;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int NT = K / 64;
+
+#define KS_STAGE_A(buf, slab, t_)                                             \
+  stage_slab<BM, NW>(A, lda, row0, M, (t_)*64 + (slab)*32,                    \
+                     &As[buf][slab][0], wid, lane)
+#define KS_STAGE_B(buf, slab, t_)                                             \
+  stage_slab<BN, NW>(B, ldb, col0, N, (t_)*64 + (slab)*32,                    \
+                     &Bs[buf][slab][0], wid, lane)
+
+  // prologue: tile0 both slabs + tile1 slab0; leave 4 units in flight.
+  KS_STAGE_A(0, 0, 0);
+  KS_STAGE_B(0, 0, 0);
+  KS_STAGE_A(0, 1, 0);
+  KS_STAGE_B(0, 1, 0);
+  if (NT > 1) {
+    KS_STAGE_A(1, 0, 1);
+    KS_STAGE_B(1, 0, 1);
+    vmcnt_wait<INFLIGHT>();
+  } else {
+    vmcnt_wait<0>();
+  }
+  S_BARRIER();
+
+  const int arow_base = wm * (FM * 16) + (lane & 15);
+  const int brow_base = wn * (FN * 16) + (lane & 15);
+  const int kq = lane >> 4;
+
+  for (int t = 0; t < NT; ++t) {
+    const int buf = t & 1;
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const bf16_t* at = &As[buf][kk][0];
+      const bf16_t* bt = &Bs[buf][kk][0];
+      s16x8 bfrag[FN];
+#pragma unroll
+      for (int ph = 0; ph < 2; ++ph) {  // global phase p = kk*2 + ph
+        s16x8 afrag[MH];
+        if (ph == 0) {
+#pragma unroll
+          for (int n = 0; n < FN; ++n)
+            bfrag[n] = lds_frag_ks(bt, brow_base + n * 16, kq);
+        }
+#pragma unroll
+        for (int g = 0; g < MH; ++g)
+          afrag[g] =
+              lds_frag_ks(at, arow_base + (ph * MH + g) * 16, kq);
+        // prefetch issue: p0: A-s1(t+1), p1: B-s1(t+1),
+        //                 p2: A-s0(t+2), p3: B-s0(t+2)
+        if (kk == 0) {
+          if (ph == 0) {
+            if (t + 1 < NT) KS_STAGE_A(buf ^ 1, 1, t + 1);
+          } else {
+            if (t + 1 < NT) KS_STAGE_B(buf ^ 1, 1, t + 1);
+          }
+        } else {
+          if (ph == 0) {
+            if (t + 2 < NT) KS_STAGE_A(buf, 0, t + 2);
+          } else {
+            if (t + 2 < NT) KS_STAGE_B(buf, 0, t + 2);
+          }
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+        for (int g = 0; g < MH; ++g)
+#pragma unroll
+          for (int n = 0; n < FN; ++n)
+            acc[ph * MH + g][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[g], bfrag[n], acc[ph * MH + g][n], 0, 0, 0);
+        if (ph == 1) {
+          // end of the k-slab pair: publish the next consumers' data
+          // (kk0: slab1(t) reads at p2; kk1: slab0(t+1) reads at t+1 p0)
+          vmcnt_wait<INFLIGHT>();
+        }
+        S_BARRIER();
+      }
+    }
+  }
+#undef KS_STAGE_A
+#undef KS_STAGE_B
 
   const int c_col_in_frag = lane & 15;
   const int c_row_base = (lane >> 4) * 4;
@@ -930,6 +1130,18 @@ void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                        dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M,   \
                        N, K, lda, ldb, ldc, (int)act, mt, nt);                \
   } while (0)
+#define LAUNCH_KS(BM, BN, WNW)                                               \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    hipLaunchKernelGGL((gemm_ks_kernel<BM, BN, WNW>), dim3(mt * nt),          \
+                       dim3(2 * WNW * 64), 0, stream.stream(), a, b, c,       \
+                       bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);   \
+  } while (0)
+    case 50: LAUNCH_KS(256, 256, 4); break;
+    case 51: LAUNCH_KS(128, 128, 4); break;
+    case 52: LAUNCH_KS(128, 128, 2); break;
+    case 53: LAUNCH_KS(256, 128, 4); break;
+#undef LAUNCH_KS
     case 30: LAUNCH_8PHP(256, 256, false); break;
     case 31: LAUNCH_8PHP(256, 256, true); break;
     case 32: LAUNCH_8PHP(256, 128, false); break;

@@ -69,7 +69,7 @@ def main():
     if not args.skip_refcheck:
         fails = 0
         cases = []
-        for variant in (0, 1, 2, 3, 10, 13, 20, 30, 31, 32, 33, 34):
+        for variant in (0, 3, 20, 30, 50, 51, 52, 53):
             for (M, N, K) in ((256, 256, 64), (512, 512, 128),
                               (1024, 768, 256), (512, 512, 64)):
                 cases.append((M, N, K, variant, 0, False))
@@ -88,7 +88,7 @@ def main():
                 fails += 1
                 print("FAIL", M, N, K, "variant", v, "act", act, "rel", rel)
         # multi-run race screen on the steady-state variant (guide m152):
-        for v in (0, 30):
+        for v in (0, 50, 51):
             for rep in range(8):
                 ok, rel = refcheck(4096, 4096, 4096, v, 0, False)
                 results["refcheck"].append(
@@ -118,10 +118,9 @@ def main():
 
         row["old_dispatch"] = tf(bench_fn(
             lambda: ext.gemm_nt_bf16(A, B, C, None, 0), args.iters))
-        for v, name in ((0, "8ph_256x256"), (3, "8ph_128x128"),
-                        (20, "8ph_256_noprio"), (30, "pipe_256x256"),
-                        (31, "pipe_256_prio"), (32, "pipe_256x128"),
-                        (33, "pipe_128x256"), (34, "pipe_128x128")):
+        for v, name in ((3, "8ph_128x128"), (20, "8ph_256_noprio"),
+                        (50, "ks_256x256"), (51, "ks_128x128_8w"),
+                        (52, "ks_128x128_4w"), (53, "ks_256x128")):
             try:
                 row[name] = tf(bench_fn(
                     lambda v=v: ext.gemm_nt8(A, B, C, None, 0, v),
