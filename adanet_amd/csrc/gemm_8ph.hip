@@ -610,18 +610,25 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
           afrag[g] =
               lds_frag_ks(at, arow_base + (ph * MH + g) * 16, kq);
         // prefetch issue: p0: A-s1(t+1), p1: B-s1(t+1),
-        //                 p2: A-s0(t+2), p3: B-s0(t+2)
+        //                 p2: A-s0(t+2), p3: B-s0(t+2).
+        // ALWAYS issued, with the tile index clamped at the tail: skipping
+        // would shrink the outstanding-load count and turn the fixed
+        // vmcnt(INFLIGHT) waits vacuous exactly when the next tile's slab0
+        // is still in flight (caught by the 4096 race screen on the LPU=1
+        // config). Clamped dummy loads land in slots no later tile reads.
         if (kk == 0) {
+          const int tn = t + 1 < NT ? t + 1 : NT - 1;
           if (ph == 0) {
-            if (t + 1 < NT) KS_STAGE_A(buf ^ 1, 1, t + 1);
+            KS_STAGE_A(buf ^ 1, 1, tn);
           } else {
-            if (t + 1 < NT) KS_STAGE_B(buf ^ 1, 1, t + 1);
+            KS_STAGE_B(buf ^ 1, 1, tn);
           }
         } else {
+          const int tn = t + 2 < NT ? t + 2 : NT - 1;
           if (ph == 0) {
-            if (t + 2 < NT) KS_STAGE_A(buf, 0, t + 2);
+            KS_STAGE_A(buf, 0, tn);
           } else {
-            if (t + 2 < NT) KS_STAGE_B(buf, 0, t + 2);
+            KS_STAGE_B(buf, 0, tn);
           }
         }
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
