@@ -1047,10 +1047,12 @@ bool gemm_nt8_try(const bf16_t* a, const bf16_t* b, bf16_t* c,
     return true;
   }
   if (t128 >= 256) {
+    // K-slab schedule: 643-687 TF at the 2048-class bench shapes vs 555
+    // for the half-tile 8-phase (profiles/gemm_ks_r02b.json, same-box).
     const int mt = (M + 127) / 128, nt = (N + 127) / 128;
-    hipLaunchKernelGGL((gemm_nt8_kernel<128, 128, false, false>),
-                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
-                       bias_ptr, M, N, K, lda, ldb, ldc, act, mt, nt);
+    hipLaunchKernelGGL((gemm_ks_kernel<128, 128, 4>), dim3(mt * nt),
+                       dim3(512), 0, stream, a, b, c, bias_ptr, M, N, K,
+                       lda, ldb, ldc, act, mt, nt);
     return true;
   }
   return false;
