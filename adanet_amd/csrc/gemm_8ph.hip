@@ -517,6 +517,54 @@ __device__ __forceinline__ void stage_slab(
   }
 }
 
+// K-major slab staging: [COLS/16][32][16] subtiles, one 1 KiB DMA each
+// (the gemm_tn.hip tr scheme per 32-k slab).
+typedef __attribute__((ext_vector_type(2))) unsigned int u32x2_k;
+
+__device__ __forceinline__ u32x2_k ks_tr_read(const bf16_t* p) {
+  const unsigned off = (unsigned)(uintptr_t)(
+      const __attribute__((address_space(3))) bf16_t*)p;
+  u32x2_k out;
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(out) : "v"(off));
+  return out;
+}
+
+template <int COLS, int NW>
+__device__ __forceinline__ void stage_slab_tr(
+    const bf16_t* __restrict__ G, int64_t ld, int col0, int max_col, int k0,
+    int max_k, bf16_t* __restrict__ lds_slab, int wid, int lane) {
+  constexpr int SUBT = COLS / 16;
+#pragma unroll
+  for (int st = wid; st < SUBT; st += NW) {
+    const int k = k0 + (lane >> 1);
+    int col = col0 + st * 16 + (lane & 1) * 8;
+    if (col + 8 > max_col) col = max(0, (max_col - 8) & ~7);
+    const bf16_t* gp = G + (int64_t)min(k, max_k - 1) * ld + col;
+    bf16_t* lp = lds_slab + st * 512;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)gp,
+        (__attribute__((address_space(3))) void*)lp, 16, 0, 0);
+  }
+}
+
+// Fragment from a K-major slab: two hardware-transpose reads of the
+// [32][16] subtile (measured cooperative semantics, gemm_tn.hip).
+__device__ __forceinline__ s16x8 lds_frag_ks_tr(const bf16_t* lds_slab,
+                                                int colblock, int lane) {
+  const bf16_t* sub = lds_slab + colblock * 512;
+  const int row = 8 * (lane >> 4) + ((lane >> 2) & 3);
+  const int col4 = 4 * (lane & 3);
+  u32x2_k r0 = ks_tr_read(sub + row * 16 + col4);
+  u32x2_k r1 = ks_tr_read(sub + (row + 4) * 16 + col4);
+  union {
+    s16x8 f;
+    struct { u32x2_k lo, hi; } u;
+  } pack;
+  pack.u.lo = r0;
+  pack.u.hi = r1;
+  return pack.f;
+}
+
 __device__ __forceinline__ s16x8 lds_frag_ks(const bf16_t* lds_slab,
                                              int row, int q) {
   const int blk = q ^ ((row >> 1) & 3);
@@ -528,7 +576,7 @@ __device__ __forceinline__ s16x8 lds_frag_ks(const bf16_t* lds_slab,
   return out;
 }
 
-template <int BM, int BN, int WN>
+template <int BM, int BN, int WN, bool TRA = false, bool TRB = false>
 __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -566,11 +614,23 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
   const int NT = K / 64;
 
 #define KS_STAGE_A(buf, slab, t_)                                             \
-  stage_slab<BM, NW>(A, lda, row0, M, (t_)*64 + (slab)*32,                    \
-                     &As[buf][slab][0], wid, lane)
+  do {                                                                        \
+    if (TRA)                                                                  \
+      stage_slab_tr<BM, NW>(A, lda, row0, M, (t_)*64 + (slab)*32, K,          \
+                            &As[buf][slab][0], wid, lane);                    \
+    else                                                                      \
+      stage_slab<BM, NW>(A, lda, row0, M, (t_)*64 + (slab)*32,                \
+                         &As[buf][slab][0], wid, lane);                       \
+  } while (0)
 #define KS_STAGE_B(buf, slab, t_)                                             \
-  stage_slab<BN, NW>(B, ldb, col0, N, (t_)*64 + (slab)*32,                    \
-                     &Bs[buf][slab][0], wid, lane)
+  do {                                                                        \
+    if (TRB)                                                                  \
+      stage_slab_tr<BN, NW>(B, ldb, col0, N, (t_)*64 + (slab)*32, K,          \
+                            &Bs[buf][slab][0], wid, lane);                    \
+    else                                                                      \
+      stage_slab<BN, NW>(B, ldb, col0, N, (t_)*64 + (slab)*32,                \
+                         &Bs[buf][slab][0], wid, lane);                       \
+  } while (0)
 
   // prologue: tile0 both slabs + tile1 slab0; leave 4 units in flight.
   KS_STAGE_A(0, 0, 0);
@@ -603,12 +663,14 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
         if (ph == 0) {
 #pragma unroll
           for (int n = 0; n < FN; ++n)
-            bfrag[n] = lds_frag_ks(bt, brow_base + n * 16, kq);
+            bfrag[n] = TRB ? lds_frag_ks_tr(bt, wn * FN + n, lane)
+                           : lds_frag_ks(bt, brow_base + n * 16, kq);
         }
 #pragma unroll
         for (int g = 0; g < MH; ++g)
-          afrag[g] =
-              lds_frag_ks(at, arow_base + (ph * MH + g) * 16, kq);
+          afrag[g] = TRA
+              ? lds_frag_ks_tr(at, wm * FM + ph * MH + g, lane)
+              : lds_frag_ks(at, arow_base + (ph * MH + g) * 16, kq);
         // prefetch issue: p0: A-s1(t+1), p1: B-s1(t+1),
         //                 p2: A-s0(t+2), p3: B-s0(t+2).
         // ALWAYS issued, with the tile index clamped at the tail: skipping
@@ -1020,6 +1082,35 @@ void gemm_x8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     case 2: LX8_T(256, 256, 4); break;
     case 3: LX8_T(256, 128, 4); break;
     case 4: LX8_T(128, 256, 4); break;
+#define LKS_T(BM, BN, WNW)                                                    \
+  do {                                                                        \
+    const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
+    if (trans_a && trans_b) {                                                 \
+      hipLaunchKernelGGL((gemm_ks_kernel<BM, BN, WNW, true, true>),           \
+                         dim3(mt * nt), dim3(2 * WNW * 64), 0,                \
+                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,    \
+                         ldb, ldc, (int)act, mt, nt);                         \
+    } else if (trans_b) {                                                     \
+      hipLaunchKernelGGL((gemm_ks_kernel<BM, BN, WNW, false, true>),          \
+                         dim3(mt * nt), dim3(2 * WNW * 64), 0,                \
+                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,    \
+                         ldb, ldc, (int)act, mt, nt);                         \
+    } else if (trans_a) {                                                     \
+      hipLaunchKernelGGL((gemm_ks_kernel<BM, BN, WNW, true, false>),          \
+                         dim3(mt * nt), dim3(2 * WNW * 64), 0,                \
+                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,    \
+                         ldb, ldc, (int)act, mt, nt);                         \
+    } else {                                                                  \
+      hipLaunchKernelGGL((gemm_ks_kernel<BM, BN, WNW, false, false>),         \
+                         dim3(mt * nt), dim3(2 * WNW * 64), 0,                \
+                         stream.stream(), a, b, c, bias_ptr, M, N, K, lda,    \
+                         ldb, ldc, (int)act, mt, nt);                         \
+    }                                                                         \
+  } while (0)
+    case 10: LKS_T(128, 128, 4); break;
+    case 11: LKS_T(256, 256, 4); break;
+    case 12: LKS_T(256, 128, 4); break;
+#undef LKS_T
     default: TORCH_CHECK(false, "gemm_x8: unknown variant");
   }
 #undef LX8

@@ -70,7 +70,7 @@ def main():
 
     if not args.skip_refcheck:
         fails = 0
-        for variant in (0, 1, 2, 3, 4):
+        for variant in (0, 1, 10, 11, 12):
             for (ta, tb) in ((True, True), (False, True), (True, False),
                              (False, False)):
                 for (M, N, K) in ((256, 256, 128), (512, 384, 192),
@@ -92,7 +92,7 @@ def main():
                 fails += 1
                 print("FAIL act1 v", variant, rel)
         # race screen on the dW shape
-        for v in (0, 1):
+        for v in (0, 10):
             for rep in range(6):
                 ok, rel = refcheck(2048, 2048, 2048, True, True, v)
                 if not ok:
@@ -124,8 +124,8 @@ def main():
         row["old_tr"] = tf(bench_fn(
             lambda: ext.gemm_tr_bf16(A, B, C, None, 0, int(ta), int(tb)),
             args.iters))
-        for v, name in ((0, "x8_128_8w"), (1, "x8_128_4w"), (2, "x8_256_8w"),
-                        (3, "x8_256x128"), (4, "x8_128x256")):
+        for v, name in ((0, "x8_128_8w"), (10, "ks_128_8w"),
+                        (11, "ks_256"), (12, "ks_256x128")):
             try:
                 row[name] = tf(bench_fn(
                     lambda v=v: ext.gemm_x8(A, B, C, None, 0, int(ta),
