@@ -381,9 +381,19 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   const int64_t b128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);
   const int64_t t_256x128 = (int64_t)((M + 255) / 256) * ((N + 127) / 128);
   const int64_t t_128x256 = (int64_t)((M + 127) / 128) * ((N + 255) / 256);
-  // measured wins at K=2048 (+9-20%) AND K=4096 (+11%): no K cap.
+  // Pipelined/K-slab routing (same-box A/B, profiles/gemm_x8_r02b.json):
+  //   dX (ft): K-slab wins at <=256-block grids (549 vs 441 @2048^2),
+  //            the half-tile pipeline wins at larger grids (588 vs 515
+  //            @2048x3072); measured at K=2048 AND 4096 (no K cap).
+  //   dW (tt): K-slab 128^2 wins the 2048-class (528 vs 442, +19%);
+  //            4096-class keeps the 2-phase 256x128 (986 vs 944).
   if (!trans_a && trans_b && K % 64 == 0 && b128 >= 256 && b128 < 1024) {
-    gemm_x8(A, B, C, bias, act, trans_a, trans_b, /*128^2 8-wave*/ 0);
+    gemm_x8(A, B, C, bias, act, trans_a, trans_b,
+            b128 <= 256 ? /*K-slab 128^2*/ 10 : /*half-tile 128^2*/ 0);
+    return;
+  }
+  if (trans_a && trans_b && K % 64 == 0 && b128 >= 256 && b128 < 1024) {
+    gemm_x8(A, B, C, bias, act, trans_a, trans_b, /*K-slab 128^2*/ 10);
     return;
   }
 #define LTR(BM, BN, FM, FN, MW, WGM, WGN)                                     \
