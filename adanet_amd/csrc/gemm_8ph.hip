@@ -581,7 +581,8 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
     int K, int64_t lda, int64_t ldb, int64_t ldc, int act, int mtiles,
-    int ntiles) {
+    int ntiles, const long long* __restrict__ seed = nullptr,
+    uint32_t pthresh = 0, float inv_keep = 1.f) {
   constexpr int NW = 2 * WN;
   constexpr int FM = BM / 32;          // m-fragments per wave (2 M-waves)
   constexpr int FN = BN / (WN * 16);
@@ -728,6 +729,12 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
         if (row >= M) continue;
         float v = acc[i][j][rr] + bv;
         if (act == 1) v = v > 0.f ? v : 0.f;
+        if (act == 3) {  // fused relu+dropout (stateless counter RNG)
+          v = v > 0.f ? v : 0.f;
+          const uint32_t rnd =
+              hash_rng((uint64_t)seed[0], (uint64_t)row * N + col);
+          v = (rnd >= pthresh) ? v * inv_keep : 0.f;
+        }
         bf16_t* cp = &C[(int64_t)row * ldc + col];
         if (act == 2) v += bf2f(*cp);
         *cp = f2bf(v);
@@ -1168,10 +1175,9 @@ bool gemm_nt8_try_dropout(const bf16_t* a, const bf16_t* b, bf16_t* c,
   }
   if (t128 >= 256) {
     const int mt = (M + 127) / 128, nt = (N + 127) / 128;
-    hipLaunchKernelGGL((gemm_nt8_kernel<128, 128, false, false>),
-                       dim3(mt * nt), dim3(512), 0, stream, a, b, c,
-                       bias_ptr, M, N, K, lda, ldb, ldc, 3, mt, nt, seed,
-                       pthresh, inv_keep);
+    hipLaunchKernelGGL((gemm_ks_kernel<128, 128, 4>), dim3(mt * nt),
+                       dim3(512), 0, stream, a, b, c, bias_ptr, M, N, K,
+                       lda, ldb, ldc, 3, mt, nt, seed, pthresh, inv_keep);
     return true;
   }
   return false;
