@@ -576,7 +576,13 @@ __device__ __forceinline__ s16x8 lds_frag_ks(const bf16_t* lds_slab,
   return out;
 }
 
-template <int BM, int BN, int WN, bool TRA = false, bool TRB = false>
+// BAR2=false drops the barrier between a slab's two compute phases:
+// both phases read the SAME already-published slab, and the next sync
+// point (vmcnt + barrier at the slab-pair end) re-aligns the waves —
+// phase-internal desync has no hazard (reads race only reads; stage
+// targets are disjoint from any in-flight reads).
+template <int BM, int BN, int WN, bool TRA = false, bool TRB = false,
+          bool BAR2 = true>
 __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -706,8 +712,10 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
           // end of the k-slab pair: publish the next consumers' data
           // (kk0: slab1(t) reads at p2; kk1: slab0(t+1) reads at t+1 p0)
           vmcnt_wait<INFLIGHT>();
+          S_BARRIER();
+        } else if (BAR2) {
+          S_BARRIER();
         }
-        S_BARRIER();
       }
     }
   }
@@ -1247,6 +1255,22 @@ void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     case 51: LAUNCH_KS(128, 128, 4); break;
     case 52: LAUNCH_KS(128, 128, 2); break;
     case 53: LAUNCH_KS(256, 128, 4); break;
+    case 55: {
+      const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+      hipLaunchKernelGGL(
+          (gemm_ks_kernel<256, 256, 4, false, false, false>), dim3(mt * nt),
+          dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M, N, K, lda,
+          ldb, ldc, (int)act, mt, nt);
+      break;
+    }
+    case 56: {
+      const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+      hipLaunchKernelGGL(
+          (gemm_ks_kernel<128, 128, 4, false, false, false>), dim3(mt * nt),
+          dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M, N, K, lda,
+          ldb, ldc, (int)act, mt, nt);
+      break;
+    }
 #undef LAUNCH_KS
     case 30: LAUNCH_8PHP(256, 256, false); break;
     case 31: LAUNCH_8PHP(256, 256, true); break;
