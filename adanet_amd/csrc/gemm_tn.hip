@@ -396,6 +396,13 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     gemm_x8(A, B, C, bias, act, trans_a, trans_b, /*K-slab 128^2*/ 10);
     return;
   }
+  const int64_t t256sq = (int64_t)((M + 255) / 256) * ((N + 255) / 256);
+  if (trans_a && trans_b && K % 64 == 0 && t256sq >= 200) {
+    // 4096-class tt: K-slab 256^2 (bar1) 1033 vs 996 TF for the 2-phase
+    // 256x128 (profiles/gemm_x8_r02c.json).
+    gemm_x8(A, B, C, bias, act, trans_a, trans_b, /*K-slab 256^2*/ 11);
+    return;
+  }
 #define LTR(BM, BN, FM, FN, MW, WGM, WGN)                                     \
   do {                                                                        \
     const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
