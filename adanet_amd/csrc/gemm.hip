@@ -356,7 +356,11 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   // improve_nas step, profiles/nasprof_summary.txt). Split the K range
   // over ~512 workgroups with fp32 atomic partials + a tiny epilogue.
   const int64_t t64 = (int64_t)((M + 63) / 64) * ((N + 63) / 64);
-  if (fast && act != 3 && t64 <= 8 && K >= 2048) {
+  const bool k_dominant =
+      (t64 <= 8 && K >= 2048) ||
+      (t64 <= 256 && K >= 4096 && (int64_t)K >= 4 * std::max(M, N));
+  if (fast && act != 3 && k_dominant &&
+      std::min<int64_t>(512 / t64, (int64_t)K / 256) >= 2) {
     auto C32 = at::zeros({M, N}, A.options().dtype(at::kFloat));
     int ksplit = (int)std::min<int64_t>(512 / t64, (int64_t)K / 256);
     if (ksplit < 2) ksplit = 2;
