@@ -582,7 +582,8 @@ __device__ __forceinline__ s16x8 lds_frag_ks(const bf16_t* lds_slab,
 // phase-internal desync has no hazard (reads race only reads; stage
 // targets are disjoint from any in-flight reads).
 template <int BM, int BN, int WN, bool TRA = false, bool TRB = false,
-          bool BAR2 = false>  // single barrier per slab measured +2-16%
+          bool BAR2 = false,  // single barrier per slab measured +2-16%
+          bool STAGGER = false>  // persistent even-wave priority boost
 __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
     const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
     bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -617,6 +618,11 @@ __global__ __launch_bounds__(2 * WN * 64, 2) void gemm_ks_kernel(
   for (int i = 0; i < FM; ++i)
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // Wave-stagger experiment: the barriers keep SIMD-sharing waves in
+  // lockstep, so both hit ds_read phases together and the MFMA pipe
+  // idles; a persistent priority split lets the scheduler drift them.
+  if (STAGGER && (wid & 1) == 0) __builtin_amdgcn_s_setprio(1);
 
   const int NT = K / 64;
 
@@ -1269,6 +1275,22 @@ void gemm_nt8(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
           (gemm_ks_kernel<128, 128, 4, false, false, false>), dim3(mt * nt),
           dim3(512), 0, stream.stream(), a, b, c, bias_ptr, M, N, K, lda,
           ldb, ldc, (int)act, mt, nt);
+      break;
+    }
+    case 57: {  // 256^2 + even-wave priority stagger
+      const int mt = (M + 255) / 256, nt = (N + 255) / 256;
+      hipLaunchKernelGGL(
+          (gemm_ks_kernel<256, 256, 4, false, false, false, true>),
+          dim3(mt * nt), dim3(512), 0, stream.stream(), a, b, c, bias_ptr,
+          M, N, K, lda, ldb, ldc, (int)act, mt, nt);
+      break;
+    }
+    case 58: {  // 128^2 8w + stagger
+      const int mt = (M + 127) / 128, nt = (N + 127) / 128;
+      hipLaunchKernelGGL(
+          (gemm_ks_kernel<128, 128, 4, false, false, false, true>),
+          dim3(mt * nt), dim3(512), 0, stream.stream(), a, b, c, bias_ptr,
+          M, N, K, lda, ldb, ldc, (int)act, mt, nt);
       break;
     }
 #undef LAUNCH_KS
