@@ -378,3 +378,96 @@ def test_allstrategy_candidate_not_split_across_ranks(tmp_path):
     for p in procs:
         p.join(timeout=120)
         assert p.exitcode == 0
+
+
+def _bench_like_worker(rank, world, port, model_dir, q):
+    """Mirrors bench.py's driver config: RoundRobin + Evaluator selection +
+    force_grow + world restarts (the exact N-GPU path the scale run
+    exercises, minus RCCL)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import functools
+
+        import adanet_amd
+        from adanet_amd.distributed import RoundRobinStrategy
+        from adanet_amd.head import MultiClassHead
+        from adanet_amd.models import simple_dnn
+        from adanet_amd.ops.optim import FusedSGD
+
+        torch.manual_seed(1234)
+        N, D, C = 256, 16, 4
+        X = torch.randn(N, D)
+        Y = (X @ torch.randn(D, C)).argmax(1)
+
+        def input_fn():
+            def gen():
+                g = torch.Generator().manual_seed(100 + rank)
+                while True:
+                    idx = torch.randint(0, N, (32,), generator=g)
+                    yield X[idx], Y[idx]
+
+            return gen()
+
+        eval_batches = [(X[i * 32:(i + 1) * 32], Y[i * 32:(i + 1) * 32])
+                        for i in range(4)]
+
+        def eval_input_fn():
+            return iter(list(eval_batches))
+
+        est = adanet_amd.Estimator(
+            head=MultiClassHead(C),
+            subnetwork_generator=simple_dnn.Generator(
+                optimizer_fn=functools.partial(FusedSGD, lr=0.05),
+                layer_size=8, initial_num_layers=1,
+                learn_mixture_weights=True, seed=77, num_restarts=world),
+            max_iteration_steps=5,
+            evaluator=adanet_amd.Evaluator(input_fn=eval_input_fn, steps=4),
+            force_grow=True,
+            model_dir=model_dir,
+            config=adanet_amd.RunConfig(tf_random_seed=42),
+            experimental_placement_strategy=RoundRobinStrategy(),
+            use_streams=False,
+        )
+        est.train(input_fn, steps=10)  # 2 full iterations
+        archs = {t: est._architectures[t] for t in sorted(est._architectures)}
+        q.put((rank, None, archs, est.iteration_number))
+        torch.distributed.destroy_process_group()
+    except Exception:
+        import traceback
+        q.put((rank, traceback.format_exc(), None, None))
+
+
+@pytest.mark.parametrize("world", [2, 8])
+def test_bench_config_evaluator_roundrobin(tmp_path, world):
+    """Evaluator-driven selection under round-robin at the driver's world
+    sizes: per-rank eval maps merge (inf placeholders for unowned
+    candidates must never win), every rank agrees on the winner, and the
+    ensemble grows every iteration (force_grow)."""
+    model_dir = str(tmp_path / "model")
+    os.makedirs(model_dir, exist_ok=True)
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_bench_like_worker,
+                         args=(r, world, port, model_dir, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, err, archs, it = _get(q)
+        assert err is None, "rank %s failed:\n%s" % (rank, err)
+        results[rank] = (archs, it)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    a0, it0 = results[0]
+    assert it0 == 2
+    for r in range(1, world):
+        assert results[r] == (a0, 2), "rank %d diverged" % r
+    arch1 = json.loads(a0[1])
+    assert len(arch1["subnetworks"]) == 2  # force_grow: +1 member/iter
