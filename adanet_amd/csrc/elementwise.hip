@@ -59,11 +59,14 @@ __global__ __launch_bounds__(256) void relu_bwd_kernel(
 // colsum kernel's full re-read of dz (8 MB at the bench shape; colsum was
 // 3.7% + relu_bwd 3.0% of step GPU time, profiles/bench_kernel_stats_r01e).
 // Layout mirrors colsum: thread-per-column (coalesced), row-chunk grid.y
-// fills the chip; db is accumulated via one atomicAdd per column per chunk
-// (direct-to-arena accum semantics — db never pre-zeroed here).
+// fills the chip (direct-to-arena accum semantics — db never pre-zeroed).
+// Deterministic: each (stripe, chunk) block writes its partial into its
+// OWN workspace slot; a fixed-order reducer finishes (fp32 atomics have
+// run-dependent ordering -> last-ulp bias-grad wobble that flips near-tie
+// candidate selections across otherwise-identical runs).
 __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
     const bf16_t* __restrict__ dy, const bf16_t* __restrict__ y,
-    bf16_t* __restrict__ dz, float* __restrict__ db, int B, int C,
+    bf16_t* __restrict__ dz, float* __restrict__ wsp, int B, int C,
     int rows_per_block, float scale) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
@@ -76,7 +79,19 @@ __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
     dz[i] = f2bf(g);
     acc += g;
   }
-  atomicAdd(&db[c], acc);
+  wsp[(int64_t)blockIdx.y * C + c] = acc;
+}
+
+// db[c] += sum_chunk wsp[chunk][c] in fixed chunk order (accumulate
+// semantics: db is the live arena grad view, never pre-zeroed here).
+__global__ __launch_bounds__(256) void colsum_chunk_reduce_kernel(
+    const float* __restrict__ wsp, float* __restrict__ db, int C,
+    int nchunks) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f;
+  for (int r = 0; r < nchunks; ++r) s += wsp[(int64_t)r * C + c];
+  db[c] += s;
 }
 
 // Batched device-to-device copy: up to 8 (src,dst) pairs per launch
@@ -194,11 +209,15 @@ void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
   const int row_chunks =
       std::max(1, std::min(1024 / stripes, (B + 7) / 8));
   const int rows_per_block = (B + row_chunks - 1) / row_chunks;
+  auto wsp = at::empty({row_chunks, C}, db.options());
   hipLaunchKernelGGL(relu_bwd_colsum_kernel,
                      dim3((unsigned)stripes, (unsigned)row_chunks), dim3(256),
                      0, stream.stream(), (const bf16_t*)dy.data_ptr(),
                      (const bf16_t*)y.data_ptr(), (bf16_t*)dz.data_ptr(),
-                     db.data_ptr<float>(), B, C, rows_per_block,
+                     wsp.data_ptr<float>(), B, C, rows_per_block,
                      (float)scale);
+  hipLaunchKernelGGL(colsum_chunk_reduce_kernel, dim3((unsigned)stripes),
+                     dim3(256), 0, stream.stream(), wsp.data_ptr<float>(),
+                     db.data_ptr<float>(), C, row_chunks);
   HIP_CHECK_KERNEL();
 }

@@ -35,7 +35,8 @@ __global__ __launch_bounds__(256) void mixer_fwd_kernel(
 // dw for SCALAR weights: dw[j] = sum_{b,c} dY[b,c] * L[j,b,c].
 __global__ __launch_bounds__(256) void mixer_bwd_dw_scalar_kernel(
     const bf16_t* __restrict__ stack, const bf16_t* __restrict__ dY,
-    float* __restrict__ dw, int J, int B, int C, int64_t stride_j, int ldy) {
+    float* __restrict__ wsp, int J, int B, int C, int64_t stride_j,
+    int ldy) {
   const int j = blockIdx.y;
   const bf16_t* L = stack + j * stride_j;
   const int64_t total = (int64_t)B * C;
@@ -50,8 +51,21 @@ __global__ __launch_bounds__(256) void mixer_bwd_dw_scalar_kernel(
   if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = acc;
   __syncthreads();
   if (threadIdx.x == 0) {
-    atomicAdd(&dw[j], partial[0] + partial[1] + partial[2] + partial[3]);
+    wsp[(int64_t)j * gridDim.x + blockIdx.x] =
+        partial[0] + partial[1] + partial[2] + partial[3];
   }
+}
+
+// dw[j] += sum_b wsp[j][b] in fixed block order (dw holds the running
+// arena gradient — accumulate, never overwrite).
+__global__ void mixer_dw_reduce_kernel(const float* __restrict__ wsp,
+                                       float* __restrict__ dw, int n,
+                                       int nblocks) {
+  const int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= n) return;
+  float s = 0.f;
+  for (int b = 0; b < nblocks; ++b) s += wsp[(int64_t)j * nblocks + b];
+  dw[j] += s;
 }
 
 // dw for VECTOR weights: dw[j][c] = sum_b dY[b,c] * L[j,b,c].
@@ -112,8 +126,11 @@ __global__ __launch_bounds__(256) void mixer_fwd_ptrs_kernel(
   }
 }
 
+// Deterministic: each block writes its partial into wsp[j][blockIdx.x];
+// mixer_dw_reduce_kernel sums in fixed block order (fp32 atomics had
+// run-dependent ordering -> last-ulp dw wobble across identical runs).
 __global__ __launch_bounds__(256) void mixer_bwd_dw_scalar_ptrs_kernel(
-    MixerPtrs Ls, const bf16_t* __restrict__ dY, float* __restrict__ dw,
+    MixerPtrs Ls, const bf16_t* __restrict__ dY, float* __restrict__ wsp,
     int B, int C, int ldy) {
   const int j = blockIdx.y;
   const bf16_t* L = Ls.p[j];
@@ -130,7 +147,8 @@ __global__ __launch_bounds__(256) void mixer_bwd_dw_scalar_ptrs_kernel(
   if ((threadIdx.x & 63) == 0) partial[threadIdx.x >> 6] = acc;
   __syncthreads();
   if (threadIdx.x == 0) {
-    atomicAdd(&dw[j], partial[0] + partial[1] + partial[2] + partial[3]);
+    wsp[(int64_t)j * gridDim.x + blockIdx.x] =
+        partial[0] + partial[1] + partial[2] + partial[3];
   }
 }
 
@@ -220,9 +238,13 @@ void mixer_bwd_dw_direct(const std::vector<at::Tensor>& members,
       dim3 grid(
           (unsigned)std::min<int64_t>(((int64_t)B * C + 2047) / 2048, 256),
           (unsigned)n);
+      auto wsp = at::empty({n, (int)grid.x}, dw.options());
       hipLaunchKernelGGL(mixer_bwd_dw_scalar_ptrs_kernel, grid, dim3(256), 0,
                          stream.stream(), P, (const bf16_t*)dY.data_ptr(),
-                         dwp, B, C, (int)dY.stride(0));
+                         wsp.data_ptr<float>(), B, C, (int)dY.stride(0));
+      hipLaunchKernelGGL(mixer_dw_reduce_kernel, dim3(1), dim3(64), 0,
+                         stream.stream(), wsp.data_ptr<float>(), dwp, n,
+                         (int)grid.x);
     }
     HIP_CHECK_KERNEL();
   }
@@ -264,10 +286,14 @@ void mixer_bwd_dw(const at::Tensor& stack, const at::Tensor& dY,
     dim3 grid(
         (unsigned)std::min<int64_t>(((int64_t)B * C + 2047) / 2048, 256),
         (unsigned)J);
+    auto wsp = at::empty({J, (int)grid.x}, dw.options());
     hipLaunchKernelGGL(mixer_bwd_dw_scalar_kernel, grid, dim3(256), 0,
                        stream.stream(), (const bf16_t*)stack.data_ptr(),
-                       (const bf16_t*)dY.data_ptr(), dw.data_ptr<float>(), J,
-                       B, C, (int64_t)stack.stride(0), (int)dY.stride(0));
+                       (const bf16_t*)dY.data_ptr(), wsp.data_ptr<float>(),
+                       J, B, C, (int64_t)stack.stride(0), (int)dY.stride(0));
+    hipLaunchKernelGGL(mixer_dw_reduce_kernel, dim3(1), dim3(64), 0,
+                       stream.stream(), wsp.data_ptr<float>(),
+                       dw.data_ptr<float>(), J, (int)grid.x);
   }
   HIP_CHECK_KERNEL();
 }

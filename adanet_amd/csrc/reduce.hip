@@ -8,26 +8,39 @@
 
 // 2-D grid: (column stripes, row chunks). Each block reduces its row chunk
 // for a 256-column stripe (coalesced: consecutive threads read consecutive
-// columns of each row) and atomically adds one partial per column — the
-// row-chunk axis is what fills the 256-CU chip (a single-stripe launch
-// occupied 8 CUs and was 54% of the training step before this).
-// `out` MUST be zero-initialized when gridDim.y > 1.
+// columns of each row); the row-chunk axis is what fills the 256-CU chip
+// (a single-stripe launch occupied 8 CUs and was 54% of the training step
+// before this). Multi-chunk partials land in `wsp` and a fixed-order
+// reducer finishes — bit-deterministic across runs.
 __global__ __launch_bounds__(256) void colsum_bf16_kernel(
     const bf16_t* __restrict__ x, float* __restrict__ out, int B, int C,
-    int ldx, int rows_per_block, int accum) {
+    int ldx, int rows_per_block, int accum, float* __restrict__ wsp) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   const int r0 = blockIdx.y * rows_per_block;
   const int r1 = min(B, r0 + rows_per_block);
   float acc = 0.f;
   for (int b = r0; b < r1; ++b) acc += bf2f(x[(int64_t)b * ldx + c]);
-  if (gridDim.y == 1 && !accum) {
-    out[c] = acc;
+  if (gridDim.y == 1) {
+    if (accum) out[c] += acc;  // single chunk: deterministic add
+    else out[c] = acc;
   } else {
-    // accum mode: out already holds the running gradient (direct-to-arena
-    // bias-grad write) — always add, never overwrite, never pre-zero.
-    atomicAdd(&out[c], acc);
+    // multi-chunk: per-chunk workspace slot; a fixed-order reducer
+    // finishes (atomics had run-dependent fp32 ordering).
+    wsp[(int64_t)blockIdx.y * C + c] = acc;
   }
+}
+
+// out[c] (+)= sum_chunk wsp[chunk][c] in fixed chunk order.
+__global__ __launch_bounds__(256) void colsum_reduce_kernel(
+    const float* __restrict__ wsp, float* __restrict__ out, int C,
+    int nchunks, int accum) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f;
+  for (int r = 0; r < nchunks; ++r) s += wsp[(int64_t)r * C + c];
+  if (accum) out[c] += s;
+  else out[c] = s;
 }
 
 // argmax over the class dim + count of matches with labels (accuracy numer).
@@ -107,15 +120,27 @@ void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum) {
   const int stripes = (C + 255) / 256;
   // Fill the chip: aim for ~1024 blocks, at least 8 rows per chunk.
   int row_chunks = std::max(1, std::min(1024 / stripes, (B + 7) / 8));
-  if (row_chunks > 1 && !accum) {
-    out.zero_();
-  }
   const int rows_per_block = (B + row_chunks - 1) / row_chunks;
+  if (row_chunks > 1) {
+    auto wsp = at::empty({row_chunks, C}, out.options());
+    hipLaunchKernelGGL(colsum_bf16_kernel,
+                       dim3((unsigned)stripes, (unsigned)row_chunks),
+                       dim3(256), 0, stream.stream(),
+                       (const bf16_t*)x.data_ptr(), out.data_ptr<float>(), B,
+                       C, (int)x.stride(0), rows_per_block, (int)accum,
+                       wsp.data_ptr<float>());
+    hipLaunchKernelGGL(colsum_reduce_kernel, dim3((unsigned)stripes),
+                       dim3(256), 0, stream.stream(),
+                       wsp.data_ptr<float>(), out.data_ptr<float>(), C,
+                       row_chunks, (int)accum);
+    HIP_CHECK_KERNEL();
+    return;
+  }
   hipLaunchKernelGGL(colsum_bf16_kernel,
                      dim3((unsigned)stripes, (unsigned)row_chunks), dim3(256),
                      0, stream.stream(), (const bf16_t*)x.data_ptr(),
                      out.data_ptr<float>(), B, C, (int)x.stride(0),
-                     rows_per_block, (int)accum);
+                     rows_per_block, (int)accum, (float*)nullptr);
   HIP_CHECK_KERNEL();
 }
 
