@@ -89,11 +89,48 @@ __global__ __launch_bounds__(256) void depthwise_fwd_lds_kernel(
   const int ni = min(P, BC - g0);
   const int npx_in = ni * H * W;
   const bf16_t* gx = x + (int64_t)g0 * H * W;
-  for (int i = threadIdx.x; i < npx_in; i += 256) xs[i] = gx[i];
+  if ((npx_in & 7) == 0) {  // image sizes are multiples of 8 in practice
+    const uint4* src = (const uint4*)gx;
+    uint4* dst = (uint4*)xs;
+    for (int i = threadIdx.x; i < (npx_in >> 3); i += 256) dst[i] = src[i];
+  } else {
+    for (int i = threadIdx.x; i < npx_in; i += 256) xs[i] = gx[i];
+  }
   __syncthreads();
   const int opx = OH * OW;
-  const int npx_out = ni * opx;
   bf16_t* gy = y + (int64_t)g0 * opx;
+  if (opx >= 256) {
+    // image-outer: whole block works one image at a time so the KS^2 tap
+    // weights live in registers (the flat loop re-read them from L1 per
+    // pixel — measured slower than the old register-tap kernel).
+    for (int img = 0; img < ni; ++img) {
+      const int c = (g0 + img) % C;
+      float wr[KS * KS];
+#pragma unroll
+      for (int i = 0; i < KS * KS; ++i) wr[i] = bf2f(w[c * KS * KS + i]);
+      const bf16_t* xi = xs + img * H * W;
+      bf16_t* yi = gy + (int64_t)img * opx;
+      for (int p = threadIdx.x; p < opx; p += 256) {
+        const int oh = p / OW, ow = p - (p / OW) * OW;
+        const int ih0 = oh * stride - pad, iw0 = ow * stride - pad;
+        float acc = 0.f;
+#pragma unroll
+        for (int kh = 0; kh < KS; ++kh) {
+          const int ih = ih0 + kh;
+          if (ih < 0 || ih >= H) continue;
+#pragma unroll
+          for (int kw = 0; kw < KS; ++kw) {
+            const int iw = iw0 + kw;
+            if (iw < 0 || iw >= W) continue;
+            acc += wr[kh * KS + kw] * bf2f(xi[ih * W + iw]);
+          }
+        }
+        yi[p] = f2bf(acc);
+      }
+    }
+    return;
+  }
+  const int npx_out = ni * opx;
   for (int p = threadIdx.x; p < npx_out; p += 256) {
     const int img = p / opx;
     const int rem = p - img * opx;
@@ -128,11 +165,52 @@ __global__ __launch_bounds__(256) void depthwise_bwd_dx_lds_kernel(
   const int g0 = blockIdx.y * P;
   const int ni = min(P, BC - g0);
   const int opx = OH * OW;
+  const int nin = ni * opx;
   const bf16_t* gdy = dy + (int64_t)g0 * opx;
-  for (int i = threadIdx.x; i < ni * opx; i += 256) ds[i] = gdy[i];
+  if ((nin & 7) == 0) {
+    const uint4* src = (const uint4*)gdy;
+    uint4* dst = (uint4*)ds;
+    for (int i = threadIdx.x; i < (nin >> 3); i += 256) dst[i] = src[i];
+  } else {
+    for (int i = threadIdx.x; i < nin; i += 256) ds[i] = gdy[i];
+  }
   __syncthreads();
   const int ipx = H * W;
   bf16_t* gdx = dx + (int64_t)g0 * ipx;
+  if (ipx >= 256) {
+    // image-outer: tap weights in registers for the whole image.
+    for (int img = 0; img < ni; ++img) {
+      const int c = (g0 + img) % C;
+      float wr[KS * KS];
+#pragma unroll
+      for (int i = 0; i < KS * KS; ++i) wr[i] = bf2f(w[c * KS * KS + i]);
+      const bf16_t* di = ds + img * opx;
+      bf16_t* xo = gdx + (int64_t)img * ipx;
+      for (int p = threadIdx.x; p < ipx; p += 256) {
+        const int ih = p / W, iw = p - (p / W) * W;
+        float acc = 0.f;
+#pragma unroll
+        for (int kh = 0; kh < KS; ++kh) {
+          const int num_h = ih + pad - kh;
+          if (num_h < 0 || (STRIDE > 1 && (num_h % STRIDE))) continue;
+          const int oh = num_h / STRIDE;
+          if (oh >= OH) continue;
+#pragma unroll
+          for (int kw = 0; kw < KS; ++kw) {
+            const int num_w = iw + pad - kw;
+            if (num_w < 0 || (STRIDE > 1 && (num_w % STRIDE))) continue;
+            const int ow = num_w / STRIDE;
+            if (ow >= OW) continue;
+            acc += wr[kh * KS + kw] * bf2f(di[oh * OW + ow]);
+          }
+        }
+        xo[p] = f2bf(acc);
+      }
+    }
+    return;
+  }
+  // small images: flat loop over every output pixel of the group (weights
+  // from L1 — C is large exactly when images are small).
   for (int p = threadIdx.x; p < ni * ipx; p += 256) {
     const int img = p / ipx;
     const int rem = p - img * ipx;
@@ -182,13 +260,29 @@ __global__ __launch_bounds__(256) void depthwise_bwd_dw_lds_kernel(
   for (int t = 0; t < KS * KS; ++t) acc[t] = 0.f;
   for (int g = b0; g < b1; g += G) {
     const int ni = min(G, b1 - g);
-    for (int i = threadIdx.x; i < ni * ipx; i += 256) {
-      const int img = i / ipx;
-      xs[i] = x[((int64_t)(g + img) * C + c) * ipx + (i - img * ipx)];
+    if ((ipx & 7) == 0) {
+      for (int i = threadIdx.x; i < ((ni * ipx) >> 3); i += 256) {
+        const int img = i / (ipx >> 3);
+        ((uint4*)xs)[i] = ((const uint4*)(
+            x + ((int64_t)(g + img) * C + c) * ipx))[i - img * (ipx >> 3)];
+      }
+    } else {
+      for (int i = threadIdx.x; i < ni * ipx; i += 256) {
+        const int img = i / ipx;
+        xs[i] = x[((int64_t)(g + img) * C + c) * ipx + (i - img * ipx)];
+      }
     }
-    for (int i = threadIdx.x; i < ni * opx; i += 256) {
-      const int img = i / opx;
-      ds[i] = dy[((int64_t)(g + img) * C + c) * opx + (i - img * opx)];
+    if ((opx & 7) == 0) {
+      for (int i = threadIdx.x; i < ((ni * opx) >> 3); i += 256) {
+        const int img = i / (opx >> 3);
+        ((uint4*)ds)[i] = ((const uint4*)(
+            dy + ((int64_t)(g + img) * C + c) * opx))[i - img * (opx >> 3)];
+      }
+    } else {
+      for (int i = threadIdx.x; i < ni * opx; i += 256) {
+        const int img = i / opx;
+        ds[i] = dy[((int64_t)(g + img) * C + c) * opx + (i - img * opx)];
+      }
     }
     __syncthreads();
     for (int p = threadIdx.x; p < ni * opx; p += 256) {
@@ -412,7 +506,10 @@ void depthwise_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
   auto stream = at::cuda::getCurrentCUDAStream();
   const int64_t bc = (int64_t)B * C;
   if ((KS == 3 || KS == 5 || KS == 7) && H * W <= 8192 && !dw_old_path()) {
-    const int P = std::max(1, std::min((int)(8192 / (H * W)), 64));
+    // P bounded by LDS (staged input) AND by output work per block (keep
+    // blocks small enough that ceil(BC/P) fills the chip).
+    const int P = std::max(1, std::min((int)(8192 / (H * W)),
+                                       std::max(1, 4096 / (OH * OW))));
     const int groups = (int)((bc + P - 1) / P);
     auto kern = KS == 3   ? depthwise_fwd_lds_kernel<3>
                 : KS == 5 ? depthwise_fwd_lds_kernel<5>
@@ -451,7 +548,8 @@ void depthwise_bwd_dx(const at::Tensor& dy, const at::Tensor& w,
   const int64_t bc = (int64_t)B * C;
   if ((KS == 3 || KS == 5 || KS == 7) && (stride == 1 || stride == 2) &&
       OH * OW <= 8192 && !dw_old_path()) {
-    const int P = std::max(1, std::min((int)(8192 / (OH * OW)), 64));
+    const int P = std::max(1, std::min((int)(8192 / (OH * OW)),
+                                       std::max(1, 4096 / (H * W))));
     const int groups = (int)((bc + P - 1) / P);
     using kern_t = void (*)(const bf16_t*, const bf16_t*, bf16_t*, int, int,
                             int, int, int, int, int, int);
