@@ -37,10 +37,13 @@ __device__ __forceinline__ float block_reduce(float v, float* scratch,
 
 // Chunked partial sums: grid (C, nchunks) so small-C shapes still fill
 // the chip (a C-block grid was 30% of the improve_nas step on 32 CUs,
-// profiles/nasprof3_summary.txt); one atomicAdd pair per block.
+// profiles/nasprof3_summary.txt). Each block writes its OWN workspace
+// slot wsp[chunk][2][C]; the finalize kernel sums chunks in fixed order —
+// deterministic, and the workspace needs no zero-fill (the at::zeros +
+// .zero_() launches were 2.9% of the step, profiles/nasprof6_summary).
 __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
-    const bf16_t* __restrict__ x, float* __restrict__ ssum,
-    float* __restrict__ s2sum, int N, int C, int64_t HW, int nchunks) {
+    const bf16_t* __restrict__ x, float* __restrict__ wsp, int N, int C,
+    int64_t HW, int nchunks) {
   __shared__ float scratch[4];
   const int c = blockIdx.x;
   const int n0 = (N * blockIdx.y) / nchunks;
@@ -70,21 +73,28 @@ __global__ __launch_bounds__(256) void bn_stats_partial_kernel(
   s = block_reduce(s, scratch, tid);
   s2 = block_reduce(s2, scratch, tid);
   if (tid == 0) {
-    atomicAdd(&ssum[c], s);
-    atomicAdd(&s2sum[c], s2);
+    float* slot = wsp + (int64_t)blockIdx.y * 2 * C;
+    slot[c] = s;
+    slot[C + c] = s2;
   }
 }
 
 __global__ __launch_bounds__(256) void bn_stats_finalize_kernel(
-    const float* __restrict__ ssum, const float* __restrict__ s2sum,
-    float* __restrict__ mean_out, float* __restrict__ rstd_out,
-    float* __restrict__ running_mean, float* __restrict__ running_var,
-    int N, int C, int64_t HW, float eps, float momentum) {
+    const float* __restrict__ wsp, float* __restrict__ mean_out,
+    float* __restrict__ rstd_out, float* __restrict__ running_mean,
+    float* __restrict__ running_var, int N, int C, int64_t HW, float eps,
+    float momentum, int nchunks) {
   const int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  float s = 0.f, s2 = 0.f;
+  for (int r = 0; r < nchunks; ++r) {
+    const float* slot = wsp + (int64_t)r * 2 * C;
+    s += slot[c];
+    s2 += slot[C + c];
+  }
   const double M = (double)N * (double)HW;
-  const float mean = (float)(ssum[c] / M);
-  float var = (float)(s2sum[c] / M) - mean * mean;
+  const float mean = (float)(s / M);
+  float var = (float)(s2 / M) - mean * mean;
   var = var > 0.f ? var : 0.f;
   mean_out[c] = mean;
   rstd_out[c] = rsqrtf(var + eps);
@@ -159,12 +169,12 @@ __global__ __launch_bounds__(256) void bn_norm_kernel(
 }
 
 // Chunked/vectorized variant of the backward reduce: grid (C, nchunks),
-// atomicAdd per block (sdy/sdyx zeroed by the host).
+// per-chunk workspace slots wsp[chunk][2][C] + fixed-order finalize
+// (deterministic; no host zero-fill).
 __global__ __launch_bounds__(256) void bn_bwd_reduce_chunked_kernel(
     const bf16_t* __restrict__ x, const bf16_t* __restrict__ dy,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ sdy, float* __restrict__ sdyx, int N, int C,
-    int64_t HW, int nchunks) {
+    float* __restrict__ wsp, int N, int C, int64_t HW, int nchunks) {
   __shared__ float scratch[4];
   const int c = blockIdx.x;
   const int n0 = (N * blockIdx.y) / nchunks;
@@ -196,9 +206,26 @@ __global__ __launch_bounds__(256) void bn_bwd_reduce_chunked_kernel(
   a = block_reduce(a, scratch, tid);
   b = block_reduce(b, scratch, tid);
   if (tid == 0) {
-    atomicAdd(&sdy[c], a);
-    atomicAdd(&sdyx[c], b);
+    float* slot = wsp + (int64_t)blockIdx.y * 2 * C;
+    slot[c] = a;
+    slot[C + c] = b;
   }
+}
+
+// sdy/sdyx[c] = sum over chunks in fixed order.
+__global__ __launch_bounds__(256) void bn_bwd_finalize_kernel(
+    const float* __restrict__ wsp, float* __restrict__ sdy,
+    float* __restrict__ sdyx, int C, int nchunks) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  for (int r = 0; r < nchunks; ++r) {
+    const float* slot = wsp + (int64_t)r * 2 * C;
+    a += slot[c];
+    b += slot[C + c];
+  }
+  sdy[c] = a;
+  sdyx[c] = b;
 }
 
 // Per-channel backward sums: sdy = sum dy (== dbeta), sdyx = sum dy*xhat
@@ -271,15 +298,14 @@ void batchnorm_stats(const at::Tensor& x, at::Tensor& mean, at::Tensor& rstd,
   const int nchunks = (int)std::max<int64_t>(
       1, std::min<int64_t>(2048 / C, std::min<int64_t>(N, 64)));
   if (nchunks > 1) {
-    auto sums = at::zeros({2, C}, x.options().dtype(at::kFloat));
-    float* ssum = sums.data_ptr<float>();
+    auto wsp = at::empty({nchunks, 2, C}, x.options().dtype(at::kFloat));
     hipLaunchKernelGGL(bn_stats_partial_kernel, dim3(C, nchunks), dim3(256),
                        0, stream.stream(), (const bf16_t*)x.data_ptr(),
-                       ssum, ssum + C, N, C, HW, nchunks);
+                       wsp.data_ptr<float>(), N, C, HW, nchunks);
     hipLaunchKernelGGL(bn_stats_finalize_kernel, dim3((C + 255) / 256),
-                       dim3(256), 0, stream.stream(), ssum, ssum + C,
+                       dim3(256), 0, stream.stream(), wsp.data_ptr<float>(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(), rm,
-                       rv, N, C, HW, (float)eps, (float)momentum);
+                       rv, N, C, HW, (float)eps, (float)momentum, nchunks);
     HIP_CHECK_KERNEL();
     return;
   }
@@ -320,14 +346,17 @@ void batchnorm_bwd(const at::Tensor& x, const at::Tensor& dy, at::Tensor& dx,
   const int nchunks = (int)std::max<int64_t>(
       1, std::min<int64_t>(2048 / C, std::min<int64_t>(N, 64)));
   if (nchunks > 1) {
-    sdy.zero_();
-    sdyx.zero_();
+    auto wsp = at::empty({nchunks, 2, C}, sdy.options());
     hipLaunchKernelGGL(bn_bwd_reduce_chunked_kernel, dim3(C, nchunks),
                        dim3(256), 0, stream.stream(),
                        (const bf16_t*)x.data_ptr(),
                        (const bf16_t*)dy.data_ptr(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), sdy.data_ptr<float>(),
-                       sdyx.data_ptr<float>(), N, C, HW, nchunks);
+                       rstd.data_ptr<float>(), wsp.data_ptr<float>(), N, C,
+                       HW, nchunks);
+    hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3((C + 255) / 256),
+                       dim3(256), 0, stream.stream(), wsp.data_ptr<float>(),
+                       sdy.data_ptr<float>(), sdyx.data_ptr<float>(), C,
+                       nchunks);
   } else {
     hipLaunchKernelGGL(bn_bwd_reduce_kernel, dim3(C), dim3(256), 0,
                        stream.stream(), (const bf16_t*)x.data_ptr(),

@@ -40,8 +40,12 @@ def drop_path(x, keep_prob: float, training: bool):
     """Per-sample stochastic branch drop (reference nasnet_utils.py:137)."""
     if not training or keep_prob >= 1.0:
         return x
-    mask = torch.rand(x.shape[0], 1, 1, 1, device=x.device) < keep_prob
-    return x * mask.to(x.dtype) / keep_prob
+    # Fold 1/keep into the tiny [B,1,1,1] mask so the full-size tensor is
+    # touched ONCE (x * mask / keep was two full-tensor kernels fwd and two
+    # in backward — the broadcast mul soup was 4%+ of the NASNet step).
+    mask = (torch.rand(x.shape[0], 1, 1, 1, device=x.device)
+            < keep_prob).to(x.dtype)
+    return x * (mask * (1.0 / keep_prob))
 
 
 class SepConv(nn.Module):
