@@ -68,8 +68,10 @@ __device__ __forceinline__ void stage_tile_nt(
 // BM x BN tile, 2x2 wave grid, FM x FN 16x16 MFMA fragments per wave,
 // KSTEP K-columns staged per barrier (KSTEP/32 MFMA K-slices).
 // SPLITK > 1: the grid carries SPLITK K-slabs per output tile; each slab
-// atomically accumulates fp32 partials into C32 (no bias/act), and a
-// separate epilogue kernel finishes bias+ReLU+bf16 — trades a little
+// writes its fp32 partial into its OWN C32 slot [split][M][N] (no bias/
+// act), and a separate epilogue kernel sums the slots in fixed order and
+// finishes bias+ReLU+bf16 — deterministic (fp32 atomic ordering used to
+// wobble conv dW), no zero-fill, no atomic contention; trades a little
 // output traffic for filling all 256 CUs with the efficient big tile.
 template <int BM, int BN, int FM, int FN, int MINWAVES, int KSTEP = 32,
           bool SPLITK = false, int WGM = 2, int WGN = 2, int GROUPM = 0>
@@ -187,7 +189,7 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
         const int row = row0 + wm * (FM * 16) + i * 16 + c_row_base + rr;
         if (row >= M) continue;
         if (SPLITK) {
-          atomicAdd(&C32[(int64_t)row * N + col], acc[i][j][rr]);
+          C32[((int64_t)split * M + row) * N + col] = acc[i][j][rr];
         } else {
           float v = acc[i][j][rr] + bv;
           if (act == 1) v = v > 0.f ? v : 0.f;
@@ -207,15 +209,17 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_nt_bf16_kernel(
   }
 }
 
-// Finishes a split-K accumulation: bf16 C = act(C32 + bias).
+// Finishes a split-K accumulation: bf16 C = act(sum_slots C32 + bias),
+// summing the ksplit slots in fixed order (deterministic).
 __global__ __launch_bounds__(256) void gemm_splitk_epilogue_kernel(
     const float* __restrict__ C32, const float* __restrict__ bias,
-    bf16_t* __restrict__ C, int M, int N, int ldc, int act) {
+    bf16_t* __restrict__ C, int M, int N, int ldc, int act, int ksplit) {
   const int64_t total = (int64_t)M * N;
   for (int64_t p = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; p < total;
        p += (int64_t)gridDim.x * blockDim.x) {
     const int m = (int)(p / N), n = (int)(p % N);
-    float v = C32[p] + (bias ? bias[n] : 0.f);
+    float v = bias ? bias[n] : 0.f;
+    for (int r = 0; r < ksplit; ++r) v += C32[(int64_t)r * total + p];
     if (act == 1) v = v > 0.f ? v : 0.f;
     if (act == 2) v += bf2f(C[(int64_t)m * ldc + n]);
     C[(int64_t)m * ldc + n] = f2bf(v);
@@ -354,16 +358,16 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
   // K=B*H*W up to ~262k): the tile grid is 1-8 workgroups, so the plain
   // kernel runs one CU for a ms-scale reduction (measured 49% of the
   // improve_nas step, profiles/nasprof_summary.txt). Split the K range
-  // over ~512 workgroups with fp32 atomic partials + a tiny epilogue.
+  // over ~512 workgroups with slotted fp32 partials + a tiny epilogue.
   const int64_t t64 = (int64_t)((M + 63) / 64) * ((N + 63) / 64);
   const bool k_dominant =
       (t64 <= 8 && K >= 2048) ||
       (t64 <= 256 && K >= 4096 && (int64_t)K >= 4 * std::max(M, N));
   if (fast && act != 3 && k_dominant &&
       std::min<int64_t>(512 / t64, (int64_t)K / 256) >= 2) {
-    auto C32 = at::zeros({M, N}, A.options().dtype(at::kFloat));
     int ksplit = (int)std::min<int64_t>(512 / t64, (int64_t)K / 256);
     if (ksplit < 2) ksplit = 2;
+    auto C32 = at::empty({ksplit, M, N}, A.options().dtype(at::kFloat));
     const int mt = (M + 63) / 64, nt = (N + 63) / 64;
     hipLaunchKernelGGL((gemm_nt_bf16_kernel<64, 64, 2, 2, 6, 32, true>),
                        dim3(mt * nt * ksplit), dim3(THREADS), 0,
@@ -373,7 +377,7 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,
                        dim3((int)std::min<int64_t>((tot + 255) / 256, 2048)),
                        dim3(256), 0, stream.stream(), C32.data_ptr<float>(),
-                       bias_ptr, c, M, N, ldc, (int)act);
+                       bias_ptr, c, M, N, ldc, (int)act, ksplit);
     HIP_CHECK_KERNEL();
     return;
   }
@@ -472,7 +476,7 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
   do {                                                                        \
     TORCH_CHECK(K % KS == 0, "probe: K %% KSTEP");                            \
     const int mt = (M + BM - 1) / BM, nt = (N + BN - 1) / BN;                 \
-    auto C32 = at::zeros({(int64_t)M, (int64_t)N},                            \
+    auto C32 = at::empty({(int64_t)SPLIT, (int64_t)M, (int64_t)N},            \
                          A.options().dtype(at::kFloat));                      \
     hipLaunchKernelGGL((gemm_nt_bf16_kernel<BM, BN, FM, FN, MW, KS, true>),   \
                        dim3(mt * nt * SPLIT), dim3(THREADS), 0,              \
@@ -482,7 +486,8 @@ void gemm_nt_bf16_probe(const at::Tensor& A, const at::Tensor& B,
     hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,                           \
                        dim3((int)std::min<int64_t>((tot + 255) / 256, 2048)), \
                        dim3(256), 0, stream.stream(),                         \
-                       C32.data_ptr<float>(), nullptr, c, M, N, ldc, 0);      \
+                       C32.data_ptr<float>(), nullptr, c, M, N, ldc, 0,      \
+                       SPLIT);                                                \
   } while (0)
 
   switch (variant) {
