@@ -1,0 +1,65 @@
+"""Bisect NAS-space GPU nondeterminism: train a bare NasNetCIFAR for N
+steps twice with identical seeds and compare parameter bytes exactly.
+Isolates the model/kernel stack from the estimator machinery (the CPU
+run of the full bench is bit-deterministic; the GPU run is not).
+
+python benchmarks/nas_det_probe.py [--steps 10] [--no-droppath]
+"""
+
+import argparse
+import hashlib
+import sys
+
+import torch
+
+sys.path.insert(0, ".")
+from adanet_amd.models.nasnet import NasNetCIFAR  # noqa: E402
+
+
+def run_once(steps, drop_path_keep, dtype=torch.bfloat16):
+    torch.manual_seed(7)
+    model = NasNetCIFAR(num_cells=3, num_conv_filters=32,
+                        drop_path_keep=drop_path_keep).cuda()
+    from adanet_amd.ops.linear import restore_fp32_params
+    model = model.to(torch.bfloat16)
+    restore_fp32_params(model)
+    model.train()
+    opt = torch.optim.SGD([p for p in model.parameters()], lr=0.025,
+                          momentum=0.9)
+    torch.manual_seed(13)
+    x = (torch.randn(64, 3, 32, 32, device="cuda") / 4).to(dtype)
+    y = torch.randint(0, 10, (64,), device="cuda")
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad(set_to_none=True)
+        logits = model(x)
+        loss = torch.nn.functional.cross_entropy(logits.float(), y)
+        loss.backward()
+        opt.step()
+        losses.append(loss.item())
+    h = hashlib.sha256()
+    for name, p in sorted(model.named_parameters()):
+        h.update(name.encode())
+        h.update(p.detach().float().cpu().numpy().tobytes())
+    return h.hexdigest(), losses
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--no-droppath", action="store_true")
+    args = ap.parse_args()
+    keep = 1.0 if args.no_droppath else 0.9
+    h1, l1 = run_once(args.steps, keep)
+    h2, l2 = run_once(args.steps, keep)
+    match = h1 == h2
+    print("param hash match:", match)
+    if not match:
+        for i, (a, b) in enumerate(zip(l1, l2)):
+            flag = "  <-- diverges" if a != b else ""
+            print("step %d loss: %.9f vs %.9f%s" % (i, a, b, flag))
+    sys.exit(0 if match else 1)
+
+
+if __name__ == "__main__":
+    main()
