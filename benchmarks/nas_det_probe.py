@@ -32,7 +32,7 @@ def run_once(steps, drop_path_keep, dtype=torch.bfloat16):
     losses = []
     for _ in range(steps):
         opt.zero_grad(set_to_none=True)
-        logits = model(x)
+        _, logits = model(x)
         loss = torch.nn.functional.cross_entropy(logits.float(), y)
         loss.backward()
         opt.step()
