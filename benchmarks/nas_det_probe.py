@@ -11,9 +11,34 @@ import hashlib
 import sys
 
 import torch
+import torch.nn.functional as F
 
 sys.path.insert(0, ".")
 from adanet_amd.models.nasnet import NasNetCIFAR  # noqa: E402
+from adanet_amd.ops import conv as aconv  # noqa: E402
+
+
+def disable_native(classes):
+    """Monkeypatch chosen Hip conv-family modules to their torch fallback
+    so the racy class can be bisected inside the full model."""
+    if "dw" in classes:
+        aconv.HipDepthwiseConv2d.forward = lambda self, x: F.conv2d(
+            x, self.weight.to(x.dtype), None, self.stride, self.padding,
+            groups=self.channels)
+    if "c1" in classes:
+        aconv.HipConv1x1.forward = lambda self, x: F.conv2d(
+            x, self.weight.to(x.dtype),
+            self.bias.to(x.dtype) if self.bias is not None else None)
+    if "cn" in classes:
+        aconv.HipConvNxN.forward = lambda self, x: F.conv2d(
+            x, self.weight.to(x.dtype),
+            self.bias.to(x.dtype) if self.bias is not None else None,
+            self.stride, self.padding)
+    if "pool" in classes:
+        aconv.HipPool2d.forward = lambda self, x: (
+            F.max_pool2d(x, 3, self.stride, 1) if self.kind == "max"
+            else F.avg_pool2d(x, 3, self.stride, 1,
+                              count_include_pad=False))
 
 
 def run_once(steps, drop_path_keep, dtype=torch.bfloat16):
@@ -48,7 +73,11 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=10)
     ap.add_argument("--no-droppath", action="store_true")
+    ap.add_argument("--disable", default="",
+                    help="comma list of dw,c1,cn,pool to force torch fallback")
     args = ap.parse_args()
+    if args.disable:
+        disable_native(args.disable.split(","))
     keep = 1.0 if args.no_droppath else 0.9
     h1, l1 = run_once(args.steps, keep)
     h2, l2 = run_once(args.steps, keep)
