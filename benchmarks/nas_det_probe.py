@@ -34,6 +34,22 @@ def disable_native(classes):
             x, self.weight.to(x.dtype),
             self.bias.to(x.dtype) if self.bias is not None else None,
             self.stride, self.padding)
+    if "bn" in classes:
+        from adanet_amd.ops import batchnorm as abn
+        def _bn_fwd(self, x):
+            return F.batch_norm(
+                x.float(), self.running_mean, self.running_var,
+                self.weight.float() if self.weight is not None else None,
+                self.bias.float() if self.bias is not None else None,
+                self.training, self.momentum, self.eps).to(x.dtype)
+        abn.HipBatchNorm2d.forward = _bn_fwd
+    if "fc" in classes:
+        from adanet_amd.ops import linear as alin
+        def _fc_fwd(self, x):
+            b = self.bias
+            return F.linear(x, self.weight.to(x.dtype),
+                            b.to(x.dtype) if b is not None else None)
+        alin.HipLinear.forward = _fc_fwd
     if "pool" in classes:
         aconv.HipPool2d.forward = lambda self, x: (
             F.max_pool2d(x, 3, self.stride, 1) if self.kind == "max"
