@@ -236,6 +236,13 @@ __global__ __launch_bounds__(WGM * WGN * 64, MINWAVES) void gemm_tr_kernel(
   STAGE_B(0, 0);
   int buf = 0;
   for (int kt = 0; kt < ktiles; ++kt) {
+    // Explicit per-wave drain of the LDS-DMA issued last iteration:
+    // __syncthreads() alone does not reliably order global_load_lds
+    // completion against cross-wave LDS reads (the tr path reads via
+    // inline-asm ds_read, invisible to the compiler's waitcnt insertion
+    // -- observed as a run-to-run race in the NASNet 1x1 conv batched
+    // GEMMs, benchmarks/nas_det_probe.py).
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     if (kt + 1 < ktiles) {
       STAGE_A(buf ^ 1, (kt + 1) * KSTEP);
