@@ -85,6 +85,46 @@ def run_once(steps, drop_path_keep, dtype=torch.bfloat16):
     return h.hexdigest(), losses
 
 
+def gradcmp(steps, keep, reps):
+    """Run fwd+bwd repeatedly on FIXED weights; report every parameter
+    whose gradient bytes differ across reps (pinpoints the racy op)."""
+    torch.manual_seed(7)
+    model = NasNetCIFAR(num_cells=3, num_conv_filters=32,
+                        drop_path_keep=keep).cuda()
+    from adanet_amd.ops.linear import restore_fp32_params
+    model = model.to(torch.bfloat16)
+    restore_fp32_params(model)
+    model.train()
+    torch.manual_seed(13)
+    x0 = (torch.randn(64, 3, 32, 32, device="cuda") / 4).to(torch.bfloat16)
+    y = torch.randint(0, 10, (64,), device="cuda")
+    ref = None
+    bad = {}
+    for rep in range(reps):
+        for p in model.parameters():
+            p.grad = None
+        torch.manual_seed(99)  # identical drop_path draws every rep
+        x = x0.clone().requires_grad_(True)
+        _, logits = model(x)
+        loss = torch.nn.functional.cross_entropy(logits.float(), y)
+        loss.backward()
+        snap = {n: p.grad.detach().clone()
+                for n, p in model.named_parameters() if p.grad is not None}
+        snap["__x__"] = x.grad.detach().clone()
+        snap["__loss__"] = loss.detach().clone()
+        if ref is None:
+            ref = snap
+        else:
+            for n, g in snap.items():
+                if not torch.equal(ref[n], g):
+                    bad.setdefault(n, 0)
+                    bad[n] += 1
+    print("diverging tensors (%d):" % len(bad))
+    for n, c in sorted(bad.items()):
+        print("  %s  (%d/%d reps)" % (n, c, reps - 1))
+    return len(bad)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--steps", type=int, default=10)
@@ -93,10 +133,13 @@ def main():
                     help="comma list of dw,c1,cn,pool to force torch fallback")
     ap.add_argument("--reps", type=int, default=1,
                     help="repeat the A/B comparison; races are flaky")
+    ap.add_argument("--gradcmp", action="store_true")
     args = ap.parse_args()
     if args.disable:
         disable_native(args.disable.split(","))
     keep = 1.0 if args.no_droppath else 0.9
+    if args.gradcmp:
+        sys.exit(1 if gradcmp(args.steps, keep, args.reps) else 0)
     mism = 0
     ref_h, ref_l = run_once(args.steps, keep)
     for _ in range(args.reps):
