@@ -374,6 +374,15 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
       std::min<int64_t>(512 / t64, (int64_t)K / 256) >= 2) {
     int ksplit = (int)std::min<int64_t>(512 / t64, (int64_t)K / 256);
     if (ksplit < 2) ksplit = 2;
+    // Clamp so every split owns >=1 K-tile: an empty split would leave its
+    // C32 slot UNWRITTEN and the epilogue would sum garbage (observed as
+    // the NASNet 1x1 dW nondeterminism, benchmarks/nas_det_probe.py).
+    {
+      const int ktiles = K / 32;
+      const int per = (ktiles + ksplit - 1) / ksplit;
+      ksplit = (ktiles + per - 1) / per;
+      if (ksplit < 2) ksplit = 2;
+    }
     auto C32 = at::empty({ksplit, M, N}, A.options().dtype(at::kFloat));
     const int mt = (M + 63) / 64, nt = (N + 63) / 64;
     hipLaunchKernelGGL((gemm_nt_bf16_kernel<64, 64, 2, 2, 6, 32, true>),
