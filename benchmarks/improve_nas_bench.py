@@ -10,6 +10,7 @@ at one). channel_multiple=32 engages the fully-native conv stack
 python benchmarks/improve_nas_bench.py [--steps 3 --warmup 1] [--out f.json]
 """
 import argparse
+import os
 import json
 import sys
 import tempfile
@@ -25,6 +26,12 @@ from adanet_amd.models.cifar import Cifar10Provider
 
 
 def main():
+    # Candidate HIP streams are disabled here: with streams on, the NASNet
+    # space trains nondeterministically AND worse (final accuracy 0.48-0.80
+    # vs a bit-identical 0.93 across runs with streams off; ~9% step-time
+    # cost). The DNN headline bench is bit-deterministic WITH streams; the
+    # NASNet interaction is an open round-3 item (TODO_ROUND3.md).
+    os.environ.setdefault("ADANET_NO_STREAMS", "1")
     p = argparse.ArgumentParser()
     p.add_argument("--steps", type=int, default=3)
     p.add_argument("--warmup", type=int, default=1)
