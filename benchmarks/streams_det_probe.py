@@ -1,0 +1,84 @@
+"""Reproduce the streams x NASNet nondeterminism outside the estimator:
+two NasNetCIFAR candidates trained in the engine's exact stream pattern
+(per-candidate HIP stream, default-stream event at step start, device
+join at step end), repeated with identical seeds; parameter bytes
+compared across reps.
+
+python benchmarks/streams_det_probe.py [--steps 8] [--reps 4]
+"""
+
+import argparse
+import hashlib
+import sys
+
+import torch
+
+sys.path.insert(0, ".")
+from adanet_amd.models.nasnet import NasNetCIFAR  # noqa: E402
+
+
+def run_once(steps, use_streams, keep):
+    torch.manual_seed(7)
+    models = [NasNetCIFAR(num_cells=3, num_conv_filters=32,
+                          drop_path_keep=keep).cuda(),
+              NasNetCIFAR(num_cells=4, num_conv_filters=32,
+                          drop_path_keep=keep).cuda()]
+    from adanet_amd.ops.linear import restore_fp32_params
+    for m in models:
+        m.to(torch.bfloat16)
+        restore_fp32_params(m)
+        m.train()
+    opts = [torch.optim.SGD(m.parameters(), lr=0.025, momentum=0.9)
+            for m in models]
+    torch.manual_seed(13)
+    x = (torch.randn(64, 3, 32, 32, device="cuda") / 4).to(torch.bfloat16)
+    y = torch.randint(0, 10, (64,), device="cuda")
+    streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+    cur = torch.cuda.current_stream()
+    for _ in range(steps):
+        ev = torch.cuda.Event()
+        ev.record()
+        for m, o, s in zip(models, opts, streams):
+            if use_streams:
+                ev.wait(s)
+                ctx = torch.cuda.stream(s)
+            else:
+                import contextlib
+                ctx = contextlib.nullcontext()
+            with ctx:
+                o.zero_grad(set_to_none=True)
+                _, logits = m(x)
+                loss = torch.nn.functional.cross_entropy(logits.float(), y)
+                loss.backward()
+                o.step()
+        if use_streams:
+            for s in streams:
+                cur.wait_stream(s)
+    torch.cuda.synchronize()
+    h = hashlib.sha256()
+    for m in models:
+        for name, p in sorted(m.named_parameters()):
+            h.update(p.detach().float().cpu().numpy().tobytes())
+    return h.hexdigest()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--steps", type=int, default=8)
+    ap.add_argument("--reps", type=int, default=4)
+    ap.add_argument("--no-streams", action="store_true")
+    ap.add_argument("--no-droppath", action="store_true")
+    args = ap.parse_args()
+    keep = 1.0 if args.no_droppath else 0.9
+    ref = run_once(args.steps, not args.no_streams, keep)
+    mism = 0
+    for _ in range(args.reps):
+        if run_once(args.steps, not args.no_streams, keep) != ref:
+            mism += 1
+    print("mismatches: %d/%d (streams=%s droppath=%s)" %
+          (mism, args.reps, not args.no_streams, keep < 1.0))
+    sys.exit(1 if mism else 0)
+
+
+if __name__ == "__main__":
+    main()
