@@ -17,7 +17,7 @@ sys.path.insert(0, ".")
 from adanet_amd.models.nasnet import NasNetCIFAR  # noqa: E402
 
 
-def run_once(steps, use_streams, keep):
+def run_once(steps, use_streams, keep, engine_opt=False):
     torch.manual_seed(7)
     models = [NasNetCIFAR(num_cells=3, num_conv_filters=32,
                           drop_path_keep=keep).cuda(),
@@ -28,8 +28,15 @@ def run_once(steps, use_streams, keep):
         m.to(torch.bfloat16)
         restore_fp32_params(m)
         m.train()
-    opts = [torch.optim.SGD(m.parameters(), lr=0.025, momentum=0.9)
-            for m in models]
+    if engine_opt:
+        from adanet_amd.ops.optim import CosineLR, FusedSGD
+        opts = [FusedSGD(list(m.parameters()), lr=0.025, momentum=0.9,
+                         weight_decay=5e-4) for m in models]
+        scheds = [CosineLR(o, steps) for o in opts]
+    else:
+        opts = [torch.optim.SGD(m.parameters(), lr=0.025, momentum=0.9)
+                for m in models]
+        scheds = []
     torch.manual_seed(13)
     x = (torch.randn(64, 3, 32, 32, device="cuda") / 4).to(torch.bfloat16)
     y = torch.randint(0, 10, (64,), device="cuda")
@@ -49,8 +56,15 @@ def run_once(steps, use_streams, keep):
                 o.zero_grad(set_to_none=True)
                 _, logits = m(x)
                 loss = torch.nn.functional.cross_entropy(logits.float(), y)
-                loss.backward()
+                if engine_opt:
+                    from adanet_amd.ops.linear import direct_grad_writes
+                    with direct_grad_writes():
+                        loss.backward()
+                else:
+                    loss.backward()
                 o.step()
+        for sc in scheds:
+            sc.step()
         if use_streams:
             for s in streams:
                 cur.wait_stream(s)
@@ -68,15 +82,19 @@ def main():
     ap.add_argument("--reps", type=int, default=4)
     ap.add_argument("--no-streams", action="store_true")
     ap.add_argument("--no-droppath", action="store_true")
+    ap.add_argument("--engine-opt", action="store_true",
+                    help="FusedSGD + CosineLR + direct-to-arena grads")
     args = ap.parse_args()
     keep = 1.0 if args.no_droppath else 0.9
-    ref = run_once(args.steps, not args.no_streams, keep)
+    ref = run_once(args.steps, not args.no_streams, keep, args.engine_opt)
     mism = 0
     for _ in range(args.reps):
-        if run_once(args.steps, not args.no_streams, keep) != ref:
+        if run_once(args.steps, not args.no_streams, keep,
+                    args.engine_opt) != ref:
             mism += 1
-    print("mismatches: %d/%d (streams=%s droppath=%s)" %
-          (mism, args.reps, not args.no_streams, keep < 1.0))
+    print("mismatches: %d/%d (streams=%s droppath=%s engine_opt=%s)" %
+          (mism, args.reps, not args.no_streams, keep < 1.0,
+           args.engine_opt))
     sys.exit(1 if mism else 0)
 
 
