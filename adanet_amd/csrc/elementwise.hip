@@ -82,16 +82,20 @@ __global__ __launch_bounds__(256) void relu_bwd_colsum_kernel(
   wsp[(int64_t)blockIdx.y * C + c] = acc;
 }
 
-// db[c] += sum_chunk wsp[chunk][c] in fixed chunk order (accumulate
-// semantics: db is the live arena grad view, never pre-zeroed here).
+// db[c] += sum_chunk wsp[chunk][c], chunks reduced lane-parallel within
+// one wave per column then wave-shuffled in FIXED TREE ORDER (still
+// deterministic). Thread-per-column serialized nchunks loads on a
+// near-empty chip (was 11% of the DNN step, bench_kernel_stats_r02c).
 __global__ __launch_bounds__(256) void colsum_chunk_reduce_kernel(
     const float* __restrict__ wsp, float* __restrict__ db, int C,
     int nchunks) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
   if (c >= C) return;
   float s = 0.f;
-  for (int r = 0; r < nchunks; ++r) s += wsp[(int64_t)r * C + c];
-  db[c] += s;
+  for (int r = lane; r < nchunks; r += 64) s += wsp[(int64_t)r * C + c];
+  s = wave_reduce_sum(s);
+  if (lane == 0) db[c] += s;
 }
 
 // Batched device-to-device copy: up to 8 (src,dst) pairs per launch
@@ -216,8 +220,9 @@ void relu_bwd_colsum(const at::Tensor& dy, const at::Tensor& y,
                      (const bf16_t*)y.data_ptr(), (bf16_t*)dz.data_ptr(),
                      wsp.data_ptr<float>(), B, C, rows_per_block,
                      (float)scale);
-  hipLaunchKernelGGL(colsum_chunk_reduce_kernel, dim3((unsigned)stripes),
-                     dim3(256), 0, stream.stream(), wsp.data_ptr<float>(),
+  hipLaunchKernelGGL(colsum_chunk_reduce_kernel,
+                     dim3((unsigned)((C + 3) / 4)), dim3(256), 0,
+                     stream.stream(), wsp.data_ptr<float>(),
                      db.data_ptr<float>(), C, row_chunks);
   HIP_CHECK_KERNEL();
 }

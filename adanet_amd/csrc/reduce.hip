@@ -31,16 +31,23 @@ __global__ __launch_bounds__(256) void colsum_bf16_kernel(
   }
 }
 
-// out[c] (+)= sum_chunk wsp[chunk][c] in fixed chunk order.
+// out[c] (+)= sum_chunk wsp[chunk][c]: one wave per column, lanes over
+// chunks, fixed-tree wave reduce — deterministic and chip-filling (the
+// thread-per-column version serialized nchunks loads; at C=10 it ran 10
+// threads x 256 loads and was 10% of the DNN step).
 __global__ __launch_bounds__(256) void colsum_reduce_kernel(
     const float* __restrict__ wsp, float* __restrict__ out, int C,
     int nchunks, int accum) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  const int c = blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
   if (c >= C) return;
   float s = 0.f;
-  for (int r = 0; r < nchunks; ++r) s += wsp[(int64_t)r * C + c];
-  if (accum) out[c] += s;
-  else out[c] = s;
+  for (int r = lane; r < nchunks; r += 64) s += wsp[(int64_t)r * C + c];
+  s = wave_reduce_sum(s);
+  if (lane == 0) {
+    if (accum) out[c] += s;
+    else out[c] = s;
+  }
 }
 
 // argmax over the class dim + count of matches with labels (accuracy numer).
@@ -129,7 +136,7 @@ void colsum_bf16(const at::Tensor& x, at::Tensor& out, int64_t accum) {
                        (const bf16_t*)x.data_ptr(), out.data_ptr<float>(), B,
                        C, (int)x.stride(0), rows_per_block, (int)accum,
                        wsp.data_ptr<float>());
-    hipLaunchKernelGGL(colsum_reduce_kernel, dim3((unsigned)stripes),
+    hipLaunchKernelGGL(colsum_reduce_kernel, dim3((unsigned)((C + 3) / 4)),
                        dim3(256), 0, stream.stream(),
                        wsp.data_ptr<float>(), out.data_ptr<float>(), C,
                        row_chunks, (int)accum);
