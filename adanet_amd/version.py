@@ -3,4 +3,4 @@
 Mirrors the reference's version module (reference: adanet/version.py:3).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.9.0"

@@ -31,8 +31,16 @@ SRC = [
     "adanet_amd/csrc/depthwise.hip",
 ]
 
+def _version():
+    ns = {}
+    with open("adanet_amd/version.py") as f:
+        exec(f.read(), ns)
+    return ns["__version__"]
+
+
 setup(
-    name="adanet_amd_hip",
+    name="adanet-amd",
+    version=_version(),
     ext_modules=[
         cpp_extension.CUDAExtension(
             name="adanet_amd._adanet_hip",
