@@ -17,7 +17,7 @@ sys.path.insert(0, ".")
 from adanet_amd.models.nasnet import NasNetCIFAR  # noqa: E402
 
 
-def run_once(steps, use_streams, keep, engine_opt=False):
+def run_once(steps, use_streams, keep, engine_opt=False, engine_full=False):
     torch.manual_seed(7)
     models = [NasNetCIFAR(num_cells=3, num_conv_filters=32,
                           drop_path_keep=keep).cuda(),
@@ -42,6 +42,23 @@ def run_once(steps, use_streams, keep, engine_opt=False):
     y = torch.randint(0, 10, (64,), device="cuda")
     streams = [torch.cuda.Stream(), torch.cuda.Stream()]
     cur = torch.cuda.current_stream()
+    head = mixers = mix_opts = frozen = loss_buf = loss_ctr = None
+    if engine_full:
+        # engine extras: fused-xent head, per-candidate mixer over
+        # [frozen static, own logits], device loss ring.
+        from adanet_amd.head import MultiClassHead
+        from adanet_amd.ops.mixer import weighted_sum_logits
+        from adanet_amd.ops.optim import FusedSGD
+        head = MultiClassHead(10)
+        frozen = (torch.randn(64, 10, device="cuda") / 4).to(torch.bfloat16)
+        mixers = [torch.nn.ParameterDict({
+            "w": torch.nn.Parameter(torch.full((2,), 0.5, device="cuda")),
+            "b": torch.nn.Parameter(torch.zeros(10, device="cuda"))})
+            for _ in models]
+        mix_opts = [FusedSGD([m["w"], m["b"]], lr=0.01) for m in mixers]
+        loss_buf = torch.full((64, 4), float("nan"), device="cuda")
+        loss_ctr = torch.zeros((), device="cuda", dtype=torch.long)
+        globals()["_wsl"] = weighted_sum_logits
     for _ in range(steps):
         ev = torch.cuda.Event()
         ev.record()
@@ -63,16 +80,38 @@ def run_once(steps, use_streams, keep, engine_opt=False):
                 else:
                     loss.backward()
                 o.step()
+                if engine_full:
+                    mi = models.index(m)
+                    md, mo = mixers[mi], mix_opts[mi]
+                    from adanet_amd.ops.linear import direct_grad_writes
+                    wsl = globals()["_wsl"]
+                    mixed = wsl([frozen, logits.detach()], md["w"], md["b"])
+                    mloss = head.loss(mixed, y) + 1e-3 * md["w"].abs().sum()
+                    mo.zero_grad(set_to_none=True)
+                    with direct_grad_writes():
+                        mloss.backward()
+                    mo.step()
         for sc in scheds:
             sc.step()
         if use_streams:
             for s in streams:
                 cur.wait_stream(s)
+        if engine_full:
+            row = torch.stack([m["w"].detach().sum() for m in mixers] +
+                              [m["b"].detach().sum() for m in mixers])
+            idx = torch.remainder(loss_ctr, 64)
+            loss_buf.index_copy_(0, idx.reshape(1), row.reshape(1, 4))
+            loss_ctr.add_(1)
     torch.cuda.synchronize()
     h = hashlib.sha256()
     for m in models:
         for name, p in sorted(m.named_parameters()):
             h.update(p.detach().float().cpu().numpy().tobytes())
+    if engine_full:
+        for md in mixers:
+            h.update(md["w"].detach().float().cpu().numpy().tobytes())
+            h.update(md["b"].detach().float().cpu().numpy().tobytes())
+        h.update(loss_buf.cpu().numpy().tobytes())
     return h.hexdigest()
 
 
@@ -84,17 +123,20 @@ def main():
     ap.add_argument("--no-droppath", action="store_true")
     ap.add_argument("--engine-opt", action="store_true",
                     help="FusedSGD + CosineLR + direct-to-arena grads")
+    ap.add_argument("--engine-full", action="store_true",
+                    help="+ fused-xent head, mixer ensemble loss, loss ring")
     args = ap.parse_args()
     keep = 1.0 if args.no_droppath else 0.9
-    ref = run_once(args.steps, not args.no_streams, keep, args.engine_opt)
+    ref = run_once(args.steps, not args.no_streams, keep, args.engine_opt,
+                   args.engine_full)
     mism = 0
     for _ in range(args.reps):
         if run_once(args.steps, not args.no_streams, keep,
-                    args.engine_opt) != ref:
+                    args.engine_opt, args.engine_full) != ref:
             mism += 1
-    print("mismatches: %d/%d (streams=%s droppath=%s engine_opt=%s)" %
+    print("mismatches: %d/%d (streams=%s droppath=%s engine_opt=%s full=%s)" %
           (mism, args.reps, not args.no_streams, keep < 1.0,
-           args.engine_opt))
+           args.engine_opt, args.engine_full))
     sys.exit(1 if mism else 0)
 
 
