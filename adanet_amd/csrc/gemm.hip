@@ -233,6 +233,31 @@ __global__ __launch_bounds__(256) void gemm_splitk_epilogue_kernel(
   }
 }
 
+// High-ksplit epilogue twin: one WAVE per output element, lanes over
+// the ksplit slots, fixed-tree wave reduce (deterministic). The
+// thread-per-element kernel serialized ksplit loads on a near-empty
+// chip for the skinny conv-dW shapes (M*N ~5k, ksplit ~150: 30 us
+// where traffic is ~1 us — 4.4% of the NASNet step,
+// profiles/nasprof7_summary.txt).
+__global__ __launch_bounds__(256) void gemm_splitk_epilogue_wave_kernel(
+    const float* __restrict__ C32, const float* __restrict__ bias,
+    bf16_t* __restrict__ C, int M, int N, int ldc, int act, int ksplit) {
+  const int64_t total = (int64_t)M * N;
+  const int64_t p = (int64_t)blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  if (p >= total) return;
+  float v = 0.f;
+  for (int r = lane; r < ksplit; r += 64) v += C32[(int64_t)r * total + p];
+  v = wave_reduce_sum(v);
+  if (lane == 0) {
+    const int m = (int)(p / N), n = (int)(p % N);
+    v += bias ? bias[n] : 0.f;
+    if (act == 1) v = v > 0.f ? v : 0.f;
+    if (act == 2) v += bf2f(C[(int64_t)m * ldc + n]);
+    C[(int64_t)m * ldc + n] = f2bf(v);
+  }
+}
+
 // GEMV path for skinny-M GEMMs (M <= 8: batch-1/small-batch serving,
 // where an MFMA tile would waste 63/64 rows): one wave per output column,
 // lanes stride K with 16 B loads, wave-shuffle reduction. Bandwidth-bound
@@ -390,10 +415,19 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
                        stream.stream(), a, b, c, nullptr, M, N, K, lda, ldb,
                        ldc, (int)act, mt, nt, C32.data_ptr<float>(), ksplit);
     const int64_t tot = (int64_t)M * N;
-    hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,
-                       dim3((int)std::min<int64_t>((tot + 255) / 256, 2048)),
-                       dim3(256), 0, stream.stream(), C32.data_ptr<float>(),
-                       bias_ptr, c, M, N, ldc, (int)act, ksplit);
+    if (ksplit > 16) {
+      hipLaunchKernelGGL(gemm_splitk_epilogue_wave_kernel,
+                         dim3((unsigned)((tot + 3) / 4)), dim3(256), 0,
+                         stream.stream(), C32.data_ptr<float>(), bias_ptr, c,
+                         M, N, ldc, (int)act, ksplit);
+    } else {
+      hipLaunchKernelGGL(gemm_splitk_epilogue_kernel,
+                         dim3((int)std::min<int64_t>((tot + 255) / 256,
+                                                     2048)),
+                         dim3(256), 0, stream.stream(),
+                         C32.data_ptr<float>(), bias_ptr, c, M, N, ldc,
+                         (int)act, ksplit);
+    }
     HIP_CHECK_KERNEL();
     return;
   }

@@ -498,7 +498,13 @@ void gemm_tr_batched(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
 #define LTRB_DISPATCH(TRA, TRB)                                               \
   do {                                                                        \
     const int64_t b128 = (int64_t)((M + 127) / 128) * ((N + 127) / 128);      \
-    if (b128 >= 64) {                                                         \
+    if (M <= 32) {                                                            \
+      /* skinny-M (pointwise conv with Co=32: a 64-row tile wastes half  */   \
+      /* its MFMA work; 12.7% of the NASNet step ran on 64x64 tiles,     */   \
+      /* profiles/nasprof7_summary.txt)                                  */   \
+      if (N >= 96) LTRB(32, 128, 2, 4, 4, 1, 2, TRA, TRB);                    \
+      else LTRB(32, 64, 2, 2, 4, 1, 2, TRA, TRB);                             \
+    } else if (b128 >= 64) {                                                  \
       LTRB(128, 128, 2, 2, 4, 4, 4, TRA, TRB);                                \
     } else {                                                                  \
       LTRB(64, 64, 2, 2, 6, 2, 2, TRA, TRB);                                  \
