@@ -305,6 +305,55 @@ __global__ __launch_bounds__(256) void gemm_nt_gemv_kernel(
   }
 }
 
+// Skinny-N path (classifier logits: N=10, M=batch, K=hidden). An MFMA
+// 64-wide tile wastes (64-N)/64 of its work AND fills only M/64 blocks
+// (measured 33 us / 2.4 TF on the bench logits GEMM, 6.7% of the step,
+// profiles/bench_kernel_stats_r02d.txt). Here the WHOLE B matrix (N*K
+// bf16 <= 64 KB) is staged in LDS once per block; each wave then streams
+// rows of A with 16 B loads and keeps all N dot products in registers —
+// VALU-bound but ~10x faster than the wasted MFMA tile at these shapes.
+__global__ __launch_bounds__(256) void gemm_nt_skinnyn_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    bf16_t* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, int lda, int ldb, int ldc, int act) {
+  __shared__ bf16_t bsmem[32768];  // 64 KB: N * K <= 32768 (host gate)
+  for (int n = 0; n < N; ++n) {
+    const bf16_t* src = B + (int64_t)n * ldb;
+    bf16_t* dst = bsmem + n * K;
+    for (int i = threadIdx.x * 8; i < K; i += 256 * 8)
+      *(bfx8*)(dst + i) = *(const bfx8*)(src + i);
+  }
+  __syncthreads();
+  const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+  for (int m = blockIdx.x * 4 + wid; m < M; m += gridDim.x * 4) {
+    float acc[16];
+#pragma unroll
+    for (int n = 0; n < 16; ++n) acc[n] = 0.f;
+    const bf16_t* arow = A + (int64_t)m * lda;
+    for (int k = lane * 8; k < K; k += 64 * 8) {
+      const bfx8 av = *(const bfx8*)(arow + k);
+      float af[8];
+#pragma unroll
+      for (int i = 0; i < 8; ++i) af[i] = bf2f(av.v[i]);
+      for (int n = 0; n < N; ++n) {
+        const bfx8 bv = *(const bfx8*)(bsmem + n * K + k);
+#pragma unroll
+        for (int i = 0; i < 8; ++i) acc[n] += af[i] * bf2f(bv.v[i]);
+      }
+    }
+    for (int n = 0; n < N; ++n) {
+      float v = wave_reduce_sum(acc[n]);
+      if (lane == 0) {
+        if (bias) v += bias[n];
+        if (act == 1) v = v > 0.f ? v : 0.f;
+        bf16_t* cp = &C[(int64_t)m * ldc + n];
+        if (act == 2) v += bf2f(*cp);
+        *cp = f2bf(v);
+      }
+    }
+  }
+}
+
 // Generic any-stride fallback (correctness net for shapes the fast path
 // can't take: lda/ldb not 8-aligned or K not a multiple of 32). VALU fp32.
 __global__ void gemm_nt_generic_kernel(const bf16_t* __restrict__ A,
@@ -383,6 +432,14 @@ void gemm_nt_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
     hipLaunchKernelGGL(gemm_nt_gemv_kernel, dim3((N + 3) / 4), dim3(256), 0,
                        stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
                        ldc, (int)act, seed_ptr, pthresh, inv_keep);
+    HIP_CHECK_KERNEL();
+    return;
+  }
+  if (fast && act != 3 && N <= 16 && (int64_t)N * K <= 32768 && M >= 64) {
+    hipLaunchKernelGGL(gemm_nt_skinnyn_kernel,
+                       dim3(std::min((M + 3) / 4, 2048)), dim3(256), 0,
+                       stream.stream(), a, b, c, bias_ptr, M, N, K, lda, ldb,
+                       ldc, (int)act);
     HIP_CHECK_KERNEL();
     return;
   }

@@ -430,7 +430,12 @@ void gemm_tr_bf16(const at::Tensor& A, const at::Tensor& B, at::Tensor& C,
           bias_ptr, M, N, K, lda, ldb, ldc, (int)act, mt, nt);                \
     }                                                                         \
   } while (0)
-  if (M >= N && t_256x128 >= 512) {
+  if (M <= 32 && N >= 256) {
+    // skinny-M (classifier dW: M=n_classes=10, N=hidden): a 64-row tile
+    // wastes >=half its MFMA work and fills M/64 of the row grid.
+    if (N >= 96) LTR(32, 128, 2, 4, 4, 1, 2);
+    else LTR(32, 64, 2, 2, 4, 1, 2);
+  } else if (M >= N && t_256x128 >= 512) {
     // 1101 TF @4096^3 tt (profiles/tr_variants_r01.json)
     LTR(256, 128, 4, 4, 2, 4, 2);
   } else if (N > M && t_128x256 >= 512) {
