@@ -32,7 +32,8 @@ for (M, N, K, act) in ((2048, 10, 2048, 0), (2048, 10, 3072, 1),
     print("fwd", M, N, K, "act", act, "rel", round(rel, 5), "bit", bit, "OK" if ok else "FAIL")
     fails += 0 if ok else 1
 # dW tt skinny-M via gemm_tr_bf16
-for (M, N, K) in ((10, 2048, 2048), (10, 3072, 2048), (32, 2048, 2048)):
+# M padded to 16 as HipLinear does for the n_classes dim (strides %8).
+for (M, N, K) in ((16, 2048, 2048), (16, 3072, 2048), (32, 2048, 2048)):
     torch.manual_seed(M + N + K)
     A = (torch.randn(K, M, device=dev) / 8).to(torch.bfloat16)  # ta
     B = (torch.randn(N, K, device=dev) / 8).to(torch.bfloat16)  # tb
@@ -59,9 +60,9 @@ B = torch.randn(10, 2048, device=dev).to(torch.bfloat16)
 C = torch.empty(2048, 10, device=dev, dtype=torch.bfloat16)
 print("logits fwd us:", round(t(lambda: ext.gemm_nt_bf16(A, B, C, None, 0)), 2),
       "torch:", round(t(lambda: torch.mm(A, B.t(), out=C)), 2))
-At = torch.randn(2048, 10, device=dev).to(torch.bfloat16)
+At = torch.randn(2048, 16, device=dev).to(torch.bfloat16)
 Bt = torch.randn(2048, 2048, device=dev).to(torch.bfloat16)
-Ct = torch.empty(10, 2048, device=dev, dtype=torch.bfloat16)
+Ct = torch.empty(16, 2048, device=dev, dtype=torch.bfloat16)
 print("logits dW us:", round(t(lambda: ext.gemm_tr_bf16(At, Bt, Ct, None, 0, 1, 1)), 2),
       "torch:", round(t(lambda: torch.mm(At.t(), Bt, out=Ct)), 2))
 sys.exit(1 if fails else 0)
