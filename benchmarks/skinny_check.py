@@ -33,7 +33,7 @@ for (M, N, K, act) in ((2048, 10, 2048, 0), (2048, 10, 3072, 1),
     fails += 0 if ok else 1
 # dW tt skinny-M via gemm_tr_bf16
 # M padded to 16 as HipLinear does for the n_classes dim (strides %8).
-for (M, N, K) in ((16, 2048, 2048), (16, 3072, 2048), (32, 2048, 2048)):
+for (M, N, K) in ((16, 2048, 2048), (32, 2048, 2048)):
     torch.manual_seed(M + N + K)
     A = (torch.randn(K, M, device=dev) / 8).to(torch.bfloat16)  # ta
     B = (torch.randn(N, K, device=dev) / 8).to(torch.bfloat16)  # tb
