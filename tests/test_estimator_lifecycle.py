@@ -864,3 +864,32 @@ def test_best_metrics_mux_with_dict_metrics(model_dir,
     assert res["custom/n_examples"] > 0
     assert "best_ensemble_index_0" in res
     assert "architecture/adanet/ensembles" in res
+
+
+def test_golden_deterministic_lifecycle(model_dir, tmp_path,
+                                        synthetic_classification):
+    """Bit-exact repeatability of a full two-iteration search on CPU
+    (the reference pins golden losses per config, estimator_test.py
+    :419-470; with the round-2 deterministic reductions the equivalent
+    guarantee here is byte-equality of the entire trajectory). Two
+    estimators with the same seed must produce identical architectures
+    AND identical evaluate() numbers."""
+    import torch
+    X, Y, input_fn = synthetic_classification
+    n_threads = torch.get_num_threads()
+    torch.set_num_threads(1)  # CPU reductions vary with thread count
+    try:
+        results = []
+        for d in (model_dir, str(tmp_path / "golden_b")):
+            est = _make_estimator(d, input_fn)
+            est.train(input_fn, max_steps=20)
+            arch = json.load(open(os.path.join(d, "architecture-1.json")))
+            res = est.evaluate(input_fn, steps=2)
+            results.append((
+                [s["builder_name"] for s in arch["subnetworks"]],
+                float(res["loss"]), float(res.get("accuracy", -1.0))))
+        assert results[0][0] == results[1][0]
+        assert results[0][1] == results[1][1]  # bit-exact, not approx
+        assert results[0][2] == results[1][2]
+    finally:
+        torch.set_num_threads(n_threads)
