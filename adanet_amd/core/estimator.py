@@ -948,6 +948,15 @@ class Estimator(object):
     def _execute_bookkeeping_phase(self, input_fn):
         iteration = self._current_iteration
         t = iteration.number
+        # Iteration-boundary barrier: the training phase's per-candidate
+        # stream work is device-joined after every step, but eval /
+        # selection / winner-freeze below interleave default-stream reads
+        # with the end of candidate-stream work — a full synchronize here
+        # removes the entire class of transition-ordering hazards for the
+        # cost of one host sync per boosting iteration (hardening for the
+        # streams interaction tracked in TODO_ROUND3.md #12).
+        if self._device.type == "cuda":
+            torch.cuda.synchronize(self._device)
 
         # (1) candidate selection (reference :1285-1329, 1415-1517).
         replay_index = None
