@@ -62,3 +62,48 @@ def test_auc_streaming_equals_single_batch(seed):
     for i in range(0, 256, 64):
         stream.update(s[i:i + 64], y[i:i + 64])
     assert torch.equal(whole._hist, stream._hist)
+
+
+@given(st.lists(st.tuples(st.text(min_size=1, max_size=20),
+                          st.floats(allow_nan=False, allow_infinity=False,
+                                    width=32)),
+                min_size=1, max_size=30))
+def test_tfevents_roundtrip_arbitrary_scalars(pairs):
+    """CRC-framed tfevents survive arbitrary tag strings and fp32 values
+    (reference summary writers guarantee the event file parses back)."""
+    import tempfile
+
+    from adanet_amd.core.tb_writer import TBEventWriter, read_tfevents
+    with tempfile.TemporaryDirectory() as d:
+        w = TBEventWriter(d)
+        for i, (tag, val) in enumerate(pairs):
+            w.scalar(tag, val, step=i)
+        events = read_tfevents(w.path)
+    scalars = [(val["tag"], val["simple_value"]) for ev in events
+               for val in ev.get("values", [])
+               if "simple_value" in val]
+    assert len(scalars) == len(pairs)
+    import struct
+    for (tag0, v0), (tag1, v1) in zip(pairs, scalars):
+        assert tag0 == tag1
+        assert struct.unpack("f", struct.pack("f", v0))[0] == v1
+
+
+@given(st.lists(st.tuples(st.integers(0, 30), st.text(
+    alphabet=st.characters(blacklist_categories=("Cs",)), min_size=1,
+    max_size=12)), min_size=1, max_size=10),
+    st.lists(st.integers(0, 20), max_size=5))
+def test_architecture_roundtrip_arbitrary(subs, replay):
+    """architecture-t.json serialize/deserialize is the identity for any
+    (iteration, builder-name) history incl. unicode names (reference
+    architecture.py serialize :132 / deserialize :153)."""
+    from adanet_amd.core.architecture import _Architecture
+    a = _Architecture("c", "e")
+    for it, name in subs:
+        a.add_subnetwork(it, name)
+    for idx in replay:
+        a.add_replay_index(idx)
+    blob = a.serialize(iteration_number=7, global_step=123)
+    b = _Architecture.deserialize(blob)
+    assert b.subnetworks == a.subnetworks
+    assert b.replay_indices == a.replay_indices
