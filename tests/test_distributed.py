@@ -471,3 +471,34 @@ def test_bench_config_evaluator_roundrobin(tmp_path, world):
         assert results[r] == (a0, 2), "rank %d diverged" % r
     arch1 = json.loads(a0[1])
     assert len(arch1["subnetworks"]) == 2  # force_grow: +1 member/iter
+
+
+def _run_bench_like(tmp_path, world, tag):
+    model_dir = str(tmp_path / ("det_" + tag))
+    os.makedirs(model_dir, exist_ok=True)
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_bench_like_worker,
+                         args=(r, world, port, model_dir, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(world):
+        rank, err, archs, it = _get(q)
+        assert err is None, "rank %s failed:\n%s" % (rank, err)
+        results[rank] = (archs, it)
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    return results[0]
+
+
+def test_distributed_run_to_run_determinism(tmp_path):
+    """The round-2 determinism guarantee extends to world>1: two
+    identical 2-rank searches (gloo) produce byte-identical architecture
+    files (winner sequence AND serialized metadata)."""
+    a = _run_bench_like(tmp_path, 2, "a")
+    b = _run_bench_like(tmp_path, 2, "b")
+    assert a == b
